@@ -1,0 +1,4 @@
+from .datasets import FooDataset, SyntheticImageDataset, build_dataset
+from .sampler import ShardedSampler
+
+__all__ = ["FooDataset", "SyntheticImageDataset", "build_dataset", "ShardedSampler"]
