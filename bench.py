@@ -1,0 +1,164 @@
+"""Flagship benchmark — ResNet-18 / synthetic CIFAR-shape DDP training step.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+(for N>1 the driver launches this under torch.distributed.run with one rank
+per GPU over RCCL).  Rank 0 prints ONE JSON line with the whole-job
+samples/sec on the BASELINE.json metric/config.
+
+The timed step is the full training step the engine runs: forward, fused
+cross-entropy, backward with the native C++ bucket reducer's overlapped
+RCCL all-reduce, global grad-norm clip, fused multi-tensor SGD(momentum)
+with fp32 master weights, LR scheduler tick.  bf16 compute, synthetic data,
+random-init weights (no datasets/checkpoints are downloadable here).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from pytorch_ddp_template_amd.models import build_model  # noqa: E402
+from pytorch_ddp_template_amd.ops import CrossEntropyLoss  # noqa: E402
+from pytorch_ddp_template_amd.optim import (  # noqa: E402
+    SGD,
+    clip_grad_norm_,
+    get_linear_schedule_with_warmup,
+)
+from pytorch_ddp_template_amd.parallel import DistributedModel  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--model", type=str, default="resnet18")
+    p.add_argument("--batch", type=int, default=1024, help="per-GPU batch")
+    p.add_argument("--bucket-mb", type=int, default=25)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world_size > 1
+    if distributed:
+        torch.cuda.set_device(local_rank)
+        torch.distributed.init_process_group(backend="nccl")
+    dev = torch.device("cuda", local_rank if distributed else 0)
+    torch.manual_seed(42)
+
+    if args.model == "resnet18":
+        num_classes, img = 10, 32
+        cfg_model = "resnet18-cifar"
+    elif args.model == "resnet50":
+        num_classes, img = 1000, 224
+        cfg_model = "resnet50"
+    else:
+        num_classes, img = 1000, 224
+        cfg_model = args.model
+
+    model = build_model(args.model, num_classes).to(torch.bfloat16).to(dev)
+    opt = SGD(
+        model.parameters(), lr=0.1, momentum=0.9, weight_decay=5e-5,
+        master_weights=True,
+    )
+    sched = get_linear_schedule_with_warmup(opt, 10, 10_000)
+    if distributed:
+        model = DistributedModel(model, bucket_bytes=args.bucket_mb << 20)
+    crit = CrossEntropyLoss()
+
+    # synthetic resident batches (rotate a few so L2/L3 can't memoize inputs)
+    n_batches = 4
+    g = torch.Generator(device="cpu").manual_seed(1234 + rank)
+    xs = [
+        torch.randn(args.batch, img, img, 3, generator=g).to(torch.bfloat16).to(dev)
+        for _ in range(n_batches)
+    ]
+    ys = [
+        torch.randint(0, num_classes, (args.batch,), generator=g).to(dev)
+        for _ in range(n_batches)
+    ]
+
+    def step(i):
+        x, y = xs[i % n_batches], ys[i % n_batches]
+        out = model(x)
+        loss = crit(out, y)
+        loss.backward()
+        if distributed:
+            model.finish_gradient_sync()
+        clip_grad_norm_(list(model.parameters()), 1000.0)
+        opt.step()
+        sched.step()
+        model.zero_grad()
+        return loss
+
+    for i in range(args.warmup):
+        step(i)
+
+    if distributed:
+        torch.distributed.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    last_loss = None
+    for i in range(args.steps):
+        last_loss = step(i)
+    if distributed:
+        torch.distributed.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks (slowest rank defines the whole-job time)
+    if distributed:
+        t = torch.tensor([elapsed], device=dev, dtype=torch.float64)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world_size if distributed else 1
+    global_batch = args.batch * n_gpus
+    samples_per_sec = global_batch * args.steps / elapsed
+
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "samples/sec (whole node) ResNet-18/CIFAR-shape DDP",
+                    "value": samples_per_sec,
+                    "unit": "samples/sec",
+                    "n_gpus": n_gpus,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": elapsed / args.steps * 1000.0,
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "bf16",
+                    "data": "synthetic",
+                    "loss": float(last_loss) if last_loss is not None else None,
+                    "config": {
+                        "model": cfg_model,
+                        "global_batch": global_batch,
+                        "image": img,
+                        "num_classes": num_classes,
+                        "parallelism": f"dp{n_gpus}",
+                        "optimizer": "sgd+momentum(master fp32)",
+                    },
+                }
+            )
+        )
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

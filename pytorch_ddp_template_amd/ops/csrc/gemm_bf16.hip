@@ -1,0 +1,537 @@
+// bf16 MFMA GEMM family for gfx950 (CDNA4) — the compute core.
+//
+// One NT kernel (C[M,N] = A[M,K] @ B[N,K]^T, both operands K-contiguous, the
+// natural MFMA layout) carries:
+//   * linear forward  (Y = X W^T; W stored [N,K] — reference model.py:11,13),
+//   * linear dgrad    (dX = dY W, via the small W transpose),
+//   * conv2d fwd/dgrad as implicit GEMM (NHWC; the A operand is gathered
+//     im2col-style at LDS-staging time — MODE_CONV loader below),
+//   * batched bmm_nt for the ViT attention path.
+// A TN kernel (C[I,J] = sum_m A[m,I] B[m,J], fp32 out, split-M atomics)
+// carries linear/conv weight gradients.
+//
+// Structure (guide §5 "canonical CDNA GEMM", 2-phase schedule):
+//   128x128 tile, BK=32, 4 waves (2x2), 64x64 per wave = 4x4 fragments of
+//   v_mfma_f32_16x16x32_bf16; global->LDS staging via
+//   __builtin_amdgcn_global_load_lds width 16 (the compiler never auto-emits
+//   it); double-buffered LDS, one barrier per K-step; XOR swizzle on the
+//   k-slot (applied to the per-lane SOURCE address + the LDS read — never
+//   the LDS destination, which glds writes lane-linearly); bijective
+//   XCD-aware block swizzle for L2 locality.
+//
+// Out-of-range rows/chunks (M/N/K tails, conv padding) are redirected to a
+// 64-byte zero page so the glds path needs no branches: zeros flow through
+// the MFMA harmlessly.  K must be a multiple of 8 (host pads when not).
+
+#include <torch/extension.h>
+
+#include <unordered_map>
+
+#include "common.h"
+#include "dispatch.h"
+
+namespace g16 {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int THREADS = 256;
+constexpr int TILE_BYTES = BM * BK * 2;  // 8 KiB per operand tile
+
+// swizzle: image[row][kp] holds global chunk kp ^ swz(row); involution.
+DEV_INLINE int kswz(int row, int kp) { return kp ^ ((row >> 2) & 3); }
+
+struct ConvMeta {
+  int H, W, C_log2, S, R, stride, pad;
+  int HO, WO;
+};
+
+enum { MODE_PLAIN = 0, MODE_CONV = 1 };
+
+// ---------------------------------------------------------------- NT -----
+template <int MODE, bool RELU, bool HAS_BIAS>
+__global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    bf16* __restrict__ C, const bf16* __restrict__ bias,
+    const bf16* __restrict__ zpad, int M, int N, int K, long long strideA,
+    long long strideB, long long strideC, ConvMeta cm) {
+  __shared__ char smem[2 * 2 * TILE_BYTES];  // [buf][A|B]
+
+  // ----- block swizzle (bijective XCD remap over the x*y grid) -----
+  int nwg = gridDim.x * gridDim.y;
+  int id = blockIdx.y * gridDim.x + blockIdx.x;
+  if (nwg >= 16) {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = id & 7, idx = id >> 3;
+    id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int tx = id % gridDim.x;  // N tile
+  const int ty = id / gridDim.x;  // M tile
+  const int m0 = ty * BM, n0 = tx * BN;
+
+  const long long batch = blockIdx.z;
+  A += batch * strideA;
+  B += batch * strideB;
+  C += batch * strideC;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  // ----- per-instr staging geometry (fixed across the K loop) -----
+  // Each wave issues 2 glds per operand tile; instr chunk = 16 rows x 64 B.
+  // lane -> (row_in_chunk = lane>>2, kp = lane&3)
+  const int rl_a[2] = {(wave * 2 + 0) * 16 + (lane >> 2),
+                       (wave * 2 + 1) * 16 + (lane >> 2)};
+  const int kp = lane & 3;
+
+  // A-side per-row precompute (conv: NHWC source decomposition)
+  int a_n[2], a_hb[2], a_wb[2];
+  bool a_ok[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int gm = m0 + rl_a[i];
+    if (MODE == MODE_CONV) {
+      a_ok[i] = gm < M;
+      if (a_ok[i]) {
+        int t = gm;
+        const int wo = t % cm.WO;
+        t /= cm.WO;
+        const int ho = t % cm.HO;
+        a_n[i] = t / cm.HO;
+        a_hb[i] = ho * cm.stride - cm.pad;
+        a_wb[i] = wo * cm.stride - cm.pad;
+      }
+    } else {
+      a_ok[i] = gm < M;
+    }
+  }
+
+  const int KT = (K + BK - 1) / BK;
+
+  // ----- staging -----
+  auto stage = [&](int buf, int kt) {
+    const int k_base = kt * BK;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      // ---- A tile ----
+      const int swz = kswz(rl_a[i], kp);
+      const int gk = k_base + swz * 8;  // first element of this 16B chunk
+      const bf16* src = zpad;
+      if (MODE == MODE_CONV) {
+        if (a_ok[i] && gk < K) {
+          const int c0 = gk & ((1 << cm.C_log2) - 1);
+          const int rs = gk >> cm.C_log2;
+          const int r = rs / cm.S, s = rs % cm.S;
+          const int hi = a_hb[i] + r, wi = a_wb[i] + s;
+          if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W)
+            src = A + (((long long)a_n[i] * cm.H + hi) * cm.W + wi) *
+                          (1LL << cm.C_log2) +
+                  c0;
+        }
+      } else {
+        if (a_ok[i] && gk < K) src = A + (long long)(m0 + rl_a[i]) * K + gk;
+      }
+      char* ldsA = &smem[buf * 2 * TILE_BYTES + (wave * 2 + i) * 1024];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)ldsA, 16, 0, 0);
+      // ---- B tile (always plain rows of [N,K]) ----
+      const int rlb = rl_a[i];
+      const int gn = n0 + rlb;
+      const int swzb = kswz(rlb, kp);
+      const int gkb = k_base + swzb * 8;
+      const bf16* srcb =
+          (gn < N && gkb < K) ? B + (long long)gn * K + gkb : zpad;
+      char* ldsB =
+          &smem[buf * 2 * TILE_BYTES + TILE_BYTES + (wave * 2 + i) * 1024];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcb,
+          (__attribute__((address_space(3))) unsigned int*)ldsB, 16, 0, 0);
+    }
+  };
+
+  // ----- main loop -----
+  f32x4 acc[4][4] = {};
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * 64;
+  const int fr = lane & 15;   // fragment row (A) / col (B, D)
+  const int fs = lane >> 4;   // k-slot
+
+  stage(0, 0);
+  __syncthreads();
+  int buf = 0;
+  for (int kt = 0; kt < KT; ++kt) {
+    if (kt + 1 < KT) stage(buf ^ 1, kt + 1);
+    const char* baseA = &smem[buf * 2 * TILE_BYTES];
+    const char* baseB = baseA + TILE_BYTES;
+    bf16x8 af[4], bf[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int ra = wm + i * 16 + fr;
+      af[i] = *reinterpret_cast<const bf16x8*>(
+          baseA + ra * 64 + kswz(ra, fs) * 16);
+      const int rb = wn + i * 16 + fr;
+      bf[i] = *reinterpret_cast<const bf16x8*>(
+          baseB + rb * 64 + kswz(rb, fs) * 16);
+    }
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  // ----- epilogue: bias + relu + bf16 store -----
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni) {
+    const int col = n0 + wn + ni * 16 + fr;
+    if (col >= N) continue;
+    float bv = HAS_BIAS ? __bfloat162float(bias[col]) : 0.f;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int row_base = m0 + wm + mi * 16 + fs * 4;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row_base + r;
+        if (row >= M) continue;
+        float v = acc[mi][ni][r] + bv;
+        if (RELU) v = fmaxf(v, 0.f);
+        C[(long long)row * N + col] = __float2bfloat16(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- TN -----
+// C[I,J] (+)= sum_m A[m,I] * B[m,J], fp32 accumulation into global memory
+// via atomics (grid.z splits the M range).  Used for dW = dY^T X.
+// Staging transposes [m][i] chunks into padded LDS images [i][m] via
+// register staging + lane-paired b32 writes (glds cannot transpose).
+//
+// MODE_CONV gathers the B operand (im2col of x) on the fly for conv wgrad;
+// the j-tile then lives inside one (r,s) slice: j = c0_tile + c.
+template <int MODE>
+__global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    float* __restrict__ C, int Mtot, int I, int J, int r, int s,
+    long long ldc, long long coff, ConvMeta cm) {
+  // C element (i, j) lives at C[coff + i*ldc + j] — conv wgrad writes a
+  // (r,s) slice of dw[Kout][R*S*C], so ldc != J there.
+  constexpr int BI = 64, BJ = 64, BMC = 32;  // m-chunk
+  constexpr int ROW = BMC + 4;               // padded LDS row (elements)
+  __shared__ bf16 lds[2 * BI * ROW];         // A image then B image
+  bf16* ldsA = lds;
+  bf16* ldsB = lds + BI * ROW;
+
+  const int i0 = blockIdx.y * BI;
+  const int j0 = blockIdx.x * BJ;
+
+  // split-M: this block handles chunk range [c0, c1)
+  const int n_chunks = (Mtot + BMC - 1) / BMC;
+  const int per_z = (n_chunks + gridDim.z - 1) / gridDim.z;
+  const int ch0 = blockIdx.z * per_z;
+  const int ch1 = min(n_chunks, ch0 + per_z);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  // staging geometry: [32 m][64 cols] bf16 = 32 rows x 128 B; 256 lanes x
+  // 16 B covers it in one pass: lane -> m = tid>>3, col0 = (tid&7)*8.
+  const int sm = threadIdx.x >> 3;        // m row in chunk (0..31)
+  const int sc0 = (threadIdx.x & 7) * 8;  // first col of this 16B piece
+
+  f32x4 acc[2][2] = {};
+  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
+  const int fr = lane & 15;
+  const int fs = lane >> 4;
+
+  for (int ch = ch0; ch < ch1; ++ch) {
+    const int mbase = ch * BMC;
+    const int gm = mbase + sm;
+    // ---- load + transpose-stage A chunk ([m][I] -> image [i][m]) ----
+    {
+      bf16x8 v = {};
+      if (gm < Mtot && i0 + sc0 < I) {
+        const long long off = (long long)gm * I + i0 + sc0;
+        if (off + 8 <= (long long)Mtot * I) {
+          v = *reinterpret_cast<const bf16x8*>(A + off);
+        } else {  // last-row partial chunk: element-wise guarded load
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (off + j < (long long)Mtot * I) v[j] = A[off + j];
+        }
+      }
+      // pair lanes (sm even/odd) to write b32 [i][m..m+1]
+      short8 mine = *reinterpret_cast<short8*>(&v);
+      short8 other;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) other[j] = __shfl_xor((int)mine[j], 8);
+      if (((threadIdx.x >> 3) & 1) == 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          unsigned int pack = ((unsigned short)mine[j]) |
+                              (((unsigned int)(unsigned short)other[j]) << 16);
+          *reinterpret_cast<unsigned int*>(&ldsA[(sc0 + j) * ROW + sm]) = pack;
+        }
+      }
+    }
+    // ---- load + transpose-stage B chunk ----
+    {
+      bf16x8 v = {};
+      if (MODE == MODE_CONV) {
+        if (gm < Mtot) {
+          int t = gm;
+          const int wo = t % cm.WO;
+          t /= cm.WO;
+          const int ho = t % cm.HO;
+          const int n = t / cm.HO;
+          const int hi = ho * cm.stride - cm.pad + r;
+          const int wi = wo * cm.stride - cm.pad + s;
+          const int c = j0 + sc0;
+          if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W &&
+              c < (1 << cm.C_log2))
+            v = *reinterpret_cast<const bf16x8*>(
+                B + (((long long)n * cm.H + hi) * cm.W + wi) *
+                        (1LL << cm.C_log2) +
+                c);
+        }
+      } else {
+        if (gm < Mtot && j0 + sc0 < J) {
+          const long long off = (long long)gm * J + j0 + sc0;
+          if (off + 8 <= (long long)Mtot * J) {
+            v = *reinterpret_cast<const bf16x8*>(B + off);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              if (off + j < (long long)Mtot * J) v[j] = B[off + j];
+          }
+        }
+      }
+      short8 mine = *reinterpret_cast<short8*>(&v);
+      short8 other;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) other[j] = __shfl_xor((int)mine[j], 8);
+      if (((threadIdx.x >> 3) & 1) == 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          unsigned int pack = ((unsigned short)mine[j]) |
+                              (((unsigned int)(unsigned short)other[j]) << 16);
+          *reinterpret_cast<unsigned int*>(&ldsB[(sc0 + j) * ROW + sm]) = pack;
+        }
+      }
+    }
+    __syncthreads();
+    // ---- MFMA: k dimension = m (BMC=32 -> one 16x16x32 per frag) ----
+    bf16x8 af[2], bfr[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      af[i] = *reinterpret_cast<const bf16x8*>(
+          &ldsA[(wm + i * 16 + fr) * ROW + fs * 8]);
+      bfr[i] = *reinterpret_cast<const bf16x8*>(
+          &ldsB[(wn + i * 16 + fr) * ROW + fs * 8]);
+    }
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+    __syncthreads();
+  }
+
+  // ---- accumulate into C (f32) ----
+#pragma unroll
+  for (int ni = 0; ni < 2; ++ni) {
+    const int col = j0 + wn + ni * 16 + fr;
+    if (col >= J) continue;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = i0 + wm + mi * 16 + fs * 4 + rr;
+        if (row >= I) continue;
+        atomicAdd(&C[coff + (long long)row * ldc + col], acc[mi][ni][rr]);
+      }
+    }
+  }
+}
+
+}  // namespace g16
+
+// ======================= host-side helpers ===============================
+
+namespace {
+
+torch::Tensor& zero_page(const torch::Device& dev) {
+  static std::unordered_map<int, torch::Tensor> cache;
+  int idx = dev.index();
+  auto it = cache.find(idx);
+  if (it == cache.end()) {
+    it = cache
+             .emplace(idx, torch::zeros(
+                               {64}, torch::dtype(torch::kBFloat16).device(dev)))
+             .first;
+  }
+  return it->second;
+}
+
+int log2_exact(int v) {
+  int l = 0;
+  while ((1 << l) < v) ++l;
+  return ((1 << l) == v) ? l : -1;
+}
+
+}  // namespace
+
+// C[b,M,N] = A[b,M,K] @ B[b,N,K]^T (+bias, +relu).  2-D inputs = batch 1.
+torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
+                          c10::optional<torch::Tensor> bias, bool relu) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  const bool batched = A.dim() == 3;
+  long long bsz = batched ? A.size(0) : 1;
+  int M = (int)A.size(batched ? 1 : 0), K = (int)A.size(batched ? 2 : 1);
+  int N = (int)B.size(batched ? 1 : 0);
+  TORCH_CHECK((int)B.size(batched ? 2 : 1) == K, "K mismatch");
+  TORCH_CHECK(K % 8 == 0, "K must be padded to a multiple of 8 (host)");
+  auto C = batched ? torch::empty({bsz, M, N}, A.options())
+                   : torch::empty({M, N}, A.options());
+  auto& zp = zero_page(A.device());
+  dim3 grid((N + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM,
+            (unsigned)bsz);
+  g16::ConvMeta cm{};
+  auto stream = c10::hip::getCurrentHIPStream();
+  const bf16* bias_p =
+      bias.has_value() ? reinterpret_cast<const bf16*>(bias->data_ptr())
+                       : nullptr;
+  long long sA = batched ? (long long)M * K : 0;
+  long long sB = batched ? (long long)N * K : 0;
+  long long sC = batched ? (long long)M * N : 0;
+#define LAUNCH_NT(RELU, HB)                                                   \
+  hipLaunchKernelGGL(                                                         \
+      (g16::gemm_nt_bf16_kernel<g16::MODE_PLAIN, RELU, HB>), grid,            \
+      dim3(g16::THREADS), 0, stream,                                          \
+      reinterpret_cast<const bf16*>(A.data_ptr()),                            \
+      reinterpret_cast<const bf16*>(B.data_ptr()),                            \
+      reinterpret_cast<bf16*>(C.data_ptr()), bias_p,                          \
+      reinterpret_cast<const bf16*>(zp.data_ptr()), M, N, K, sA, sB, sC, cm)
+  if (relu) {
+    if (bias_p) LAUNCH_NT(true, true);
+    else LAUNCH_NT(true, false);
+  } else {
+    if (bias_p) LAUNCH_NT(false, true);
+    else LAUNCH_NT(false, false);
+  }
+#undef LAUNCH_NT
+  return C;
+}
+
+// conv2d forward, NHWC x[N,H,W,C] * w[Kout,R,S,C] -> y[N,HO,WO,Kout]
+torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
+                              c10::optional<torch::Tensor> bias, int64_t stride,
+                              int64_t pad, bool relu) {
+  int N = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
+      Cin = (int)x.size(3);
+  int Kout = (int)w.size(0), R = (int)w.size(1), S = (int)w.size(2);
+  TORCH_CHECK((int)w.size(3) == Cin);
+  int cl = log2_exact(Cin);
+  TORCH_CHECK(cl >= 3, "conv fast path needs pow2 C >= 8 (stem uses im2col)");
+  int HO = (H + 2 * (int)pad - R) / (int)stride + 1;
+  int WO = (W + 2 * (int)pad - S) / (int)stride + 1;
+  int M = N * HO * WO, K = R * S * Cin;
+  auto y = torch::empty({N, HO, WO, Kout}, x.options());
+  auto& zp = zero_page(x.device());
+  dim3 grid((Kout + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
+  g16::ConvMeta cm{H, W, cl, S, R, (int)stride, (int)pad, HO, WO};
+  auto stream = c10::hip::getCurrentHIPStream();
+  const bf16* bias_p =
+      bias.has_value() ? reinterpret_cast<const bf16*>(bias->data_ptr())
+                       : nullptr;
+#define LAUNCH_CV(RELU, HB)                                                   \
+  hipLaunchKernelGGL(                                                         \
+      (g16::gemm_nt_bf16_kernel<g16::MODE_CONV, RELU, HB>), grid,             \
+      dim3(g16::THREADS), 0, stream,                                          \
+      reinterpret_cast<const bf16*>(x.data_ptr()),                            \
+      reinterpret_cast<const bf16*>(w.data_ptr()),                            \
+      reinterpret_cast<bf16*>(y.data_ptr()), bias_p,                          \
+      reinterpret_cast<const bf16*>(zp.data_ptr()), M, Kout, K, 0, 0, 0, cm)
+  if (relu) {
+    if (bias_p) LAUNCH_CV(true, true);
+    else LAUNCH_CV(true, false);
+  } else {
+    if (bias_p) LAUNCH_CV(false, true);
+    else LAUNCH_CV(false, false);
+  }
+#undef LAUNCH_CV
+  return y;
+}
+
+// C[I,J] = A[...,M,I]^T @ B[...,M,J] summed over batch? No — per batch.
+// Returns f32.  2-D inputs only here; batched variant loops z on grid.
+torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  const bool batched = A.dim() == 3;
+  TORCH_CHECK(!batched || B.dim() == 3);
+  long long bsz = batched ? A.size(0) : 1;
+  int M = (int)A.size(batched ? 1 : 0), I = (int)A.size(batched ? 2 : 1);
+  int J = (int)B.size(batched ? 2 : 1);
+  TORCH_CHECK((int)B.size(batched ? 1 : 0) == (batched ? M : M));
+  auto C = batched
+               ? torch::zeros({bsz, I, J},
+                              A.options().dtype(torch::kFloat32))
+               : torch::zeros({I, J}, A.options().dtype(torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream();
+  // split-M for parallelism: aim for >= 512 blocks
+  int n_chunks = (M + 31) / 32;
+  int tiles = ((J + 63) / 64) * ((I + 63) / 64);
+  int z = std::max(1, std::min(n_chunks, 512 / std::max(1, tiles) + 1));
+  g16::ConvMeta cm{};
+  for (long long b = 0; b < bsz; ++b) {
+    dim3 grid((J + 63) / 64, (I + 63) / 64, z);
+    hipLaunchKernelGGL(
+        (g16::gemm_tn_bf16_kernel<g16::MODE_PLAIN>), grid, dim3(g16::THREADS),
+        0, stream,
+        reinterpret_cast<const bf16*>(A.data_ptr()) + b * (long long)M * I,
+        reinterpret_cast<const bf16*>(B.data_ptr()) + b * (long long)M * J,
+        C.data_ptr<float>() + b * (long long)I * J, M, I, J, 0, 0,
+        /*ldc=*/J, /*coff=*/0, cm);
+  }
+  return C;
+}
+
+// conv wgrad: dw[Kout,R,S,C] (f32) from dy[N,HO,WO,Kout], x[N,H,W,C]
+torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
+                                int64_t stride, int64_t pad, int64_t R,
+                                int64_t S) {
+  int N = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
+      Cin = (int)x.size(3);
+  int HO = (int)dy.size(1), WO = (int)dy.size(2), Kout = (int)dy.size(3);
+  int cl = log2_exact(Cin);
+  TORCH_CHECK(cl >= 3, "conv wgrad fast path needs pow2 C (stem uses im2col)");
+  int M = N * HO * WO;
+  auto dw = torch::zeros({(long long)Kout, R, S, (long long)Cin},
+                         x.options().dtype(torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream();
+  g16::ConvMeta cm{H, W, cl, (int)S, (int)R, (int)stride, (int)pad, HO, WO};
+  int n_chunks = (M + 31) / 32;
+  int tiles = ((Cin + 63) / 64) * ((Kout + 63) / 64);
+  int z = std::max(1, std::min(n_chunks, 512 / std::max(1, tiles) + 1));
+  for (int r = 0; r < (int)R; ++r) {
+    for (int s = 0; s < (int)S; ++s) {
+      dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
+      // dw slice for (r,s): offset (r*S+s)*Cin within each Kout row; row
+      // stride of the f32 output is R*S*Cin.
+      hipLaunchKernelGGL(
+          (g16::gemm_tn_bf16_kernel<g16::MODE_CONV>), grid,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const bf16*>(dy.data_ptr()),
+          reinterpret_cast<const bf16*>(x.data_ptr()),
+          dw.data_ptr<float>(), M, Kout, Cin, r, s,
+          /*ldc=*/(long long)R * S * Cin,
+          /*coff=*/(long long)(r * (int)S + s) * Cin, cm);
+    }
+  }
+  return dw;
+}

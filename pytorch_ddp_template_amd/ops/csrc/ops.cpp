@@ -1,0 +1,261 @@
+// Binding + routing layer for the in-tree HIP extension (gfx950).
+//
+// Exposes the op surface ops/native.py expects.  Dtype routing picks the
+// bf16 MFMA kernels or the exact-f32 MFMA kernels; shape glue (K padding to
+// the GEMM kernels' chunk granularity, the im2col fallback for the
+// non-pow2-channel stem, weight rotation / zero-stuffing for strided conv
+// dgrad) lives here so the kernels stay branch-free on the hot path.
+
+#include <torch/extension.h>
+
+#include <vector>
+
+// ---- implemented in the .hip translation units ----
+// elementwise.hip
+torch::Tensor relu_fwd(torch::Tensor x);
+torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
+torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b);
+torch::Tensor gelu_fwd(torch::Tensor x);
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x);
+torch::Tensor col_sum(torch::Tensor dy);
+torch::Tensor transpose2d(torch::Tensor x);
+torch::Tensor avgpool_global(torch::Tensor x);
+torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W);
+std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k, int64_t s,
+                                         int64_t p);
+torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
+                            int64_t W);
+torch::Tensor weight_rot(torch::Tensor w);
+torch::Tensor zero_stuff(torch::Tensor dy, int64_t s);
+torch::Tensor im2col(torch::Tensor x, int64_t R, int64_t S, int64_t stride,
+                     int64_t pad, int64_t Kpad);
+// gemm_bf16.hip
+torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
+                          c10::optional<torch::Tensor> bias, bool relu);
+torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B);
+torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
+                              c10::optional<torch::Tensor> bias, int64_t stride,
+                              int64_t pad, bool relu);
+torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
+                                int64_t stride, int64_t pad, int64_t R,
+                                int64_t S);
+// gemm_f32.hip
+torch::Tensor bmm_nt_f32(torch::Tensor A, torch::Tensor B,
+                         c10::optional<torch::Tensor> bias, bool relu);
+torch::Tensor bmm_tn_f32(torch::Tensor A, torch::Tensor B);
+torch::Tensor conv2d_fwd_f32(torch::Tensor x, torch::Tensor w,
+                             c10::optional<torch::Tensor> bias, int64_t stride,
+                             int64_t pad, bool relu);
+torch::Tensor conv2d_wgrad_f32(torch::Tensor dy, torch::Tensor x,
+                               int64_t stride, int64_t pad, int64_t R,
+                               int64_t S);
+// norm.hip
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
+                                  torch::Tensor beta,
+                                  c10::optional<torch::Tensor> running_mean,
+                                  c10::optional<torch::Tensor> running_var,
+                                  double momentum, double eps, bool relu);
+torch::Tensor bn_infer(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
+                       torch::Tensor rmean, torch::Tensor rvar, double eps,
+                       bool relu);
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor gamma, torch::Tensor mean,
+                                  torch::Tensor rstd);
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
+                                         torch::Tensor beta, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor gamma,
+                                         torch::Tensor mean,
+                                         torch::Tensor rstd);
+torch::Tensor softmax_fwd(torch::Tensor x, double scale);
+torch::Tensor softmax_bwd(torch::Tensor dy, torch::Tensor y, double scale);
+// loss.hip
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor target);
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor target,
+                     torch::Tensor lse, torch::Tensor dloss);
+torch::Tensor mse_fwd(torch::Tensor p, torch::Tensor t);
+torch::Tensor mse_bwd(torch::Tensor p, torch::Tensor t, torch::Tensor dloss);
+// multi_tensor.hip
+void sgd_step(std::vector<torch::Tensor> params,
+              std::vector<torch::Tensor> grads,
+              std::vector<torch::Tensor> moms,
+              std::vector<torch::Tensor> masters, double lr, double momentum,
+              double wd, double damp, bool nesterov);
+torch::Tensor l2norm_sq(std::vector<torch::Tensor> grads);
+void scale_(std::vector<torch::Tensor> ts, double s);
+void scale_by_tensor_(std::vector<torch::Tensor> ts, torch::Tensor s);
+
+// ======================= routing helpers =================================
+
+namespace {
+
+bool is_bf16(const torch::Tensor& t) {
+  return t.scalar_type() == torch::kBFloat16;
+}
+
+int k_granule(const torch::Tensor& t) { return is_bf16(t) ? 8 : 4; }
+
+// pad the trailing (K) dim up to the kernel's chunk granularity
+torch::Tensor pad_k(const torch::Tensor& t, int mult) {
+  long long K = t.size(-1);
+  long long pad = (mult - K % mult) % mult;
+  if (pad == 0) return t;
+  return torch::constant_pad_nd(t, {0, pad});
+}
+
+bool pow2_ge(int v, int lo) {
+  return v >= lo && (v & (v - 1)) == 0;
+}
+
+}  // namespace
+
+// C = A @ B^T (+bias)(+relu); A [.., M, K], B [N, K]
+torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B,
+                      c10::optional<torch::Tensor> bias, bool relu,
+                      bool /*out_f32_unused*/) {
+  TORCH_CHECK(A.scalar_type() == B.scalar_type(), "dtype mismatch");
+  int g = k_granule(A);
+  if (A.size(-1) % g) {
+    A = pad_k(A.contiguous(), g);
+    B = pad_k(B.contiguous(), g);
+  }
+  if (is_bf16(A)) return bmm_nt_bf16(A.contiguous(), B.contiguous(), bias, relu);
+  return bmm_nt_f32(A.contiguous(), B.contiguous(), bias, relu);
+}
+
+torch::Tensor gemm_tn(torch::Tensor A, torch::Tensor B) {
+  if (is_bf16(A)) return bmm_tn_bf16(A.contiguous(), B.contiguous());
+  return bmm_tn_f32(A.contiguous(), B.contiguous());
+}
+
+torch::Tensor bmm_nt(torch::Tensor A, torch::Tensor B) {
+  int g = k_granule(A);
+  if (A.size(-1) % g) {
+    A = pad_k(A.contiguous(), g);
+    B = pad_k(B.contiguous(), g);
+  }
+  if (is_bf16(A)) return bmm_nt_bf16(A.contiguous(), B.contiguous(), {}, false);
+  return bmm_nt_f32(A.contiguous(), B.contiguous(), {}, false);
+}
+
+torch::Tensor bmm_tn(torch::Tensor A, torch::Tensor B) {
+  auto C = is_bf16(A) ? bmm_tn_bf16(A.contiguous(), B.contiguous())
+                      : bmm_tn_f32(A.contiguous(), B.contiguous());
+  return C.to(A.scalar_type());
+}
+
+// C[b,M,N] = A[b,M,K] @ B[b,K,N]: realized as NT against B^T (fast batched
+// LDS transpose, then the NT MFMA kernel).
+torch::Tensor bmm_nn(torch::Tensor A, torch::Tensor B) {
+  auto Bt = transpose2d(B.contiguous());  // [b, N, K]
+  return bmm_nt(A, Bt);
+}
+
+// ---- conv2d ----
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias, int64_t stride,
+                         int64_t pad, bool relu) {
+  int Cin = (int)x.size(3);
+  int R = (int)w.size(1), S = (int)w.size(2), Kout = (int)w.size(0);
+  const bool fast = is_bf16(x) ? pow2_ge(Cin, 8) : pow2_ge(Cin, 4);
+  if (fast) {
+    if (is_bf16(x))
+      return conv2d_fwd_bf16(x.contiguous(), w.contiguous(), bias, stride, pad,
+                             relu);
+    return conv2d_fwd_f32(x.contiguous(), w.contiguous(), bias, stride, pad,
+                          relu);
+  }
+  // generic fallback: explicit im2col + NT GEMM (used by the 3-channel stem)
+  int K = R * S * Cin;
+  int g = k_granule(x);
+  int Kpad = (K + g - 1) / g * g;
+  auto cols = im2col(x.contiguous(), R, S, stride, pad, Kpad);
+  auto wf = pad_k(w.contiguous().reshape({Kout, K}), g);
+  auto y = is_bf16(x) ? bmm_nt_bf16(cols, wf.contiguous(), bias, relu)
+                      : bmm_nt_f32(cols, wf.contiguous(), bias, relu);
+  int N = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2);
+  int HO = (H + 2 * (int)pad - R) / (int)stride + 1;
+  int WO = (W + 2 * (int)pad - S) / (int)stride + 1;
+  return y.reshape({N, HO, WO, Kout});
+}
+
+torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
+                           int64_t pad, int64_t H, int64_t W) {
+  // dx = conv(zero_stuffed(dy), rot(w), stride=1, pad=R-1-pad)
+  int R = (int)w.size(1), S = (int)w.size(2);
+  auto wr = weight_rot(w.contiguous());  // [C, R, S, Kout]
+  auto dys = stride > 1 ? zero_stuff(dy.contiguous(), stride) : dy.contiguous();
+  auto dx = conv2d_fwd(dys, wr, {}, 1, R - 1 - pad, false);
+  TORCH_CHECK(dx.size(1) == H && dx.size(2) == W,
+              "dgrad shape mismatch: got ", dx.sizes(), " want H=", H, " W=",
+              W, " (input H+2p-R must be divisible by stride)");
+  return dx;
+}
+
+torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
+                           int64_t pad, int64_t R, int64_t S) {
+  int Cin = (int)x.size(3);
+  const bool fast = is_bf16(x) ? pow2_ge(Cin, 8) : pow2_ge(Cin, 4);
+  dy = dy.contiguous();
+  x = x.contiguous();
+  if (fast) {
+    return is_bf16(x) ? conv2d_wgrad_bf16(dy, x, stride, pad, R, S)
+                      : conv2d_wgrad_f32(dy, x, stride, pad, R, S);
+  }
+  // generic fallback: im2col + TN GEMM
+  int K = (int)(R * S * Cin);
+  int g = k_granule(x);
+  int Kpad = (K + g - 1) / g * g;
+  auto cols = im2col(x, R, S, stride, pad, Kpad);  // [M, Kpad]
+  int Kout = (int)dy.size(3);
+  auto dw = is_bf16(x)
+                ? bmm_tn_bf16(dy.reshape({-1, Kout}).contiguous(), cols)
+                : bmm_tn_f32(dy.reshape({-1, Kout}).contiguous(), cols);
+  // [Kout, Kpad] f32 -> [Kout, R, S, Cin]
+  return dw.narrow(1, 0, K).reshape({Kout, R, S, Cin}).contiguous();
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm_nt", &gemm_nt, py::arg("A"), py::arg("B"),
+        py::arg("bias") = c10::nullopt, py::arg("relu") = false,
+        py::arg("out_f32") = false);
+  m.def("gemm_tn", &gemm_tn);
+  m.def("bmm_nt", &bmm_nt);
+  m.def("bmm_nn", &bmm_nn);
+  m.def("bmm_tn", &bmm_tn);
+  m.def("transpose2d", &transpose2d);
+  m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("w"),
+        py::arg("bias") = c10::nullopt, py::arg("stride") = 1,
+        py::arg("pad") = 0, py::arg("relu") = false);
+  m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_wgrad", &conv2d_wgrad);
+  m.def("relu_fwd", &relu_fwd);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("col_sum", &col_sum);
+  m.def("avgpool_global", &avgpool_global);
+  m.def("avgpool_global_bwd", &avgpool_global_bwd);
+  m.def("maxpool2d_fwd", &maxpool2d_fwd);
+  m.def("maxpool2d_bwd", &maxpool2d_bwd);
+  m.def("bn_fwd", &bn_fwd);
+  m.def("bn_infer", &bn_infer);
+  m.def("bn_bwd", &bn_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("softmax_fwd", &softmax_fwd);
+  m.def("softmax_bwd", &softmax_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("mse_fwd", &mse_fwd);
+  m.def("mse_bwd", &mse_bwd);
+  m.def("sgd_step", &sgd_step);
+  m.def("l2norm_sq", &l2norm_sq);
+  m.def("scale_", &scale_);
+  m.def("scale_by_tensor_", &scale_by_tensor_);
+  m.def("im2col", &im2col);
+  m.def("weight_rot", &weight_rot);
+  m.def("zero_stuff", &zero_stuff);
+}

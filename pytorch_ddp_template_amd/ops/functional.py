@@ -438,11 +438,18 @@ class _MaxPool2dFn(torch.autograd.Function):
         if use_native(dy):
             dx = native().maxpool2d_bwd(dy.contiguous(), idx, shape[1], shape[2])
         else:
+            # scatter-add (max_unpool2d overwrites ties instead of summing)
             N, H, W, C = shape
-            dyc = dy.permute(0, 3, 1, 2).float()
-            idxc = idx.permute(0, 3, 1, 2)
-            dxc = F.max_unpool2d(dyc, idxc, k, stride, pad, output_size=(H, W))
-            dx = dxc.permute(0, 2, 3, 1).contiguous().to(dy.dtype)
+            dyc = dy.permute(0, 3, 1, 2).reshape(N, C, -1).float()
+            idxc = idx.permute(0, 3, 1, 2).reshape(N, C, -1)
+            dxc = torch.zeros(N, C, H * W, dtype=torch.float32)
+            dxc.scatter_add_(2, idxc, dyc)
+            dx = (
+                dxc.reshape(N, C, H, W)
+                .permute(0, 2, 3, 1)
+                .contiguous()
+                .to(dy.dtype)
+            )
         return dx, None, None, None
 
 
