@@ -238,7 +238,10 @@ __global__ void weight_rot_kernel(const T* __restrict__ w, T* __restrict__ wr,
   }
 }
 
-// zero-stuff dy for strided dgrad: [N,HO,WO,K] -> [N,(HO-1)*s+1,(WO-1)*s+1,K]
+// zero-stuff dy for strided dgrad:
+// [N,HO,WO,K] -> [N,(HO-1)*s+1+opad_h,(WO-1)*s+1+opad_w,K]
+// (opad covers inputs the strided forward never reached when
+//  (H+2p-R) % stride != 0 — transposed-conv output_padding semantics)
 template <typename T>
 __global__ void zero_stuff_kernel(const T* __restrict__ dy, T* __restrict__ out,
                                   int N, int HO, int WO, int K, int S, int HS,
@@ -252,7 +255,7 @@ __global__ void zero_stuff_kernel(const T* __restrict__ dy, T* __restrict__ out,
     int hs = (int)(t % HS);
     int n = (int)(t / HS);
     T v = to_t<T>(0.f);
-    if (hs % S == 0 && ws % S == 0)
+    if (hs % S == 0 && ws % S == 0 && hs / S < HO && ws / S < WO)
       v = dy[(((long long)n * HO + hs / S) * WO + ws / S) * K + k];
     out[i] = v;
   }
@@ -522,11 +525,13 @@ torch::Tensor im2col(torch::Tensor x, int64_t R, int64_t S, int64_t stride,
   return cols;
 }
 
-torch::Tensor zero_stuff(torch::Tensor dy, int64_t s) {
+torch::Tensor zero_stuff(torch::Tensor dy, int64_t s, int64_t opad_h,
+                         int64_t opad_w) {
   CHECK_GPU(dy);
   int N = (int)dy.size(0), HO = (int)dy.size(1), WO = (int)dy.size(2),
       K = (int)dy.size(3);
-  int HS = (HO - 1) * (int)s + 1, WS = (WO - 1) * (int)s + 1;
+  int HS = (HO - 1) * (int)s + 1 + (int)opad_h;
+  int WS = (WO - 1) * (int)s + 1 + (int)opad_w;
   auto out = torch::empty({N, HS, WS, K}, dy.options());
   long long total = (long long)N * HS * WS * K;
   DDP_DISPATCH_FLOAT(dy.scalar_type(), "zero_stuff", [&] {

@@ -56,7 +56,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     bf16* __restrict__ C, const bf16* __restrict__ bias,
     const bf16* __restrict__ zpad, int M, int N, int K, long long strideA,
     long long strideB, long long strideC, ConvMeta cm) {
-  __shared__ char smem[2 * 2 * TILE_BYTES];  // [buf][A|B]
+  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * TILE_BYTES];  // [buf][A|B]
 
   // ----- block swizzle (bijective XCD remap over the x*y grid) -----
   int nwg = gridDim.x * gridDim.y;
@@ -221,8 +221,11 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
   // C element (i, j) lives at C[coff + i*ldc + j] — conv wgrad writes a
   // (r,s) slice of dw[Kout][R*S*C], so ldc != J there.
   constexpr int BI = 64, BJ = 64, BMC = 32;  // m-chunk
-  constexpr int ROW = BMC + 4;               // padded LDS row (elements)
-  __shared__ bf16 lds[2 * BI * ROW];         // A image then B image
+  // padded LDS row: +8 keeps 16-B alignment for the b128 fragment reads
+  // (row stride 80 B) while staying bank-conflict-free (banks r*20 mod 64
+  // are distinct over any 16 consecutive rows).
+  constexpr int ROW = BMC + 8;
+  __shared__ __attribute__((aligned(16))) bf16 lds[2 * BI * ROW];
   bf16* ldsA = lds;
   bf16* ldsB = lds + BI * ROW;
 

@@ -33,7 +33,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_f32_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     float* __restrict__ C, const float* __restrict__ bias, int M, int N, int K,
     long long strideA, long long strideB, long long strideC, ConvMeta cm) {
-  __shared__ float lds[2 * BM * ROW];  // A image then B image
+  __shared__ __attribute__((aligned(16))) float lds[2 * BM * ROW];  // A image then B image
   float* ldsA = lds;
   float* ldsB = lds + BM * ROW;
 
@@ -143,7 +143,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_f32_kernel(
     long long coff, ConvMeta cm) {
   constexpr int BI = 64, BJ = 64, BMC = 32;
   constexpr int TROW = BMC + 4;
-  __shared__ float lds[2 * BI * TROW];
+  __shared__ __attribute__((aligned(16))) float lds[2 * BI * TROW];
   float* ldsA = lds;
   float* ldsB = lds + BI * TROW;
 

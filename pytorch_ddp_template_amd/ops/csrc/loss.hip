@@ -19,7 +19,7 @@ namespace loss {
 // one block per row: lse + per-row loss contribution (atomic mean)
 template <typename T>
 __global__ void ce_fwd_kernel(const T* __restrict__ logits,
-                              const long long* __restrict__ target,
+                              const int64_t* __restrict__ target,
                               float* __restrict__ loss_out,
                               float* __restrict__ lse_out, int D) {
   __shared__ float lds[4];
@@ -48,7 +48,7 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
 // dlogits = (softmax - onehot) * (dloss / M); dloss read on device
 template <typename T>
 __global__ void ce_bwd_kernel(const T* __restrict__ logits,
-                              const long long* __restrict__ target,
+                              const int64_t* __restrict__ target,
                               const float* __restrict__ lse,
                               const float* __restrict__ dloss,
                               T* __restrict__ dlogits, long long M, int D) {
@@ -104,7 +104,7 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor target) {
     hipLaunchKernelGGL((loss::ce_fwd_kernel<scalar_t>), dim3((unsigned)M),
                        dim3(256), 0, stream,
                        reinterpret_cast<const scalar_t*>(logits.data_ptr()),
-                       target.data_ptr<long long>(), out.data_ptr<float>(),
+                       target.data_ptr<int64_t>(), out.data_ptr<float>(),
                        lse.data_ptr<float>(), D);
   });
   return {out, lse};
@@ -121,7 +121,7 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor target,
     hipLaunchKernelGGL((loss::ce_bwd_kernel<scalar_t>),
                        dim3(grid_1d(M * D, 256)), dim3(256), 0, stream,
                        reinterpret_cast<const scalar_t*>(logits.data_ptr()),
-                       target.data_ptr<long long>(), lse.data_ptr<float>(),
+                       target.data_ptr<int64_t>(), lse.data_ptr<float>(),
                        dloss_f.data_ptr<float>(),
                        reinterpret_cast<scalar_t*>(dl.data_ptr()), M, D);
   });

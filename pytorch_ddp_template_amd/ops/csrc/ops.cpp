@@ -26,7 +26,8 @@ std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k, int64_t s,
 torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                             int64_t W);
 torch::Tensor weight_rot(torch::Tensor w);
-torch::Tensor zero_stuff(torch::Tensor dy, int64_t s);
+torch::Tensor zero_stuff(torch::Tensor dy, int64_t s, int64_t opad_h,
+                         int64_t opad_w);
 torch::Tensor im2col(torch::Tensor x, int64_t R, int64_t S, int64_t stride,
                      int64_t pad, int64_t Kpad);
 // gemm_bf16.hip
@@ -182,10 +183,15 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
 
 torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
                            int64_t pad, int64_t H, int64_t W) {
-  // dx = conv(zero_stuffed(dy), rot(w), stride=1, pad=R-1-pad)
+  // dx = conv(zero_stuffed(dy), rot(w), stride=1, pad=R-1-pad); output
+  // padding covers (H+2p-R) % stride != 0 (inputs the fwd never reached).
   int R = (int)w.size(1), S = (int)w.size(2);
+  TORCH_CHECK(pad <= R - 1 && pad <= S - 1, "dgrad needs pad <= kernel-1");
+  int64_t opad_h = (H + 2 * pad - R) % stride;
+  int64_t opad_w = (W + 2 * pad - S) % stride;
   auto wr = weight_rot(w.contiguous());  // [C, R, S, Kout]
-  auto dys = stride > 1 ? zero_stuff(dy.contiguous(), stride) : dy.contiguous();
+  auto dys = stride > 1 ? zero_stuff(dy.contiguous(), stride, opad_h, opad_w)
+                        : dy.contiguous();
   auto dx = conv2d_fwd(dys, wr, {}, 1, R - 1 - pad, false);
   TORCH_CHECK(dx.size(1) == H && dx.size(2) == W,
               "dgrad shape mismatch: got ", dx.sizes(), " want H=", H, " W=",

@@ -1,0 +1,139 @@
+"""End-to-end model paths on MI355X: HIP kernels vs fp32 CPU references,
+loss descent, and the native extension actually being used."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def test_native_extension_loaded():
+    from pytorch_ddp_template_amd.ops import native_available
+
+    assert native_available(), "HIP extension must load on the GPU box"
+
+
+def test_foo_model_matches_cpu_f32():
+    from pytorch_ddp_template_amd.models import FooModel
+
+    torch.manual_seed(0)
+    m = FooModel()
+    mg = FooModel()
+    mg.load_state_dict(m.state_dict())
+    mg = mg.to(DEV)
+    x = torch.randn(64, 10)
+    y = m(x)
+    yg = mg(x.to(DEV))
+    torch.testing.assert_close(yg.cpu(), y, rtol=1e-3, atol=1e-4)
+    # grads
+    y.pow(2).sum().backward()
+    yg.pow(2).sum().backward()
+    for p, pg in zip(m.parameters(), mg.parameters()):
+        torch.testing.assert_close(pg.grad.cpu(), p.grad, rtol=2e-2, atol=2e-3)
+
+
+def test_resnet18_bf16_forward_close_to_cpu_f32():
+    from pytorch_ddp_template_amd.models import resnet18
+
+    torch.manual_seed(1)
+    m = resnet18(num_classes=10, stem="cifar")
+    m.eval()
+    x = torch.randn(4, 32, 32, 3)
+    with torch.no_grad():
+        ref = m(x)
+        mg = resnet18(num_classes=10, stem="cifar")
+        mg.load_state_dict(m.state_dict())
+        mg.eval()
+        out = mg.to(torch.bfloat16).to(DEV)(x.to(torch.bfloat16).to(DEV))
+    err = (out.float().cpu() - ref).abs().max()
+    scale = ref.abs().max().clamp(min=1.0)
+    assert err / scale < 0.12, f"relative error {err/scale}"
+
+
+def test_resnet18_bf16_training_loss_descends():
+    from pytorch_ddp_template_amd.models import resnet18
+    from pytorch_ddp_template_amd.ops import CrossEntropyLoss
+    from pytorch_ddp_template_amd.optim import SGD
+
+    torch.manual_seed(2)
+    m = resnet18(num_classes=10, stem="cifar").to(torch.bfloat16).to(DEV)
+    opt = SGD(m.parameters(), lr=0.05, momentum=0.9, master_weights=True)
+    crit = CrossEntropyLoss()
+    x = torch.randn(64, 32, 32, 3).to(torch.bfloat16).to(DEV)
+    y = torch.randint(0, 10, (64,)).to(DEV)
+    losses = []
+    for _ in range(20):
+        out = m(x)
+        loss = crit(out, y)
+        loss.backward()
+        opt.step()
+        m.zero_grad()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.7, losses[:3] + losses[-3:]
+    assert all(l == l for l in losses), "NaN loss"
+
+
+def test_foo_training_with_mse_descends():
+    from pytorch_ddp_template_amd.models import FooModel
+    from pytorch_ddp_template_amd.ops import MSELoss
+    from pytorch_ddp_template_amd.optim import SGD, clip_grad_norm_
+
+    torch.manual_seed(3)
+    m = FooModel().to(DEV)
+    opt = SGD(m.parameters(), lr=0.05)
+    crit = MSELoss()
+    x = torch.randn(256, 10).to(DEV)
+    y = torch.randn(256, 5).to(DEV)
+    losses = []
+    for _ in range(50):
+        loss = crit(m(x), y)
+        loss.backward()
+        clip_grad_norm_(list(m.parameters()), 1000.0)
+        opt.step()
+        m.zero_grad()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.9
+
+
+def test_vit_tiny_bf16_fwd_bwd_runs():
+    from pytorch_ddp_template_amd.models.vit import ViT
+
+    torch.manual_seed(4)
+    m = (
+        ViT(image_size=32, patch_size=8, dim=64, depth=2, heads=4, num_classes=10)
+        .to(torch.bfloat16)
+        .to(DEV)
+    )
+    x = torch.randn(4, 32, 32, 3).to(torch.bfloat16).to(DEV)
+    y = m(x)
+    assert y.shape == (4, 10)
+    y.float().sum().backward()
+    assert m.blocks[0].attn.qkv.weight.grad is not None
+    assert torch.isfinite(y.float()).all()
+
+
+def test_resnet50_bf16_fwd_bwd_runs():
+    from pytorch_ddp_template_amd.models import resnet50
+
+    m = resnet50(num_classes=1000).to(torch.bfloat16).to(DEV)
+    x = torch.randn(2, 224, 224, 3).to(torch.bfloat16).to(DEV)
+    y = m(x)
+    assert y.shape == (2, 1000)
+    y.float().sum().backward()
+    assert torch.isfinite(y.float()).all()
+
+
+def test_checkpoint_roundtrip_gpu(tmp_path):
+    from pytorch_ddp_template_amd.ddp import save_model
+    from pytorch_ddp_template_amd.models import resnet18
+
+    m = resnet18().to(torch.bfloat16).to(DEV)
+    save_model(m, str(tmp_path / "ck"))
+    m2 = resnet18().to(torch.bfloat16).to(DEV)
+    sd = torch.load(tmp_path / "ck" / "model.bin", map_location="cpu",
+                    weights_only=True)
+    m2.load_state_dict(sd)
+    for a, b in zip(m.parameters(), m2.parameters()):
+        assert torch.equal(a.cpu(), b.cpu())

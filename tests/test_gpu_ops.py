@@ -99,7 +99,7 @@ def test_bmm_nt_nn_batched():
     b = t32(6, 29, 16, seed=13).to(torch.bfloat16)
     out = EXT.bmm_nt(a.to(DEV), b.to(DEV))
     close_bf16(out, torch.einsum("bmk,bnk->bmn", a.float(), b.float()), scale=4.0)
-    c = t32(6, 16, 29, seed=14).to(torch.bfloat16)
+    c = t32(6, 29, 16, seed=14).to(torch.bfloat16)  # [b, K=29, N=16]
     out2 = EXT.bmm_nn(out.to(DEV), c.to(DEV))
     ref2 = torch.einsum(
         "bmn,bnj->bmj",
