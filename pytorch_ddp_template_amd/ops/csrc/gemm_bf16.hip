@@ -489,7 +489,7 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   // split-M for parallelism: aim for >= 512 blocks
   int n_chunks = (M + 31) / 32;
   int tiles = ((J + 63) / 64) * ((I + 63) / 64);
-  int z = std::max(1, std::min(n_chunks, 512 / std::max(1, tiles) + 1));
+  int z = std::max(1, std::min(n_chunks, 2048 / std::max(1, tiles) + 1));
   g16::ConvMeta cm{};
   for (long long b = 0; b < bsz; ++b) {
     dim3 grid((J + 63) / 64, (I + 63) / 64, z);
@@ -520,7 +520,7 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   g16::ConvMeta cm{H, W, cl, (int)S, (int)R, (int)stride, (int)pad, HO, WO};
   int n_chunks = (M + 31) / 32;
   int tiles = ((Cin + 63) / 64) * ((Kout + 63) / 64);
-  int z = std::max(1, std::min(n_chunks, 512 / std::max(1, tiles) + 1));
+  int z = std::max(1, std::min(n_chunks, 2048 / std::max(1, tiles) + 1));
   for (int r = 0; r < (int)R; ++r) {
     for (int s = 0; s < (int)S; ++s) {
       dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);

@@ -101,6 +101,15 @@ class BatchNorm2dNHWC(nn.Module):
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
 
+    def _apply(self, fn, recurse=True):
+        # Running stats stay fp32 regardless of model dtype: the BN kernels
+        # accumulate and update them in fp32 (model.to(bf16) must not
+        # downcast them).
+        super()._apply(fn, recurse)
+        self.running_mean.data = self.running_mean.data.float()
+        self.running_var.data = self.running_var.data.float()
+        return self
+
     def forward(self, x):
         if self.training:
             self.num_batches_tracked += 1
