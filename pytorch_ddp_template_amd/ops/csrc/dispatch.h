@@ -6,7 +6,7 @@
 
 #include "common.h"
 
-// Map torch scalar types onto the kernel types (float / bf16).
+// Map torch scalar types onto the kernel types (float / bf16 / f16).
 #define DDP_DISPATCH_FLOAT(TYPE, NAME, ...)                         \
   [&] {                                                             \
     switch (TYPE) {                                                 \
@@ -16,6 +16,10 @@
       }                                                             \
       case torch::kBFloat16: {                                      \
         using scalar_t = bf16;                                      \
+        return __VA_ARGS__();                                       \
+      }                                                             \
+      case torch::kHalf: {                                          \
+        using scalar_t = _Float16;                                  \
         return __VA_ARGS__();                                       \
       }                                                             \
       default:                                                      \
@@ -34,6 +38,10 @@ template <>
 DEV_INLINE float to_f<bf16>(bf16 v) {
   return __bfloat162float(v);
 }
+template <>
+DEV_INLINE float to_f<_Float16>(_Float16 v) {
+  return (float)v;
+}
 
 template <typename T>
 DEV_INLINE T to_t(float v);
@@ -44,6 +52,10 @@ DEV_INLINE float to_t<float>(float v) {
 template <>
 DEV_INLINE bf16 to_t<bf16>(float v) {
   return __float2bfloat16(v);
+}
+template <>
+DEV_INLINE _Float16 to_t<_Float16>(float v) {
+  return (_Float16)v;
 }
 
 // 16-byte vectorized access helpers
@@ -64,4 +76,12 @@ struct VecIO<bf16> {
   using Vec = short8;
   DEV_INLINE static float get(const Vec& v, int i) { return bfbits2f(v[i]); }
   DEV_INLINE static void set(Vec& v, int i, float x) { v[i] = f2bfbits(x); }
+};
+
+template <>
+struct VecIO<_Float16> {
+  static constexpr int kPerLane = 8;  // 16 B
+  using Vec = half8;
+  DEV_INLINE static float get(const Vec& v, int i) { return (float)v[i]; }
+  DEV_INLINE static void set(Vec& v, int i, float x) { v[i] = (_Float16)x; }
 };

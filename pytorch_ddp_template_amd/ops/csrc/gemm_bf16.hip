@@ -33,7 +33,26 @@
 namespace g16 {
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+// 16-bit dtype traits: fragment vector type + the matching MFMA intrinsic.
+template <typename T16>
+struct M16;
+template <>
+struct M16<bf16> {
+  using vec = bf16x8;
+  DEV_INLINE static f32x4 mma(vec a, vec b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <>
+struct M16<_Float16> {
+  using vec = f16x8;
+  DEV_INLINE static f32x4 mma(vec a, vec b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+};
 
 constexpr int BM = 128, BN = 128, BK = 32;
 constexpr int THREADS = 256;
@@ -50,11 +69,11 @@ struct ConvMeta {
 enum { MODE_PLAIN = 0, MODE_CONV = 1 };
 
 // ---------------------------------------------------------------- NT -----
-template <int MODE, bool RELU, bool HAS_BIAS>
+template <typename T16, int MODE, bool RELU, bool HAS_BIAS>
 __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
-    const bf16* __restrict__ A, const bf16* __restrict__ B,
-    bf16* __restrict__ C, const bf16* __restrict__ bias,
-    const bf16* __restrict__ zpad, int M, int N, int K, long long strideA,
+    const T16* __restrict__ A, const T16* __restrict__ B,
+    T16* __restrict__ C, const T16* __restrict__ bias,
+    const T16* __restrict__ zpad, int M, int N, int K, long long strideA,
     long long strideB, long long strideC, ConvMeta cm) {
   __shared__ __attribute__((aligned(16))) char smem[2 * 2 * TILE_BYTES];  // [buf][A|B]
 
@@ -117,7 +136,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
       // ---- A tile ----
       const int swz = kswz(rl_a[i], kp);
       const int gk = k_base + swz * 8;  // first element of this 16B chunk
-      const bf16* src = zpad;
+      const T16* src = zpad;
       if (MODE == MODE_CONV) {
         if (a_ok[i] && gk < K) {
           const int c0 = gk & ((1 << cm.C_log2) - 1);
@@ -141,7 +160,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
       const int gn = n0 + rlb;
       const int swzb = kswz(rlb, kp);
       const int gkb = k_base + swzb * 8;
-      const bf16* srcb =
+      const T16* srcb =
           (gn < N && gkb < K) ? B + (long long)gn * K + gkb : zpad;
       char* ldsB =
           &smem[buf * 2 * TILE_BYTES + TILE_BYTES + (wave * 2 + i) * 1024];
@@ -164,22 +183,22 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     if (kt + 1 < KT) stage(buf ^ 1, kt + 1);
     const char* baseA = &smem[buf * 2 * TILE_BYTES];
     const char* baseB = baseA + TILE_BYTES;
-    bf16x8 af[4], bf[4];
+    using vec16 = typename M16<T16>::vec;
+    vec16 af[4], bf[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int ra = wm + i * 16 + fr;
-      af[i] = *reinterpret_cast<const bf16x8*>(
+      af[i] = *reinterpret_cast<const vec16*>(
           baseA + ra * 64 + kswz(ra, fs) * 16);
       const int rb = wn + i * 16 + fr;
-      bf[i] = *reinterpret_cast<const bf16x8*>(
+      bf[i] = *reinterpret_cast<const vec16*>(
           baseB + rb * 64 + kswz(rb, fs) * 16);
     }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+        acc[mi][ni] = M16<T16>::mma(af[mi], bf[ni], acc[mi][ni]);
     __syncthreads();
     buf ^= 1;
   }
@@ -189,7 +208,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   for (int ni = 0; ni < 4; ++ni) {
     const int col = n0 + wn + ni * 16 + fr;
     if (col >= N) continue;
-    float bv = HAS_BIAS ? __bfloat162float(bias[col]) : 0.f;
+    float bv = HAS_BIAS ? to_f(bias[col]) : 0.f;
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const int row_base = m0 + wm + mi * 16 + fs * 4;
@@ -199,7 +218,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
         if (row >= M) continue;
         float v = acc[mi][ni][r] + bv;
         if (RELU) v = fmaxf(v, 0.f);
-        C[(long long)row * N + col] = __float2bfloat16(v);
+        C[(long long)row * N + col] = to_t<T16>(v);
       }
     }
   }
@@ -213,9 +232,9 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
 //
 // MODE_CONV gathers the B operand (im2col of x) on the fly for conv wgrad;
 // the j-tile then lives inside one (r,s) slice: j = c0_tile + c.
-template <int MODE>
+template <typename T16, int MODE>
 __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
-    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    const T16* __restrict__ A, const T16* __restrict__ B,
     float* __restrict__ C, int Mtot, int I, int J, int r, int s,
     long long ldc, long long coff, ConvMeta cm) {
   // C element (i, j) lives at C[coff + i*ldc + j] — conv wgrad writes a
@@ -225,9 +244,10 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
   // (row stride 80 B) while staying bank-conflict-free (banks r*20 mod 64
   // are distinct over any 16 consecutive rows).
   constexpr int ROW = BMC + 8;
-  __shared__ __attribute__((aligned(16))) bf16 lds[2 * BI * ROW];
-  bf16* ldsA = lds;
-  bf16* ldsB = lds + BI * ROW;
+  using vec16 = typename M16<T16>::vec;
+  __shared__ __attribute__((aligned(16))) T16 lds[2 * BI * ROW];
+  T16* ldsA = lds;
+  T16* ldsB = lds + BI * ROW;
 
   const int i0 = blockIdx.y * BI;
   const int j0 = blockIdx.x * BJ;
@@ -256,11 +276,11 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
     const int gm = mbase + sm;
     // ---- load + transpose-stage A chunk ([m][I] -> image [i][m]) ----
     {
-      bf16x8 v = {};
+      vec16 v = {};
       if (gm < Mtot && i0 + sc0 < I) {
         const long long off = (long long)gm * I + i0 + sc0;
         if (off + 8 <= (long long)Mtot * I) {
-          v = *reinterpret_cast<const bf16x8*>(A + off);
+          v = *reinterpret_cast<const vec16*>(A + off);
         } else {  // last-row partial chunk: element-wise guarded load
 #pragma unroll
           for (int j = 0; j < 8; ++j)
@@ -283,7 +303,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
     }
     // ---- load + transpose-stage B chunk ----
     {
-      bf16x8 v = {};
+      vec16 v = {};
       if (MODE == MODE_CONV) {
         if (gm < Mtot) {
           int t = gm;
@@ -296,7 +316,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
           const int c = j0 + sc0;
           if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W &&
               c < (1 << cm.C_log2))
-            v = *reinterpret_cast<const bf16x8*>(
+            v = *reinterpret_cast<const vec16*>(
                 B + (((long long)n * cm.H + hi) * cm.W + wi) *
                         (1LL << cm.C_log2) +
                 c);
@@ -305,7 +325,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
         if (gm < Mtot && j0 + sc0 < J) {
           const long long off = (long long)gm * J + j0 + sc0;
           if (off + 8 <= (long long)Mtot * J) {
-            v = *reinterpret_cast<const bf16x8*>(B + off);
+            v = *reinterpret_cast<const vec16*>(B + off);
           } else {
 #pragma unroll
             for (int j = 0; j < 8; ++j)
@@ -328,20 +348,19 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
     }
     __syncthreads();
     // ---- MFMA: k dimension = m (BMC=32 -> one 16x16x32 per frag) ----
-    bf16x8 af[2], bfr[2];
+    vec16 af[2], bfr[2];
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      af[i] = *reinterpret_cast<const bf16x8*>(
+      af[i] = *reinterpret_cast<const vec16*>(
           &ldsA[(wm + i * 16 + fr) * ROW + fs * 8]);
-      bfr[i] = *reinterpret_cast<const bf16x8*>(
+      bfr[i] = *reinterpret_cast<const vec16*>(
           &ldsB[(wn + i * 16 + fr) * ROW + fs * 8]);
     }
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        acc[mi][ni] = M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni]);
     __syncthreads();
   }
 
