@@ -387,17 +387,43 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 
 namespace {
 
-torch::Tensor& zero_page(const torch::Device& dev) {
+torch::Tensor& zero_page(const torch::Device& dev, torch::ScalarType st) {
   static std::unordered_map<int, torch::Tensor> cache;
-  int idx = dev.index();
+  int idx = dev.index() * 4 + (st == torch::kBFloat16 ? 0 : 1);
   auto it = cache.find(idx);
   if (it == cache.end()) {
-    it = cache
-             .emplace(idx, torch::zeros(
-                               {64}, torch::dtype(torch::kBFloat16).device(dev)))
+    it = cache.emplace(idx, torch::zeros({64}, torch::dtype(st).device(dev)))
              .first;
   }
   return it->second;
+}
+
+template <typename t16, int MODE>
+void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
+                 torch::Tensor& C, const c10::optional<torch::Tensor>& bias,
+                 bool relu, const torch::Tensor& zp, dim3 grid, int M, int N,
+                 int K, long long sA, long long sB, long long sC,
+                 g16::ConvMeta cm) {
+  auto stream = c10::hip::getCurrentHIPStream();
+  const t16* bias_p =
+      bias.has_value() ? reinterpret_cast<const t16*>(bias->data_ptr())
+                       : nullptr;
+#define LAUNCH_NT16(RELU, HB)                                                \
+  hipLaunchKernelGGL((g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB>), grid,  \
+                     dim3(g16::THREADS), 0, stream,                          \
+                     reinterpret_cast<const t16*>(A.data_ptr()),             \
+                     reinterpret_cast<const t16*>(B.data_ptr()),             \
+                     reinterpret_cast<t16*>(C.data_ptr()), bias_p,           \
+                     reinterpret_cast<const t16*>(zp.data_ptr()), M, N, K,   \
+                     sA, sB, sC, cm)
+  if (relu) {
+    if (bias_p) LAUNCH_NT16(true, true);
+    else LAUNCH_NT16(true, false);
+  } else {
+    if (bias_p) LAUNCH_NT16(false, true);
+    else LAUNCH_NT16(false, false);
+  }
+#undef LAUNCH_NT16
 }
 
 int log2_exact(int v) {
@@ -420,33 +446,20 @@ torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
   TORCH_CHECK(K % 8 == 0, "K must be padded to a multiple of 8 (host)");
   auto C = batched ? torch::empty({bsz, M, N}, A.options())
                    : torch::empty({M, N}, A.options());
-  auto& zp = zero_page(A.device());
+  auto& zp = zero_page(A.device(), A.scalar_type());
   dim3 grid((N + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM,
             (unsigned)bsz);
   g16::ConvMeta cm{};
   auto stream = c10::hip::getCurrentHIPStream();
-  const bf16* bias_p =
-      bias.has_value() ? reinterpret_cast<const bf16*>(bias->data_ptr())
-                       : nullptr;
   long long sA = batched ? (long long)M * K : 0;
   long long sB = batched ? (long long)N * K : 0;
   long long sC = batched ? (long long)M * N : 0;
-#define LAUNCH_NT(RELU, HB)                                                   \
-  hipLaunchKernelGGL(                                                         \
-      (g16::gemm_nt_bf16_kernel<g16::MODE_PLAIN, RELU, HB>), grid,            \
-      dim3(g16::THREADS), 0, stream,                                          \
-      reinterpret_cast<const bf16*>(A.data_ptr()),                            \
-      reinterpret_cast<const bf16*>(B.data_ptr()),                            \
-      reinterpret_cast<bf16*>(C.data_ptr()), bias_p,                          \
-      reinterpret_cast<const bf16*>(zp.data_ptr()), M, N, K, sA, sB, sC, cm)
-  if (relu) {
-    if (bias_p) LAUNCH_NT(true, true);
-    else LAUNCH_NT(true, false);
-  } else {
-    if (bias_p) LAUNCH_NT(false, true);
-    else LAUNCH_NT(false, false);
-  }
-#undef LAUNCH_NT
+  if (A.scalar_type() == torch::kBFloat16)
+    launch_nt16<bf16, g16::MODE_PLAIN>(A, B, C, bias, relu, zp, grid, M, N, K,
+                                       sA, sB, sC, cm);
+  else
+    launch_nt16<_Float16, g16::MODE_PLAIN>(A, B, C, bias, relu, zp, grid, M,
+                                           N, K, sA, sB, sC, cm);
   return C;
 }
 
@@ -464,29 +477,15 @@ torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
   int WO = (W + 2 * (int)pad - S) / (int)stride + 1;
   int M = N * HO * WO, K = R * S * Cin;
   auto y = torch::empty({N, HO, WO, Kout}, x.options());
-  auto& zp = zero_page(x.device());
+  auto& zp = zero_page(x.device(), x.scalar_type());
   dim3 grid((Kout + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
   g16::ConvMeta cm{H, W, cl, S, R, (int)stride, (int)pad, HO, WO};
-  auto stream = c10::hip::getCurrentHIPStream();
-  const bf16* bias_p =
-      bias.has_value() ? reinterpret_cast<const bf16*>(bias->data_ptr())
-                       : nullptr;
-#define LAUNCH_CV(RELU, HB)                                                   \
-  hipLaunchKernelGGL(                                                         \
-      (g16::gemm_nt_bf16_kernel<g16::MODE_CONV, RELU, HB>), grid,             \
-      dim3(g16::THREADS), 0, stream,                                          \
-      reinterpret_cast<const bf16*>(x.data_ptr()),                            \
-      reinterpret_cast<const bf16*>(w.data_ptr()),                            \
-      reinterpret_cast<bf16*>(y.data_ptr()), bias_p,                          \
-      reinterpret_cast<const bf16*>(zp.data_ptr()), M, Kout, K, 0, 0, 0, cm)
-  if (relu) {
-    if (bias_p) LAUNCH_CV(true, true);
-    else LAUNCH_CV(true, false);
-  } else {
-    if (bias_p) LAUNCH_CV(false, true);
-    else LAUNCH_CV(false, false);
-  }
-#undef LAUNCH_CV
+  if (x.scalar_type() == torch::kBFloat16)
+    launch_nt16<bf16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M, Kout,
+                                      K, 0, 0, 0, cm);
+  else
+    launch_nt16<_Float16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M,
+                                          Kout, K, 0, 0, 0, cm);
   return y;
 }
 
@@ -510,16 +509,21 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   int tiles = ((J + 63) / 64) * ((I + 63) / 64);
   int z = std::max(1, std::min(n_chunks, 2048 / std::max(1, tiles) + 1));
   g16::ConvMeta cm{};
-  for (long long b = 0; b < bsz; ++b) {
-    dim3 grid((J + 63) / 64, (I + 63) / 64, z);
-    hipLaunchKernelGGL(
-        (g16::gemm_tn_bf16_kernel<g16::MODE_PLAIN>), grid, dim3(g16::THREADS),
-        0, stream,
-        reinterpret_cast<const bf16*>(A.data_ptr()) + b * (long long)M * I,
-        reinterpret_cast<const bf16*>(B.data_ptr()) + b * (long long)M * J,
-        C.data_ptr<float>() + b * (long long)I * J, M, I, J, 0, 0,
-        /*ldc=*/J, /*coff=*/0, cm);
-  }
+  auto run = [&](auto tag) {
+    using t16 = decltype(tag);
+    for (long long b = 0; b < bsz; ++b) {
+      dim3 grid((J + 63) / 64, (I + 63) / 64, z);
+      hipLaunchKernelGGL(
+          (g16::gemm_tn_bf16_kernel<t16, g16::MODE_PLAIN>), grid,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()) + b * (long long)M * I,
+          reinterpret_cast<const t16*>(B.data_ptr()) + b * (long long)M * J,
+          C.data_ptr<float>() + b * (long long)I * J, M, I, J, 0, 0,
+          /*ldc=*/J, /*coff=*/0, cm);
+    }
+  };
+  if (A.scalar_type() == torch::kBFloat16) run(bf16{});
+  else run(_Float16{});
   return C;
 }
 
@@ -540,20 +544,25 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   int n_chunks = (M + 31) / 32;
   int tiles = ((Cin + 63) / 64) * ((Kout + 63) / 64);
   int z = std::max(1, std::min(n_chunks, 2048 / std::max(1, tiles) + 1));
-  for (int r = 0; r < (int)R; ++r) {
-    for (int s = 0; s < (int)S; ++s) {
-      dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
-      // dw slice for (r,s): offset (r*S+s)*Cin within each Kout row; row
-      // stride of the f32 output is R*S*Cin.
-      hipLaunchKernelGGL(
-          (g16::gemm_tn_bf16_kernel<g16::MODE_CONV>), grid,
-          dim3(g16::THREADS), 0, stream,
-          reinterpret_cast<const bf16*>(dy.data_ptr()),
-          reinterpret_cast<const bf16*>(x.data_ptr()),
-          dw.data_ptr<float>(), M, Kout, Cin, r, s,
-          /*ldc=*/(long long)R * S * Cin,
-          /*coff=*/(long long)(r * (int)S + s) * Cin, cm);
+  auto run = [&](auto tag) {
+    using t16 = decltype(tag);
+    for (int r = 0; r < (int)R; ++r) {
+      for (int s = 0; s < (int)S; ++s) {
+        dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
+        // dw slice for (r,s): offset (r*S+s)*Cin within each Kout row; row
+        // stride of the f32 output is R*S*Cin.
+        hipLaunchKernelGGL(
+            (g16::gemm_tn_bf16_kernel<t16, g16::MODE_CONV>), grid,
+            dim3(g16::THREADS), 0, stream,
+            reinterpret_cast<const t16*>(dy.data_ptr()),
+            reinterpret_cast<const t16*>(x.data_ptr()),
+            dw.data_ptr<float>(), M, Kout, Cin, r, s,
+            /*ldc=*/(long long)R * S * Cin,
+            /*coff=*/(long long)(r * (int)S + s) * Cin, cm);
+      }
     }
-  }
+  };
+  if (x.scalar_type() == torch::kBFloat16) run(bf16{});
+  else run(_Float16{});
   return dw;
 }

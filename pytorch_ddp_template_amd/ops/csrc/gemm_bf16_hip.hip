@@ -34,7 +34,26 @@
 namespace g16 {
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+// 16-bit dtype traits: fragment vector type + the matching MFMA intrinsic.
+template <typename T16>
+struct M16;
+template <>
+struct M16<bf16> {
+  using vec = bf16x8;
+  DEV_INLINE static f32x4 mma(vec a, vec b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <>
+struct M16<_Float16> {
+  using vec = f16x8;
+  DEV_INLINE static f32x4 mma(vec a, vec b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+};
 
 constexpr int BM = 128, BN = 128, BK = 32;
 constexpr int THREADS = 256;
@@ -51,11 +70,11 @@ struct ConvMeta {
 enum { MODE_PLAIN = 0, MODE_CONV = 1 };
 
 // ---------------------------------------------------------------- NT -----
-template <int MODE, bool RELU, bool HAS_BIAS>
+template <typename T16, int MODE, bool RELU, bool HAS_BIAS>
 __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
-    const bf16* __restrict__ A, const bf16* __restrict__ B,
-    bf16* __restrict__ C, const bf16* __restrict__ bias,
-    const bf16* __restrict__ zpad, int M, int N, int K, long long strideA,
+    const T16* __restrict__ A, const T16* __restrict__ B,
+    T16* __restrict__ C, const T16* __restrict__ bias,
+    const T16* __restrict__ zpad, int M, int N, int K, long long strideA,
     long long strideB, long long strideC, ConvMeta cm) {
   __shared__ __attribute__((aligned(16))) char smem[2 * 2 * TILE_BYTES];  // [buf][A|B]
 
@@ -118,7 +137,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
       // ---- A tile ----
       const int swz = kswz(rl_a[i], kp);
       const int gk = k_base + swz * 8;  // first element of this 16B chunk
-      const bf16* src = zpad;
+      const T16* src = zpad;
       if (MODE == MODE_CONV) {
         if (a_ok[i] && gk < K) {
           const int c0 = gk & ((1 << cm.C_log2) - 1);
@@ -142,7 +161,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
       const int gn = n0 + rlb;
       const int swzb = kswz(rlb, kp);
       const int gkb = k_base + swzb * 8;
-      const bf16* srcb =
+      const T16* srcb =
           (gn < N && gkb < K) ? B + (long long)gn * K + gkb : zpad;
       char* ldsB =
           &smem[buf * 2 * TILE_BYTES + TILE_BYTES + (wave * 2 + i) * 1024];
@@ -165,22 +184,22 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     if (kt + 1 < KT) stage(buf ^ 1, kt + 1);
     const char* baseA = &smem[buf * 2 * TILE_BYTES];
     const char* baseB = baseA + TILE_BYTES;
-    bf16x8 af[4], bf[4];
+    using vec16 = typename M16<T16>::vec;
+    vec16 af[4], bf[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int ra = wm + i * 16 + fr;
-      af[i] = *reinterpret_cast<const bf16x8*>(
+      af[i] = *reinterpret_cast<const vec16*>(
           baseA + ra * 64 + kswz(ra, fs) * 16);
       const int rb = wn + i * 16 + fr;
-      bf[i] = *reinterpret_cast<const bf16x8*>(
+      bf[i] = *reinterpret_cast<const vec16*>(
           baseB + rb * 64 + kswz(rb, fs) * 16);
     }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+        acc[mi][ni] = M16<T16>::mma(af[mi], bf[ni], acc[mi][ni]);
     __syncthreads();
     buf ^= 1;
   }
@@ -190,7 +209,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   for (int ni = 0; ni < 4; ++ni) {
     const int col = n0 + wn + ni * 16 + fr;
     if (col >= N) continue;
-    float bv = HAS_BIAS ? __bfloat162float(bias[col]) : 0.f;
+    float bv = HAS_BIAS ? to_f(bias[col]) : 0.f;
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const int row_base = m0 + wm + mi * 16 + fs * 4;
@@ -200,7 +219,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
         if (row >= M) continue;
         float v = acc[mi][ni][r] + bv;
         if (RELU) v = fmaxf(v, 0.f);
-        C[(long long)row * N + col] = __float2bfloat16(v);
+        C[(long long)row * N + col] = to_t<T16>(v);
       }
     }
   }
@@ -214,9 +233,9 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
 //
 // MODE_CONV gathers the B operand (im2col of x) on the fly for conv wgrad;
 // the j-tile then lives inside one (r,s) slice: j = c0_tile + c.
-template <int MODE>
+template <typename T16, int MODE>
 __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
-    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    const T16* __restrict__ A, const T16* __restrict__ B,
     float* __restrict__ C, int Mtot, int I, int J, int r, int s,
     long long ldc, long long coff, ConvMeta cm) {
   // C element (i, j) lives at C[coff + i*ldc + j] — conv wgrad writes a
@@ -226,9 +245,10 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
   // (row stride 80 B) while staying bank-conflict-free (banks r*20 mod 64
   // are distinct over any 16 consecutive rows).
   constexpr int ROW = BMC + 8;
-  __shared__ __attribute__((aligned(16))) bf16 lds[2 * BI * ROW];
-  bf16* ldsA = lds;
-  bf16* ldsB = lds + BI * ROW;
+  using vec16 = typename M16<T16>::vec;
+  __shared__ __attribute__((aligned(16))) T16 lds[2 * BI * ROW];
+  T16* ldsA = lds;
+  T16* ldsB = lds + BI * ROW;
 
   const int i0 = blockIdx.y * BI;
   const int j0 = blockIdx.x * BJ;
@@ -257,11 +277,11 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
     const int gm = mbase + sm;
     // ---- load + transpose-stage A chunk ([m][I] -> image [i][m]) ----
     {
-      bf16x8 v = {};
+      vec16 v = {};
       if (gm < Mtot && i0 + sc0 < I) {
         const long long off = (long long)gm * I + i0 + sc0;
         if (off + 8 <= (long long)Mtot * I) {
-          v = *reinterpret_cast<const bf16x8*>(A + off);
+          v = *reinterpret_cast<const vec16*>(A + off);
         } else {  // last-row partial chunk: element-wise guarded load
 #pragma unroll
           for (int j = 0; j < 8; ++j)
@@ -284,7 +304,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
     }
     // ---- load + transpose-stage B chunk ----
     {
-      bf16x8 v = {};
+      vec16 v = {};
       if (MODE == MODE_CONV) {
         if (gm < Mtot) {
           int t = gm;
@@ -297,7 +317,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
           const int c = j0 + sc0;
           if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W &&
               c < (1 << cm.C_log2))
-            v = *reinterpret_cast<const bf16x8*>(
+            v = *reinterpret_cast<const vec16*>(
                 B + (((long long)n * cm.H + hi) * cm.W + wi) *
                         (1LL << cm.C_log2) +
                 c);
@@ -306,7 +326,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
         if (gm < Mtot && j0 + sc0 < J) {
           const long long off = (long long)gm * J + j0 + sc0;
           if (off + 8 <= (long long)Mtot * J) {
-            v = *reinterpret_cast<const bf16x8*>(B + off);
+            v = *reinterpret_cast<const vec16*>(B + off);
           } else {
 #pragma unroll
             for (int j = 0; j < 8; ++j)
@@ -329,20 +349,19 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
     }
     __syncthreads();
     // ---- MFMA: k dimension = m (BMC=32 -> one 16x16x32 per frag) ----
-    bf16x8 af[2], bfr[2];
+    vec16 af[2], bfr[2];
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      af[i] = *reinterpret_cast<const bf16x8*>(
+      af[i] = *reinterpret_cast<const vec16*>(
           &ldsA[(wm + i * 16 + fr) * ROW + fs * 8]);
-      bfr[i] = *reinterpret_cast<const bf16x8*>(
+      bfr[i] = *reinterpret_cast<const vec16*>(
           &ldsB[(wn + i * 16 + fr) * ROW + fs * 8]);
     }
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+        acc[mi][ni] = M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni]);
     __syncthreads();
   }
 
@@ -369,17 +388,43 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 
 namespace {
 
-torch::Tensor& zero_page(const torch::Device& dev) {
+torch::Tensor& zero_page(const torch::Device& dev, torch::ScalarType st) {
   static std::unordered_map<int, torch::Tensor> cache;
-  int idx = dev.index();
+  int idx = dev.index() * 4 + (st == torch::kBFloat16 ? 0 : 1);
   auto it = cache.find(idx);
   if (it == cache.end()) {
-    it = cache
-             .emplace(idx, torch::zeros(
-                               {64}, torch::dtype(torch::kBFloat16).device(dev)))
+    it = cache.emplace(idx, torch::zeros({64}, torch::dtype(st).device(dev)))
              .first;
   }
   return it->second;
+}
+
+template <typename t16, int MODE>
+void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
+                 torch::Tensor& C, const c10::optional<torch::Tensor>& bias,
+                 bool relu, const torch::Tensor& zp, dim3 grid, int M, int N,
+                 int K, long long sA, long long sB, long long sC,
+                 g16::ConvMeta cm) {
+  auto stream = c10::hip::getCurrentHIPStream();
+  const t16* bias_p =
+      bias.has_value() ? reinterpret_cast<const t16*>(bias->data_ptr())
+                       : nullptr;
+#define LAUNCH_NT16(RELU, HB)                                                \
+  hipLaunchKernelGGL((g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB>), grid,  \
+                     dim3(g16::THREADS), 0, stream,                          \
+                     reinterpret_cast<const t16*>(A.data_ptr()),             \
+                     reinterpret_cast<const t16*>(B.data_ptr()),             \
+                     reinterpret_cast<t16*>(C.data_ptr()), bias_p,           \
+                     reinterpret_cast<const t16*>(zp.data_ptr()), M, N, K,   \
+                     sA, sB, sC, cm)
+  if (relu) {
+    if (bias_p) LAUNCH_NT16(true, true);
+    else LAUNCH_NT16(true, false);
+  } else {
+    if (bias_p) LAUNCH_NT16(false, true);
+    else LAUNCH_NT16(false, false);
+  }
+#undef LAUNCH_NT16
 }
 
 int log2_exact(int v) {
@@ -402,33 +447,20 @@ torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
   TORCH_CHECK(K % 8 == 0, "K must be padded to a multiple of 8 (host)");
   auto C = batched ? torch::empty({bsz, M, N}, A.options())
                    : torch::empty({M, N}, A.options());
-  auto& zp = zero_page(A.device());
+  auto& zp = zero_page(A.device(), A.scalar_type());
   dim3 grid((N + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM,
             (unsigned)bsz);
   g16::ConvMeta cm{};
   auto stream = c10::hip::getCurrentHIPStream();
-  const bf16* bias_p =
-      bias.has_value() ? reinterpret_cast<const bf16*>(bias->data_ptr())
-                       : nullptr;
   long long sA = batched ? (long long)M * K : 0;
   long long sB = batched ? (long long)N * K : 0;
   long long sC = batched ? (long long)M * N : 0;
-#define LAUNCH_NT(RELU, HB)                                                   \
-  hipLaunchKernelGGL(                                                         \
-      (g16::gemm_nt_bf16_kernel<g16::MODE_PLAIN, RELU, HB>), grid,            \
-      dim3(g16::THREADS), 0, stream,                                          \
-      reinterpret_cast<const bf16*>(A.data_ptr()),                            \
-      reinterpret_cast<const bf16*>(B.data_ptr()),                            \
-      reinterpret_cast<bf16*>(C.data_ptr()), bias_p,                          \
-      reinterpret_cast<const bf16*>(zp.data_ptr()), M, N, K, sA, sB, sC, cm)
-  if (relu) {
-    if (bias_p) LAUNCH_NT(true, true);
-    else LAUNCH_NT(true, false);
-  } else {
-    if (bias_p) LAUNCH_NT(false, true);
-    else LAUNCH_NT(false, false);
-  }
-#undef LAUNCH_NT
+  if (A.scalar_type() == torch::kBFloat16)
+    launch_nt16<bf16, g16::MODE_PLAIN>(A, B, C, bias, relu, zp, grid, M, N, K,
+                                       sA, sB, sC, cm);
+  else
+    launch_nt16<_Float16, g16::MODE_PLAIN>(A, B, C, bias, relu, zp, grid, M,
+                                           N, K, sA, sB, sC, cm);
   return C;
 }
 
@@ -446,29 +478,15 @@ torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
   int WO = (W + 2 * (int)pad - S) / (int)stride + 1;
   int M = N * HO * WO, K = R * S * Cin;
   auto y = torch::empty({N, HO, WO, Kout}, x.options());
-  auto& zp = zero_page(x.device());
+  auto& zp = zero_page(x.device(), x.scalar_type());
   dim3 grid((Kout + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
   g16::ConvMeta cm{H, W, cl, S, R, (int)stride, (int)pad, HO, WO};
-  auto stream = c10::hip::getCurrentHIPStream();
-  const bf16* bias_p =
-      bias.has_value() ? reinterpret_cast<const bf16*>(bias->data_ptr())
-                       : nullptr;
-#define LAUNCH_CV(RELU, HB)                                                   \
-  hipLaunchKernelGGL(                                                         \
-      (g16::gemm_nt_bf16_kernel<g16::MODE_CONV, RELU, HB>), grid,             \
-      dim3(g16::THREADS), 0, stream,                                          \
-      reinterpret_cast<const bf16*>(x.data_ptr()),                            \
-      reinterpret_cast<const bf16*>(w.data_ptr()),                            \
-      reinterpret_cast<bf16*>(y.data_ptr()), bias_p,                          \
-      reinterpret_cast<const bf16*>(zp.data_ptr()), M, Kout, K, 0, 0, 0, cm)
-  if (relu) {
-    if (bias_p) LAUNCH_CV(true, true);
-    else LAUNCH_CV(true, false);
-  } else {
-    if (bias_p) LAUNCH_CV(false, true);
-    else LAUNCH_CV(false, false);
-  }
-#undef LAUNCH_CV
+  if (x.scalar_type() == torch::kBFloat16)
+    launch_nt16<bf16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M, Kout,
+                                      K, 0, 0, 0, cm);
+  else
+    launch_nt16<_Float16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M,
+                                          Kout, K, 0, 0, 0, cm);
   return y;
 }
 
@@ -492,16 +510,21 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   int tiles = ((J + 63) / 64) * ((I + 63) / 64);
   int z = std::max(1, std::min(n_chunks, 2048 / std::max(1, tiles) + 1));
   g16::ConvMeta cm{};
-  for (long long b = 0; b < bsz; ++b) {
-    dim3 grid((J + 63) / 64, (I + 63) / 64, z);
-    hipLaunchKernelGGL(
-        (g16::gemm_tn_bf16_kernel<g16::MODE_PLAIN>), grid, dim3(g16::THREADS),
-        0, stream,
-        reinterpret_cast<const bf16*>(A.data_ptr()) + b * (long long)M * I,
-        reinterpret_cast<const bf16*>(B.data_ptr()) + b * (long long)M * J,
-        C.data_ptr<float>() + b * (long long)I * J, M, I, J, 0, 0,
-        /*ldc=*/J, /*coff=*/0, cm);
-  }
+  auto run = [&](auto tag) {
+    using t16 = decltype(tag);
+    for (long long b = 0; b < bsz; ++b) {
+      dim3 grid((J + 63) / 64, (I + 63) / 64, z);
+      hipLaunchKernelGGL(
+          (g16::gemm_tn_bf16_kernel<t16, g16::MODE_PLAIN>), grid,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()) + b * (long long)M * I,
+          reinterpret_cast<const t16*>(B.data_ptr()) + b * (long long)M * J,
+          C.data_ptr<float>() + b * (long long)I * J, M, I, J, 0, 0,
+          /*ldc=*/J, /*coff=*/0, cm);
+    }
+  };
+  if (A.scalar_type() == torch::kBFloat16) run(bf16{});
+  else run(_Float16{});
   return C;
 }
 
@@ -522,20 +545,25 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   int n_chunks = (M + 31) / 32;
   int tiles = ((Cin + 63) / 64) * ((Kout + 63) / 64);
   int z = std::max(1, std::min(n_chunks, 2048 / std::max(1, tiles) + 1));
-  for (int r = 0; r < (int)R; ++r) {
-    for (int s = 0; s < (int)S; ++s) {
-      dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
-      // dw slice for (r,s): offset (r*S+s)*Cin within each Kout row; row
-      // stride of the f32 output is R*S*Cin.
-      hipLaunchKernelGGL(
-          (g16::gemm_tn_bf16_kernel<g16::MODE_CONV>), grid,
-          dim3(g16::THREADS), 0, stream,
-          reinterpret_cast<const bf16*>(dy.data_ptr()),
-          reinterpret_cast<const bf16*>(x.data_ptr()),
-          dw.data_ptr<float>(), M, Kout, Cin, r, s,
-          /*ldc=*/(long long)R * S * Cin,
-          /*coff=*/(long long)(r * (int)S + s) * Cin, cm);
+  auto run = [&](auto tag) {
+    using t16 = decltype(tag);
+    for (int r = 0; r < (int)R; ++r) {
+      for (int s = 0; s < (int)S; ++s) {
+        dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
+        // dw slice for (r,s): offset (r*S+s)*Cin within each Kout row; row
+        // stride of the f32 output is R*S*Cin.
+        hipLaunchKernelGGL(
+            (g16::gemm_tn_bf16_kernel<t16, g16::MODE_CONV>), grid,
+            dim3(g16::THREADS), 0, stream,
+            reinterpret_cast<const t16*>(dy.data_ptr()),
+            reinterpret_cast<const t16*>(x.data_ptr()),
+            dw.data_ptr<float>(), M, Kout, Cin, r, s,
+            /*ldc=*/(long long)R * S * Cin,
+            /*coff=*/(long long)(r * (int)S + s) * Cin, cm);
+      }
     }
-  }
+  };
+  if (x.scalar_type() == torch::kBFloat16) run(bf16{});
+  else run(_Float16{});
   return dw;
 }

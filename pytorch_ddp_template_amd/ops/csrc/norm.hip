@@ -14,24 +14,67 @@
 
 namespace nrm {
 
-// ---- BN pass 1: per-channel sum / sumsq (atomics over M-split grid) ----
+// ---- BN pass 1: per-channel sum / sumsq ----
+// Thread t owns channel slot c0 = (t % (C/8))*8 and strides rows; every
+// load is a 16-B vector and consecutive threads cover consecutive channel
+// slots, so the sweep is fully coalesced and HBM-bound.  Per-block LDS
+// accumulation, then one global atomic per channel per block.
+constexpr int kBnMaxC = 2048;  // largest channel count (ResNet-50 layer4)
+
 template <typename T>
 __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
                                 float* __restrict__ sumsq, long long M, int C) {
-  const int c = blockIdx.x * kWave + lane_id();
-  if (c >= C) return;
-  const int nw = blockDim.x / kWave;
-  const long long rows_per = (M + gridDim.y - 1) / gridDim.y;
-  const long long m0 = blockIdx.y * rows_per;
-  const long long m1 = min(M, m0 + rows_per);
-  float s = 0.f, ss = 0.f;
-  for (long long m = m0 + wave_id(); m < m1; m += nw) {
-    float v = to_f(x[m * C + c]);
-    s += v;
-    ss += v * v;
+  __shared__ float ssum[kBnMaxC];
+  __shared__ float ssq[kBnMaxC];
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    ssum[c] = 0.f;
+    ssq[c] = 0.f;
   }
-  atomicAdd(&sum[c], s);
-  atomicAdd(&sumsq[c], ss);
+  __syncthreads();
+  using IO = VecIO<T>;
+  const int slots = C / 8;  // C % 8 == 0 (host checks)
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const int c0 = (int)(tid % slots) * 8;
+  const long long row0 = tid / slots;
+  const long long rstride = nthreads / slots;
+  float a0 = 0, a1 = 0, a2 = 0, a3 = 0, a4 = 0, a5 = 0, a6 = 0, a7 = 0;
+  float q0 = 0, q1 = 0, q2 = 0, q3 = 0, q4 = 0, q5 = 0, q6 = 0, q7 = 0;
+  for (long long m = row0; m < M; m += rstride) {
+    const T* p = x + m * C + c0;
+    float v[8];
+    if (IO::kPerLane == 8) {
+      auto vec = *reinterpret_cast<const typename VecIO<T>::Vec*>(p);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = IO::get(vec, j);
+    } else {  // f32: two 16-B loads
+      auto va = *reinterpret_cast<const float4v*>(p);
+      auto vb = *reinterpret_cast<const float4v*>(p + 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        v[j] = va[j];
+        v[4 + j] = vb[j];
+      }
+    }
+    a0 += v[0]; a1 += v[1]; a2 += v[2]; a3 += v[3];
+    a4 += v[4]; a5 += v[5]; a6 += v[6]; a7 += v[7];
+    q0 += v[0]*v[0]; q1 += v[1]*v[1]; q2 += v[2]*v[2]; q3 += v[3]*v[3];
+    q4 += v[4]*v[4]; q5 += v[5]*v[5]; q6 += v[6]*v[6]; q7 += v[7]*v[7];
+  }
+  float a[8] = {a0,a1,a2,a3,a4,a5,a6,a7};
+  float q[8] = {q0,q1,q2,q3,q4,q5,q6,q7};
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&ssum[c0 + j], a[j]);
+    atomicAdd(&ssq[c0 + j], q[j]);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    if (ssum[c] != 0.f || ssq[c] != 0.f) {
+      atomicAdd(&sum[c], ssum[c]);
+      atomicAdd(&sumsq[c], ssq[c]);
+    }
+  }
 }
 
 // ---- BN pass 2: finalize mean/rstd + update running stats ----
@@ -58,18 +101,44 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
 }
 
 // ---- BN pass 3: normalize (+optional fused relu) ----
+// Vectorized: each thread owns one 8-channel slot and strides rows, so the
+// per-channel scale/shift loads hoist out of the row loop.
 template <typename T, bool RELU>
 __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
                                const float* __restrict__ mean,
                                const float* __restrict__ rstd,
                                const T* __restrict__ gamma,
                                const T* __restrict__ beta, long long M, int C) {
-  const long long total = M * C;
-  GRID_STRIDE(i, total) {
-    const int c = (int)(i % C);
-    float v = (to_f(x[i]) - mean[c]) * rstd[c] * to_f(gamma[c]) + to_f(beta[c]);
-    if (RELU) v = fmaxf(v, 0.f);
-    y[i] = to_t<T>(v);
+  using IO = VecIO<T>;
+  const int slots = C / 8;
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const int c0 = (int)(tid % slots) * 8;
+  const long long row0 = tid / slots;
+  const long long rstride = nthreads / slots;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    sc[j] = rstd[c0 + j] * to_f(gamma[c0 + j]);
+    sh[j] = to_f(beta[c0 + j]) - mean[c0 + j] * sc[j];
+  }
+  for (long long m = row0; m < M; m += rstride) {
+    if (IO::kPerLane == 8) {
+      auto v = *reinterpret_cast<const typename VecIO<T>::Vec*>(x + m * C + c0);
+      typename VecIO<T>::Vec o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float r = IO::get(v, j) * sc[j] + sh[j];
+        IO::set(o, j, RELU ? fmaxf(r, 0.f) : r);
+      }
+      *reinterpret_cast<typename VecIO<T>::Vec*>(y + m * C + c0) = o;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float r = to_f(x[m * C + c0 + j]) * sc[j] + sh[j];
+        y[m * C + c0 + j] = to_t<T>(RELU ? fmaxf(r, 0.f) : r);
+      }
+    }
   }
 }
 
@@ -92,6 +161,7 @@ __global__ void bn_infer_kernel(const T* __restrict__ x, T* __restrict__ y,
 }
 
 // ---- BN bwd pass 1: dgamma = sum dy*xhat, dbeta = sum dy ----
+// Same coalesced 8-channel-per-thread layout as bn_stats_kernel.
 template <typename T>
 __global__ void bn_bwd_stats_kernel(const T* __restrict__ dy,
                                     const T* __restrict__ x,
@@ -100,25 +170,71 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ dy,
                                     float* __restrict__ dgamma,
                                     float* __restrict__ dbeta, long long M,
                                     int C) {
-  const int c = blockIdx.x * kWave + lane_id();
-  if (c >= C) return;
-  const int nw = blockDim.x / kWave;
-  const long long rows_per = (M + gridDim.y - 1) / gridDim.y;
-  const long long m0 = blockIdx.y * rows_per;
-  const long long m1 = min(M, m0 + rows_per);
-  const float mu = mean[c], rs = rstd[c];
-  float dg = 0.f, db = 0.f;
-  for (long long m = m0 + wave_id(); m < m1; m += nw) {
-    float g = to_f(dy[m * C + c]);
-    float xh = (to_f(x[m * C + c]) - mu) * rs;
-    dg += g * xh;
-    db += g;
+  __shared__ float sdg[kBnMaxC];
+  __shared__ float sdb[kBnMaxC];
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    sdg[c] = 0.f;
+    sdb[c] = 0.f;
   }
-  atomicAdd(&dgamma[c], dg);
-  atomicAdd(&dbeta[c], db);
+  __syncthreads();
+  using IO = VecIO<T>;
+  const int slots = C / 8;
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const int c0 = (int)(tid % slots) * 8;
+  const long long row0 = tid / slots;
+  const long long rstride = nthreads / slots;
+  float mu[8], rs[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mu[j] = mean[c0 + j];
+    rs[j] = rstd[c0 + j];
+  }
+  float dg[8] = {}, db[8] = {};
+  for (long long m = row0; m < M; m += rstride) {
+    const T* pg = dy + m * C + c0;
+    const T* px = x + m * C + c0;
+    float g[8], v[8];
+    if (IO::kPerLane == 8) {
+      auto vg = *reinterpret_cast<const typename VecIO<T>::Vec*>(pg);
+      auto vx = *reinterpret_cast<const typename VecIO<T>::Vec*>(px);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        g[j] = IO::get(vg, j);
+        v[j] = IO::get(vx, j);
+      }
+    } else {
+      auto ga = *reinterpret_cast<const float4v*>(pg);
+      auto gb = *reinterpret_cast<const float4v*>(pg + 4);
+      auto xa = *reinterpret_cast<const float4v*>(px);
+      auto xb = *reinterpret_cast<const float4v*>(px + 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        g[j] = ga[j]; g[4 + j] = gb[j];
+        v[j] = xa[j]; v[4 + j] = xb[j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      db[j] += g[j];
+      dg[j] += g[j] * (v[j] - mu[j]) * rs[j];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&sdg[c0 + j], dg[j]);
+    atomicAdd(&sdb[c0 + j], db[j]);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    if (sdg[c] != 0.f || sdb[c] != 0.f) {
+      atomicAdd(&dgamma[c], sdg[c]);
+      atomicAdd(&dbeta[c], sdb[c]);
+    }
+  }
 }
 
-// ---- BN bwd pass 2: dx ----
+// ---- BN bwd pass 2: dx (vectorized 8-channel slots) ----
 template <typename T>
 __global__ void bn_bwd_dx_kernel(const T* __restrict__ dy,
                                  const T* __restrict__ x, T* __restrict__ dx,
@@ -128,16 +244,44 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ dy,
                                  const float* __restrict__ dgamma,
                                  const float* __restrict__ dbeta, long long M,
                                  int C) {
+  using IO = VecIO<T>;
   const float invM = 1.f / (float)M;
-  const long long total = M * C;
-  GRID_STRIDE(i, total) {
-    const int c = (int)(i % C);
-    const float mu = mean[c], rs = rstd[c];
-    float g = to_f(dy[i]);
-    float xh = (to_f(x[i]) - mu) * rs;
-    float v = to_f(gamma[c]) * rs *
-              (g - invM * (dbeta[c] + xh * dgamma[c]));
-    dx[i] = to_t<T>(v);
+  const int slots = C / 8;
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const int c0 = (int)(tid % slots) * 8;
+  const long long row0 = tid / slots;
+  const long long rstride = nthreads / slots;
+  float mu[8], rs[8], gm[8], dgv[8], dbv[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mu[j] = mean[c0 + j];
+    rs[j] = rstd[c0 + j];
+    gm[j] = to_f(gamma[c0 + j]);
+    dgv[j] = dgamma[c0 + j];
+    dbv[j] = dbeta[c0 + j];
+  }
+  for (long long m = row0; m < M; m += rstride) {
+    if (IO::kPerLane == 8) {
+      auto vg = *reinterpret_cast<const typename VecIO<T>::Vec*>(dy + m * C + c0);
+      auto vx = *reinterpret_cast<const typename VecIO<T>::Vec*>(x + m * C + c0);
+      typename VecIO<T>::Vec o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = IO::get(vg, j);
+        float xh = (IO::get(vx, j) - mu[j]) * rs[j];
+        IO::set(o, j, gm[j] * rs[j] * (g - invM * (dbv[j] + xh * dgv[j])));
+      }
+      *reinterpret_cast<typename VecIO<T>::Vec*>(dx + m * C + c0) = o;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = to_f(dy[m * C + c0 + j]);
+        float xh = (to_f(x[m * C + c0 + j]) - mu[j]) * rs[j];
+        dx[m * C + c0 + j] =
+            to_t<T>(gm[j] * rs[j] * (g - invM * (dbv[j] + xh * dgv[j])));
+      }
+    }
   }
 }
 
@@ -272,8 +416,9 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   auto rstd = torch::empty({C}, f32);
   auto y = torch::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream();
-  dim3 sgrid((C + kWave - 1) / kWave,
-             (unsigned)std::min<long long>(64, (M + 4095) / 4096) );
+  TORCH_CHECK(C % 8 == 0 && C <= 2048,
+              "bn kernels need C % 8 == 0 and C <= 2048 (got ", C, ")");
+  dim3 sgrid(grid_1d(M * C / 8, 256));
   DDP_DISPATCH_FLOAT(x.scalar_type(), "bn_fwd", [&] {
     const auto* xp = reinterpret_cast<const scalar_t*>(x.data_ptr());
     hipLaunchKernelGGL((nrm::bn_stats_kernel<scalar_t>), sgrid, dim3(256), 0,
@@ -291,14 +436,14 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     auto* yp = reinterpret_cast<scalar_t*>(y.data_ptr());
     if (relu)
       hipLaunchKernelGGL((nrm::bn_norm_kernel<scalar_t, true>),
-                         dim3(grid_1d(M * C, 256)), dim3(256), 0, stream, xp,
-                         yp, mean.data_ptr<float>(), rstd.data_ptr<float>(), gp,
-                         bp, M, C);
+                         dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
+                         xp, yp, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         gp, bp, M, C);
     else
       hipLaunchKernelGGL((nrm::bn_norm_kernel<scalar_t, false>),
-                         dim3(grid_1d(M * C, 256)), dim3(256), 0, stream, xp,
-                         yp, mean.data_ptr<float>(), rstd.data_ptr<float>(), gp,
-                         bp, M, C);
+                         dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
+                         xp, yp, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         gp, bp, M, C);
   });
   return {y, mean, rstd};
 }
@@ -339,8 +484,9 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto dbeta = torch::zeros({C}, f32);
   auto dx = torch::empty_like(dy);
   auto stream = c10::hip::getCurrentHIPStream();
-  dim3 sgrid((C + kWave - 1) / kWave,
-             (unsigned)std::min<long long>(64, (M + 4095) / 4096));
+  TORCH_CHECK(C % 8 == 0 && C <= 2048,
+              "bn kernels need C % 8 == 0 and C <= 2048 (got ", C, ")");
+  dim3 sgrid(grid_1d(M * C / 8, 256));
   DDP_DISPATCH_FLOAT(x.scalar_type(), "bn_bwd", [&] {
     const auto* dyp = reinterpret_cast<const scalar_t*>(dy.data_ptr());
     const auto* xp = reinterpret_cast<const scalar_t*>(x.data_ptr());
@@ -350,8 +496,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                        rstd.data_ptr<float>(), dgamma.data_ptr<float>(),
                        dbeta.data_ptr<float>(), M, C);
     hipLaunchKernelGGL((nrm::bn_bwd_dx_kernel<scalar_t>),
-                       dim3(grid_1d(M * C, 256)), dim3(256), 0, stream, dyp,
-                       xp, reinterpret_cast<scalar_t*>(dx.data_ptr()),
+                       dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
+                       dyp, xp, reinterpret_cast<scalar_t*>(dx.data_ptr()),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(), gp,
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), M, C);
   });

@@ -91,7 +91,9 @@ void scale_by_tensor_(std::vector<torch::Tensor> ts, torch::Tensor s);
 namespace {
 
 bool is_bf16(const torch::Tensor& t) {
-  return t.scalar_type() == torch::kBFloat16;
+  // both 16-bit dtypes route to the (templated) 16-bit MFMA kernels
+  return t.scalar_type() == torch::kBFloat16 ||
+         t.scalar_type() == torch::kHalf;
 }
 
 int k_granule(const torch::Tensor& t) { return is_bf16(t) ? 8 : 4; }
