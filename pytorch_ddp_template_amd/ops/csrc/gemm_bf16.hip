@@ -381,6 +381,146 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
   }
 }
 
+// ------------------------------------------------------------ TN conv ----
+// Conv wgrad with the S kernel columns of one kernel row fused into a single
+// launch: the dy chunk is transpose-staged ONCE per m-chunk and reused for
+// all NS column shifts (the x windows of adjacent s overlap, so their
+// re-reads hit L1/L2).  Cuts wgrad HBM traffic ~S x vs the per-(r,s) launch.
+template <typename T16, int NS>
+__global__ __launch_bounds__(THREADS) void gemm_tn_conv_kernel(
+    const T16* __restrict__ A /*dy*/, const T16* __restrict__ B /*x*/,
+    float* __restrict__ C, int Mtot, int I /*Kout*/, int r, long long ldc,
+    long long coff_base /*offset of (r, s=0) slice*/, ConvMeta cm) {
+  constexpr int BI = 64, BJ = 64, BMC = 32;
+  constexpr int ROW = BMC + 8;
+  using vec16 = typename M16<T16>::vec;
+  __shared__ __attribute__((aligned(16))) T16 lds[2 * BI * ROW];
+  T16* ldsA = lds;
+  T16* ldsB = lds + BI * ROW;
+
+  const int Cin = 1 << cm.C_log2;
+  const int i0 = blockIdx.y * BI;
+  const int j0 = blockIdx.x * BJ;  // channel tile
+  const int n_chunks = (Mtot + BMC - 1) / BMC;
+  const int per_z = (n_chunks + gridDim.z - 1) / gridDim.z;
+  const int ch0 = blockIdx.z * per_z;
+  const int ch1 = min(n_chunks, ch0 + per_z);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
+  const int fr = lane & 15;
+  const int fs = lane >> 4;
+  const int sm = threadIdx.x >> 3;
+  const int sc0 = (threadIdx.x & 7) * 8;
+
+  f32x4 acc[2][2][NS] = {};
+
+  for (int ch = ch0; ch < ch1; ++ch) {
+    const int mbase = ch * BMC;
+    const int gm = mbase + sm;
+    // per-chunk source decode (reused across the NS shifts)
+    int n = 0, hb = 0, wb = 0;
+    if (gm < Mtot) {
+      int t = gm;
+      const int wo = t % cm.WO;
+      t /= cm.WO;
+      const int ho = t % cm.HO;
+      n = t / cm.HO;
+      hb = ho * cm.stride - cm.pad + r;
+      wb = wo * cm.stride - cm.pad;
+    }
+    // ---- stage dy chunk once ----
+    {
+      vec16 v = {};
+      if (gm < Mtot && i0 + sc0 < I) {
+        const long long off = (long long)gm * I + i0 + sc0;
+        if (off + 8 <= (long long)Mtot * I) {
+          v = *reinterpret_cast<const vec16*>(A + off);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (off + j < (long long)Mtot * I) v[j] = A[off + j];
+        }
+      }
+      short8 mine = *reinterpret_cast<short8*>(&v);
+      short8 other;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) other[j] = __shfl_xor((int)mine[j], 8);
+      if (((threadIdx.x >> 3) & 1) == 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          unsigned int pack = ((unsigned short)mine[j]) |
+                              (((unsigned int)(unsigned short)other[j]) << 16);
+          *reinterpret_cast<unsigned int*>(&ldsA[(sc0 + j) * ROW + sm]) = pack;
+        }
+      }
+    }
+#pragma unroll
+    for (int si = 0; si < NS; ++si) {
+      // ---- stage x chunk for shift s = si ----
+      {
+        vec16 v = {};
+        const int hi = hb;
+        const int wi = wb + si;
+        if (gm < Mtot && hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W &&
+            j0 + sc0 < Cin)
+          v = *reinterpret_cast<const vec16*>(
+              B + (((long long)n * cm.H + hi) * cm.W + wi) * Cin + j0 + sc0);
+        short8 mine = *reinterpret_cast<short8*>(&v);
+        short8 other;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) other[j] = __shfl_xor((int)mine[j], 8);
+        if (((threadIdx.x >> 3) & 1) == 0) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            unsigned int pack =
+                ((unsigned short)mine[j]) |
+                (((unsigned int)(unsigned short)other[j]) << 16);
+            *reinterpret_cast<unsigned int*>(&ldsB[(sc0 + j) * ROW + sm]) =
+                pack;
+          }
+        }
+      }
+      __syncthreads();
+      vec16 af[2], bfr[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        af[i] = *reinterpret_cast<const vec16*>(
+            &ldsA[(wm + i * 16 + fr) * ROW + fs * 8]);
+        bfr[i] = *reinterpret_cast<const vec16*>(
+            &ldsB[(wn + i * 16 + fr) * ROW + fs * 8]);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni][si] =
+              M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni][si]);
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int si = 0; si < NS; ++si) {
+    const long long coff = coff_base + (long long)si * Cin;
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int col = j0 + wn + ni * 16 + fr;
+      if (col >= Cin) continue;
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          const int row = i0 + wm + mi * 16 + fs * 4 + rr;
+          if (row >= I) continue;
+          atomicAdd(&C[coff + (long long)row * ldc + col],
+                    acc[mi][ni][si][rr]);
+        }
+    }
+  }
+}
+
 }  // namespace g16
 
 // ======================= host-side helpers ===============================
@@ -547,19 +687,32 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
     for (int r = 0; r < (int)R; ++r) {
-      for (int s = 0; s < (int)S; ++s) {
-        dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
-        // dw slice for (r,s): offset (r*S+s)*Cin within each Kout row; row
-        // stride of the f32 output is R*S*Cin.
-        hipLaunchKernelGGL(
-            (g16::gemm_tn_bf16_kernel<t16, g16::MODE_CONV>), grid,
-            dim3(g16::THREADS), 0, stream,
-            reinterpret_cast<const t16*>(dy.data_ptr()),
-            reinterpret_cast<const t16*>(x.data_ptr()),
-            dw.data_ptr<float>(), M, Kout, Cin, r, s,
-            /*ldc=*/(long long)R * S * Cin,
-            /*coff=*/(long long)(r * (int)S + s) * Cin, cm);
-      }
+      dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
+      // one launch per kernel ROW: the S column shifts are fused in-kernel
+      // (dy staged once, x windows overlap in cache).
+      const long long ldc = (long long)R * S * Cin;
+      const long long coff = (long long)r * (int)S * Cin;
+      if (S == 3)
+        hipLaunchKernelGGL((g16::gemm_tn_conv_kernel<t16, 3>), grid,
+                           dim3(g16::THREADS), 0, stream,
+                           reinterpret_cast<const t16*>(dy.data_ptr()),
+                           reinterpret_cast<const t16*>(x.data_ptr()),
+                           dw.data_ptr<float>(), M, Kout, r, ldc, coff, cm);
+      else if (S == 1)
+        hipLaunchKernelGGL((g16::gemm_tn_conv_kernel<t16, 1>), grid,
+                           dim3(g16::THREADS), 0, stream,
+                           reinterpret_cast<const t16*>(dy.data_ptr()),
+                           reinterpret_cast<const t16*>(x.data_ptr()),
+                           dw.data_ptr<float>(), M, Kout, r, ldc, coff, cm);
+      else
+        for (int s2 = 0; s2 < (int)S; ++s2)
+          hipLaunchKernelGGL(
+              (g16::gemm_tn_bf16_kernel<t16, g16::MODE_CONV>), grid,
+              dim3(g16::THREADS), 0, stream,
+              reinterpret_cast<const t16*>(dy.data_ptr()),
+              reinterpret_cast<const t16*>(x.data_ptr()),
+              dw.data_ptr<float>(), M, Kout, Cin, r, s2, ldc,
+              coff + (long long)s2 * Cin, cm);
     }
   };
   if (x.scalar_type() == torch::kBFloat16) run(bf16{});
