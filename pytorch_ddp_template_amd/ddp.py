@@ -42,6 +42,7 @@ from torch.utils.data import DataLoader, RandomSampler
 from .data import ShardedSampler, build_dataset
 from .models import build_model
 from .optim import SGD, clip_grad_norm_, get_linear_schedule_with_warmup
+from .ops import scale_grads_
 from .ops import CrossEntropyLoss, MSELoss
 from .parallel import DistributedModel
 from .utils import (
@@ -197,6 +198,31 @@ def load_checkpoint(args, model, optimizer=None, scheduler=None, path=None):
     return step
 
 
+class LossScaler:
+    """fp16 loss scaling: static (``--loss_scale N``) or dynamic (N == 0,
+    the reference's apex-style default — ddp.py:174-180 intent, natively)."""
+
+    def __init__(self, static_scale: float = 0.0):
+        self.dynamic = static_scale == 0
+        self.scale = static_scale if static_scale > 0 else 65536.0
+        self.growth_interval = 2000
+        self._good_steps = 0
+
+    def step_ok(self, total_norm) -> bool:
+        finite = bool(torch.isfinite(total_norm).all())
+        if not self.dynamic:
+            return finite
+        if finite:
+            self._good_steps += 1
+            if self._good_steps >= self.growth_interval:
+                self.scale *= 2.0
+                self._good_steps = 0
+            return True
+        self.scale = max(1.0, self.scale / 2.0)
+        self._good_steps = 0
+        return False
+
+
 def _criterion_for(args):
     if args.model in ("foo", "foomodel", "mlp"):
         return MSELoss()  # reference ddp.py:164
@@ -334,6 +360,12 @@ def train(args, model):
         ),
     )
 
+    # Native fp16 loss scaling (replaces the reference's broken apex branch,
+    # ddp.py:165-181: static scale when --loss_scale > 0, else dynamic).
+    scaler = (
+        LossScaler(args.loss_scale) if (args.fp16 and not args.bf16) else None
+    )
+
     model.train()
     model.zero_grad()
     device_loss = torch.zeros((), device=args.device)
@@ -365,16 +397,28 @@ def train(args, model):
                 loss = criterion(outputs, y)
                 if args.gradient_accumulation_steps > 1:
                     loss = loss / args.gradient_accumulation_steps
-                loss.backward()
+                if scaler is not None:
+                    (loss * scaler.scale).backward()
+                else:
+                    loss.backward()
             device_loss += loss.detach()
             steps_since_log += 1
 
             if accum_boundary:
                 if isinstance(model, DistributedModel):
                     model.finish_gradient_sync()
-                clip_grad_norm_(
+                if scaler is not None:
+                    grads = [
+                        p.grad for p in model.parameters() if p.grad is not None
+                    ]
+                    scale_grads_(grads, 1.0 / scaler.scale)
+                total_norm = clip_grad_norm_(
                     [p for p in model.parameters()], args.max_grad_norm
                 )
+                if scaler is not None and not scaler.step_ok(total_norm):
+                    # inf/nan grads: skip the step, shrink the scale
+                    model.zero_grad()
+                    continue
                 optimizer.step()
                 scheduler.step()
                 model.zero_grad()
