@@ -64,6 +64,11 @@ DEV_INLINE int kswz(int row, int kp) { return kp ^ ((row >> 2) & 3); }
 struct ConvMeta {
   int H, W, C_log2, S, R, stride, pad;
   int HO, WO;
+  // dgrad zero-stuffing, folded into the im2col gather: the logical [H,W]
+  // image is the stride-`ss` zero-stuffing of a physical [SH,SW] source
+  // (dy).  ss <= 1 means dense; brace-inits that omit these get 0 = dense,
+  // so no materialized zero-stuffed tensor (or its 2x HBM round trip) exists.
+  int ss, SH, SW;
 };
 
 enum { MODE_PLAIN = 0, MODE_CONV = 1 };
@@ -143,10 +148,20 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
           const int rs = gk >> cm.C_log2;
           const int r = rs / cm.S, s = rs % cm.S;
           const int hi = a_hb[i] + r, wi = a_wb[i] + s;
-          if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W)
-            src = A + (((long long)a_n[i] * cm.H + hi) * cm.W + wi) *
-                          (1LL << cm.C_log2) +
-                  c0;
+          if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W) {
+            if (cm.ss <= 1) {
+              src = A + (((long long)a_n[i] * cm.H + hi) * cm.W + wi) *
+                            (1LL << cm.C_log2) +
+                    c0;
+            } else {  // stuffed dgrad source: only every ss-th point is real
+              const int hs = hi / cm.ss, ws = wi / cm.ss;
+              if (hi == hs * cm.ss && wi == ws * cm.ss && hs < cm.SH &&
+                  ws < cm.SW)
+                src = A + (((long long)a_n[i] * cm.SH + hs) * cm.SW + ws) *
+                              (1LL << cm.C_log2) +
+                      c0;
+            }
+          }
         }
       } else {
         if (a_ok[i] && gk < K) src = A + (long long)(m0 + rl_a[i]) * K + gk;
@@ -383,9 +398,12 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 
 // ------------------------------------------------------------ TN conv ----
 // Conv wgrad with the S kernel columns of one kernel row fused into a single
-// launch: the dy chunk is transpose-staged ONCE per m-chunk and reused for
-// all NS column shifts (the x windows of adjacent s overlap, so their
-// re-reads hit L1/L2).  Cuts wgrad HBM traffic ~S x vs the per-(r,s) launch.
+// launch, software-pipelined: per 32-m chunk the dy tile and all NS shifted
+// x tiles are transpose-staged into one of two LDS buffers, the next chunk's
+// global loads are issued before the MFMAs of the current chunk (their
+// latency hides under the matrix work), and there is exactly ONE barrier per
+// chunk.  The x windows of adjacent s overlap so their re-reads hit L1.
+// Host keeps total blocks ~768 so the final f32 atomic fan-in stays small.
 template <typename T16, int NS>
 __global__ __launch_bounds__(THREADS) void gemm_tn_conv_kernel(
     const T16* __restrict__ A /*dy*/, const T16* __restrict__ B /*x*/,
@@ -393,10 +411,10 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_conv_kernel(
     long long coff_base /*offset of (r, s=0) slice*/, ConvMeta cm) {
   constexpr int BI = 64, BJ = 64, BMC = 32;
   constexpr int ROW = BMC + 8;
+  constexpr int TILE = BI * ROW;  // elements per staged operand tile
   using vec16 = typename M16<T16>::vec;
-  __shared__ __attribute__((aligned(16))) T16 lds[2 * BI * ROW];
-  T16* ldsA = lds;
-  T16* ldsB = lds + BI * ROW;
+  // [2 pipeline buffers][A | B shift 0..NS-1]
+  __shared__ __attribute__((aligned(16))) T16 lds[2 * (1 + NS) * TILE];
 
   const int Cin = 1 << cm.C_log2;
   const int i0 = blockIdx.y * BI;
@@ -405,99 +423,112 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_conv_kernel(
   const int per_z = (n_chunks + gridDim.z - 1) / gridDim.z;
   const int ch0 = blockIdx.z * per_z;
   const int ch1 = min(n_chunks, ch0 + per_z);
+  if (ch0 >= ch1) return;  // uniform per block: no barrier divergence
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
   const int fr = lane & 15;
   const int fs = lane >> 4;
-  const int sm = threadIdx.x >> 3;
-  const int sc0 = (threadIdx.x & 7) * 8;
+  const int sm = threadIdx.x >> 3;        // m row in chunk (0..31)
+  const int sc0 = (threadIdx.x & 7) * 8;  // first col of this 16B piece
 
   f32x4 acc[2][2][NS] = {};
+  vec16 va, vb[NS];  // in-flight global data for the chunk being staged
 
-  for (int ch = ch0; ch < ch1; ++ch) {
-    const int mbase = ch * BMC;
-    const int gm = mbase + sm;
-    // per-chunk source decode (reused across the NS shifts)
-    int n = 0, hb = 0, wb = 0;
+  // global fetch of chunk ch into registers (dy tile + NS shifted x tiles)
+  auto fetch = [&](int ch) {
+    const int gm = ch * BMC + sm;
+    va = vec16{};
+#pragma unroll
+    for (int si = 0; si < NS; ++si) vb[si] = vec16{};
     if (gm < Mtot) {
+      if (i0 + sc0 < I) {
+        const long long off = (long long)gm * I + i0 + sc0;
+        if (off + 8 <= (long long)Mtot * I) {
+          va = *reinterpret_cast<const vec16*>(A + off);
+        } else {  // last-row partial chunk: element-wise guarded load
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (off + j < (long long)Mtot * I) va[j] = A[off + j];
+        }
+      }
       int t = gm;
       const int wo = t % cm.WO;
       t /= cm.WO;
       const int ho = t % cm.HO;
-      n = t / cm.HO;
-      hb = ho * cm.stride - cm.pad + r;
-      wb = wo * cm.stride - cm.pad;
-    }
-    // ---- stage dy chunk once ----
-    {
-      vec16 v = {};
-      if (gm < Mtot && i0 + sc0 < I) {
-        const long long off = (long long)gm * I + i0 + sc0;
-        if (off + 8 <= (long long)Mtot * I) {
-          v = *reinterpret_cast<const vec16*>(A + off);
-        } else {
+      const int n = t / cm.HO;
+      const int hi = ho * cm.stride - cm.pad + r;
+      const int wb = wo * cm.stride - cm.pad;
+      if (hi >= 0 && hi < cm.H && j0 + sc0 < Cin) {
+        const long long rowoff =
+            ((long long)n * cm.H + hi) * cm.W * Cin + j0 + sc0;
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (off + j < (long long)Mtot * I) v[j] = A[off + j];
+        for (int si = 0; si < NS; ++si) {
+          const int wi = wb + si;
+          if (wi >= 0 && wi < cm.W)
+            vb[si] = *reinterpret_cast<const vec16*>(
+                B + rowoff + (long long)wi * Cin);
         }
       }
+    }
+  };
+
+  // transpose-pack registers -> LDS buffer `buf` (lane pairs 8 apart pack
+  // two adjacent m rows into one b32; image layout [col][m])
+  auto stage = [&](int buf) {
+    T16* base = lds + buf * (1 + NS) * TILE;
+    const bool writer = ((threadIdx.x >> 3) & 1) == 0;
+#pragma unroll
+    for (int op = 0; op < 1 + NS; ++op) {
+      vec16 v = (op == 0) ? va : vb[op - 1];
       short8 mine = *reinterpret_cast<short8*>(&v);
       short8 other;
 #pragma unroll
       for (int j = 0; j < 8; ++j) other[j] = __shfl_xor((int)mine[j], 8);
-      if (((threadIdx.x >> 3) & 1) == 0) {
+      if (writer) {
+        T16* dst = base + op * TILE;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           unsigned int pack = ((unsigned short)mine[j]) |
                               (((unsigned int)(unsigned short)other[j]) << 16);
-          *reinterpret_cast<unsigned int*>(&ldsA[(sc0 + j) * ROW + sm]) = pack;
+          *reinterpret_cast<unsigned int*>(&dst[(sc0 + j) * ROW + sm]) = pack;
         }
       }
     }
+  };
+
+  fetch(ch0);
+  stage(0);
+  __syncthreads();
+
+  for (int ch = ch0; ch < ch1; ++ch) {
+    const int buf = (ch - ch0) & 1;
+    const bool more = ch + 1 < ch1;
+    if (more) fetch(ch + 1);  // latency hides under the MFMAs below
+    const T16* base = lds + buf * (1 + NS) * TILE;
+    vec16 af[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      af[i] = *reinterpret_cast<const vec16*>(
+          &base[(wm + i * 16 + fr) * ROW + fs * 8]);
 #pragma unroll
     for (int si = 0; si < NS; ++si) {
-      // ---- stage x chunk for shift s = si ----
-      {
-        vec16 v = {};
-        const int hi = hb;
-        const int wi = wb + si;
-        if (gm < Mtot && hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W &&
-            j0 + sc0 < Cin)
-          v = *reinterpret_cast<const vec16*>(
-              B + (((long long)n * cm.H + hi) * cm.W + wi) * Cin + j0 + sc0);
-        short8 mine = *reinterpret_cast<short8*>(&v);
-        short8 other;
+      const T16* bb = base + (1 + si) * TILE;
+      vec16 bfr[2];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) other[j] = __shfl_xor((int)mine[j], 8);
-        if (((threadIdx.x >> 3) & 1) == 0) {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            unsigned int pack =
-                ((unsigned short)mine[j]) |
-                (((unsigned int)(unsigned short)other[j]) << 16);
-            *reinterpret_cast<unsigned int*>(&ldsB[(sc0 + j) * ROW + sm]) =
-                pack;
-          }
-        }
-      }
-      __syncthreads();
-      vec16 af[2], bfr[2];
-#pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        af[i] = *reinterpret_cast<const vec16*>(
-            &ldsA[(wm + i * 16 + fr) * ROW + fs * 8]);
+      for (int i = 0; i < 2; ++i)
         bfr[i] = *reinterpret_cast<const vec16*>(
-            &ldsB[(wn + i * 16 + fr) * ROW + fs * 8]);
-      }
+            &bb[(wn + i * 16 + fr) * ROW + fs * 8]);
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 2; ++ni)
-          acc[mi][ni][si] =
-              M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni][si]);
-      __syncthreads();
+          acc[mi][ni][si] = M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni][si]);
+    }
+    if (more) {
+      stage(buf ^ 1);
+      __syncthreads();  // the ONE barrier per chunk
     }
   }
 
@@ -520,6 +551,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_conv_kernel(
     }
   }
 }
+
 
 }  // namespace g16
 
@@ -629,6 +661,43 @@ torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
+// conv2d dgrad, direct: dx = conv(zero_stuffed(dy), wr, 1, R-1-pad) with the
+// zero-stuffing folded into the im2col gather (ConvMeta.ss) — the stuffed
+// tensor is never materialized.  dy[N,HO,WO,Kout], wr[Cin,R,S,Kout] (the
+// 180°-rotated weight), stride/pad are the ORIGINAL forward conv's.
+torch::Tensor conv2d_dgrad_bf16(torch::Tensor dy, torch::Tensor wr,
+                                int64_t stride, int64_t pad, int64_t H,
+                                int64_t W) {
+  int N = (int)dy.size(0), HOs = (int)dy.size(1), WOs = (int)dy.size(2),
+      Kout = (int)dy.size(3);
+  int Cin = (int)wr.size(0), R = (int)wr.size(1), S = (int)wr.size(2);
+  TORCH_CHECK((int)wr.size(3) == Kout);
+  int cl = log2_exact(Kout);
+  TORCH_CHECK(cl >= 3, "dgrad fast path needs pow2 Kout >= 8");
+  // logical (stuffed) image dims: (HO-1)*stride + 1 + output-padding
+  const int opad_h = (int)((H + 2 * pad - R) % stride);
+  const int opad_w = (int)((W + 2 * pad - S) % stride);
+  const int HS = (HOs - 1) * (int)stride + 1 + opad_h;
+  const int WS = (WOs - 1) * (int)stride + 1 + opad_w;
+  const int dpad = R - 1 - (int)pad;
+  TORCH_CHECK(dpad >= 0 && S - 1 - pad == dpad, "dgrad needs pad <= kernel-1");
+  const int HOut = HS + 2 * dpad - R + 1, WOut = WS + 2 * dpad - S + 1;
+  TORCH_CHECK(HOut == H && WOut == W, "dgrad shape mismatch");
+  int M = N * HOut * WOut, K = R * S * Kout;
+  auto dx = torch::empty({N, HOut, WOut, Cin}, dy.options());
+  auto& zp = zero_page(dy.device(), dy.scalar_type());
+  dim3 grid((Cin + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
+  g16::ConvMeta cm{HS,   WS,   cl,   S,  R, 1, dpad, HOut, WOut,
+                   (int)stride, HOs, WOs};
+  if (dy.scalar_type() == torch::kBFloat16)
+    launch_nt16<bf16, g16::MODE_CONV>(dy, wr, dx, {}, false, zp, grid, M, Cin,
+                                      K, 0, 0, 0, cm);
+  else
+    launch_nt16<_Float16, g16::MODE_CONV>(dy, wr, dx, {}, false, zp, grid, M,
+                                          Cin, K, 0, 0, 0, cm);
+  return dx;
+}
+
 // C[I,J] = A[...,M,I]^T @ B[...,M,J] summed over batch? No — per batch.
 // Returns f32.  2-D inputs only here; batched variant loops z on grid.
 torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
@@ -647,7 +716,9 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   // split-M for parallelism: aim for >= 512 blocks
   int n_chunks = (M + 31) / 32;
   int tiles = ((J + 63) / 64) * ((I + 63) / 64);
-  int z = std::max(1, std::min(n_chunks, 2048 / std::max(1, tiles) + 1));
+  // ~768 total blocks: enough to fill 256 CUs ~3 deep while keeping the
+  // f32 atomic writeback fan-in (z * tile bytes) small.
+  int z = std::max(1, std::min(n_chunks, (768 + tiles - 1) / std::max(1, tiles)));
   g16::ConvMeta cm{};
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
@@ -683,7 +754,9 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   g16::ConvMeta cm{H, W, cl, (int)S, (int)R, (int)stride, (int)pad, HO, WO};
   int n_chunks = (M + 31) / 32;
   int tiles = ((Cin + 63) / 64) * ((Kout + 63) / 64);
-  int z = std::max(1, std::min(n_chunks, 2048 / std::max(1, tiles) + 1));
+  // ~768 total blocks: enough to fill 256 CUs ~3 deep while keeping the
+  // f32 atomic writeback fan-in (z * tile bytes) small.
+  int z = std::max(1, std::min(n_chunks, (768 + tiles - 1) / std::max(1, tiles)));
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
     for (int r = 0; r < (int)R; ++r) {

@@ -40,6 +40,9 @@ torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
 torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                                 int64_t stride, int64_t pad, int64_t R,
                                 int64_t S);
+torch::Tensor conv2d_dgrad_bf16(torch::Tensor dy, torch::Tensor wr,
+                                int64_t stride, int64_t pad, int64_t H,
+                                int64_t W);
 // gemm_f32.hip
 torch::Tensor bmm_nt_f32(torch::Tensor A, torch::Tensor B,
                          c10::optional<torch::Tensor> bias, bool relu);
@@ -189,9 +192,15 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
   // padding covers (H+2p-R) % stride != 0 (inputs the fwd never reached).
   int R = (int)w.size(1), S = (int)w.size(2);
   TORCH_CHECK(pad <= R - 1 && pad <= S - 1, "dgrad needs pad <= kernel-1");
+  auto wr = weight_rot(w.contiguous());  // [C, R, S, Kout]
+  int Kout = (int)dy.size(3);
+  if (is_bf16(dy) ? pow2_ge(Kout, 8) : false) {
+    // bf16/f16 fast path: zero-stuffing folded into the GEMM's im2col
+    // gather (never materialized) — gemm_bf16.hip conv2d_dgrad_bf16.
+    return conv2d_dgrad_bf16(dy.contiguous(), wr, stride, pad, H, W);
+  }
   int64_t opad_h = (H + 2 * pad - R) % stride;
   int64_t opad_w = (W + 2 * pad - S) % stride;
-  auto wr = weight_rot(w.contiguous());  // [C, R, S, Kout]
   auto dys = stride > 1 ? zero_stuff(dy.contiguous(), stride, opad_h, opad_w)
                         : dy.contiguous();
   auto dx = conv2d_fwd(dys, wr, {}, 1, R - 1 - pad, false);
