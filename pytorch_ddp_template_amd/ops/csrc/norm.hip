@@ -22,8 +22,13 @@ namespace nrm {
 constexpr int kBnMaxC = 2048;  // largest channel count (ResNet-50 layer4)
 
 template <typename T>
-__global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
-                                float* __restrict__ sumsq, long long M, int C) {
+__global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ ws,
+                                long long M, int C) {
+  // ws layout: [2 * gridDim.x][C] — per-block plain-store partials (row b =
+  // block b's sums, row gridDim.x + b its sumsqs).  No global atomics: with
+  // 2048 blocks the old per-block atomic tail hammered a C-sized array from
+  // every block and dominated small layers; the finalize pass reduces the
+  // workspace instead.
   __shared__ float ssum[kBnMaxC];
   __shared__ float ssq[kBnMaxC];
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
@@ -69,17 +74,16 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
     atomicAdd(&ssq[c0 + j], q[j]);
   }
   __syncthreads();
+  float* wsum = ws + (long long)blockIdx.x * C;
+  float* wsq = ws + ((long long)gridDim.x + blockIdx.x) * C;
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    if (ssum[c] != 0.f || ssq[c] != 0.f) {
-      atomicAdd(&sum[c], ssum[c]);
-      atomicAdd(&sumsq[c], ssq[c]);
-    }
+    wsum[c] = ssum[c];
+    wsq[c] = ssq[c];
   }
 }
 
 // ---- BN pass 2: finalize mean/rstd + update running stats ----
-__global__ void bn_finalize_kernel(const float* __restrict__ sum,
-                                   const float* __restrict__ sumsq,
+__global__ void bn_finalize_kernel(const float* __restrict__ ws, int nblocks,
                                    float* __restrict__ mean,
                                    float* __restrict__ rstd,
                                    float* __restrict__ running_mean,
@@ -88,8 +92,13 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
                                    int C) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  const float mu = sum[c] / (float)M;
-  const float var = fmaxf(sumsq[c] / (float)M - mu * mu, 0.f);
+  float s = 0.f, sq = 0.f;
+  for (int b = 0; b < nblocks; ++b) {
+    s += ws[(long long)b * C + c];
+    sq += ws[((long long)nblocks + b) * C + c];
+  }
+  const float mu = s / (float)M;
+  const float var = fmaxf(sq / (float)M - mu * mu, 0.f);
   mean[c] = mu;
   rstd[c] = rsqrtf(var + eps);
   if (running_mean) {
@@ -162,13 +171,16 @@ __global__ void bn_infer_kernel(const T* __restrict__ x, T* __restrict__ y,
 
 // ---- BN bwd pass 1: dgamma = sum dy*xhat, dbeta = sum dy ----
 // Same coalesced 8-channel-per-thread layout as bn_stats_kernel.
-template <typename T>
+// RELU: the forward fused relu(bn(x)); dy is masked by y > 0 inline (the
+// separate relu_bwd pass and its extra tensor round-trip disappear).
+// Partials go to ws ([2 * gridDim.x][C], like bn_stats) — no global atomics.
+template <typename T, bool RELU>
 __global__ void bn_bwd_stats_kernel(const T* __restrict__ dy,
                                     const T* __restrict__ x,
+                                    const T* __restrict__ y,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ rstd,
-                                    float* __restrict__ dgamma,
-                                    float* __restrict__ dbeta, long long M,
+                                    float* __restrict__ ws, long long M,
                                     int C) {
   __shared__ float sdg[kBnMaxC];
   __shared__ float sdb[kBnMaxC];
@@ -203,6 +215,13 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ dy,
         g[j] = IO::get(vg, j);
         v[j] = IO::get(vx, j);
       }
+      if (RELU) {
+        auto vy = *reinterpret_cast<const typename VecIO<T>::Vec*>(
+            y + m * C + c0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (!(IO::get(vy, j) > 0.f)) g[j] = 0.f;
+      }
     } else {
       auto ga = *reinterpret_cast<const float4v*>(pg);
       auto gb = *reinterpret_cast<const float4v*>(pg + 4);
@@ -212,6 +231,15 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ dy,
       for (int j = 0; j < 4; ++j) {
         g[j] = ga[j]; g[4 + j] = gb[j];
         v[j] = xa[j]; v[4 + j] = xb[j];
+      }
+      if (RELU) {
+        auto ya = *reinterpret_cast<const float4v*>(y + m * C + c0);
+        auto yb = *reinterpret_cast<const float4v*>(y + m * C + c0 + 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          if (!(ya[j] > 0.f)) g[j] = 0.f;
+          if (!(yb[j] > 0.f)) g[4 + j] = 0.f;
+        }
       }
     }
 #pragma unroll
@@ -226,18 +254,35 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ dy,
     atomicAdd(&sdb[c0 + j], db[j]);
   }
   __syncthreads();
+  float* wdg = ws + (long long)blockIdx.x * C;
+  float* wdb = ws + ((long long)gridDim.x + blockIdx.x) * C;
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    if (sdg[c] != 0.f || sdb[c] != 0.f) {
-      atomicAdd(&dgamma[c], sdg[c]);
-      atomicAdd(&dbeta[c], sdb[c]);
-    }
+    wdg[c] = sdg[c];
+    wdb[c] = sdb[c];
   }
 }
 
+// reduce the bwd-stats workspace into dgamma/dbeta
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ ws,
+                                       int nblocks,
+                                       float* __restrict__ dgamma,
+                                       float* __restrict__ dbeta, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float dg = 0.f, db = 0.f;
+  for (int b = 0; b < nblocks; ++b) {
+    dg += ws[(long long)b * C + c];
+    db += ws[((long long)nblocks + b) * C + c];
+  }
+  dgamma[c] = dg;
+  dbeta[c] = db;
+}
+
 // ---- BN bwd pass 2: dx (vectorized 8-channel slots) ----
-template <typename T>
+template <typename T, bool RELU>
 __global__ void bn_bwd_dx_kernel(const T* __restrict__ dy,
-                                 const T* __restrict__ x, T* __restrict__ dx,
+                                 const T* __restrict__ x,
+                                 const T* __restrict__ y, T* __restrict__ dx,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ rstd,
                                  const T* __restrict__ gamma,
@@ -265,10 +310,14 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ dy,
     if (IO::kPerLane == 8) {
       auto vg = *reinterpret_cast<const typename VecIO<T>::Vec*>(dy + m * C + c0);
       auto vx = *reinterpret_cast<const typename VecIO<T>::Vec*>(x + m * C + c0);
+      typename VecIO<T>::Vec vy{};
+      if (RELU)
+        vy = *reinterpret_cast<const typename VecIO<T>::Vec*>(y + m * C + c0);
       typename VecIO<T>::Vec o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = IO::get(vg, j);
+        if (RELU && !(IO::get(vy, j) > 0.f)) g = 0.f;
         float xh = (IO::get(vx, j) - mu[j]) * rs[j];
         IO::set(o, j, gm[j] * rs[j] * (g - invM * (dbv[j] + xh * dgv[j])));
       }
@@ -277,6 +326,7 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ dy,
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = to_f(dy[m * C + c0 + j]);
+        if (RELU && !(to_f(y[m * C + c0 + j]) > 0.f)) g = 0.f;
         float xh = (to_f(x[m * C + c0 + j]) - mu[j]) * rs[j];
         dx[m * C + c0 + j] =
             to_t<T>(gm[j] * rs[j] * (g - invM * (dbv[j] + xh * dgv[j])));
@@ -410,23 +460,21 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   long long M = x.size(0);
   int C = (int)x.size(1);
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto sum = torch::zeros({C}, f32);
-  auto sumsq = torch::zeros({C}, f32);
   auto mean = torch::empty({C}, f32);
   auto rstd = torch::empty({C}, f32);
   auto y = torch::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream();
   TORCH_CHECK(C % 8 == 0 && C <= 2048,
               "bn kernels need C % 8 == 0 and C <= 2048 (got ", C, ")");
-  dim3 sgrid(grid_1d(M * C / 8, 256));
+  dim3 sgrid(grid_1d(M * C / 8, 256, 1024));
+  auto ws = torch::empty({2 * (long long)sgrid.x, C}, f32);
   DDP_DISPATCH_FLOAT(x.scalar_type(), "bn_fwd", [&] {
     const auto* xp = reinterpret_cast<const scalar_t*>(x.data_ptr());
     hipLaunchKernelGGL((nrm::bn_stats_kernel<scalar_t>), sgrid, dim3(256), 0,
-                       stream, xp, sum.data_ptr<float>(),
-                       sumsq.data_ptr<float>(), M, C);
+                       stream, xp, ws.data_ptr<float>(), M, C);
     hipLaunchKernelGGL(
         nrm::bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
-        sum.data_ptr<float>(), sumsq.data_ptr<float>(), mean.data_ptr<float>(),
+        ws.data_ptr<float>(), (int)sgrid.x, mean.data_ptr<float>(),
         rstd.data_ptr<float>(),
         running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
         running_var.has_value() ? running_var->data_ptr<float>() : nullptr,
@@ -476,30 +524,54 @@ torch::Tensor bn_infer(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor gamma, torch::Tensor mean,
-                                  torch::Tensor rstd) {
+                                  torch::Tensor rstd,
+                                  c10::optional<torch::Tensor> y_relu) {
   long long M = x.size(0);
   int C = (int)x.size(1);
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto dgamma = torch::zeros({C}, f32);
-  auto dbeta = torch::zeros({C}, f32);
+  auto dgamma = torch::empty({C}, f32);
+  auto dbeta = torch::empty({C}, f32);
   auto dx = torch::empty_like(dy);
   auto stream = c10::hip::getCurrentHIPStream();
   TORCH_CHECK(C % 8 == 0 && C <= 2048,
               "bn kernels need C % 8 == 0 and C <= 2048 (got ", C, ")");
-  dim3 sgrid(grid_1d(M * C / 8, 256));
+  const bool relu = y_relu.has_value();
+  dim3 sgrid(grid_1d(M * C / 8, 256, 1024));
+  auto ws = torch::empty({2 * (long long)sgrid.x, C}, f32);
   DDP_DISPATCH_FLOAT(x.scalar_type(), "bn_bwd", [&] {
     const auto* dyp = reinterpret_cast<const scalar_t*>(dy.data_ptr());
     const auto* xp = reinterpret_cast<const scalar_t*>(x.data_ptr());
     const auto* gp = reinterpret_cast<const scalar_t*>(gamma.data_ptr());
-    hipLaunchKernelGGL((nrm::bn_bwd_stats_kernel<scalar_t>), sgrid, dim3(256),
-                       0, stream, dyp, xp, mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), dgamma.data_ptr<float>(),
-                       dbeta.data_ptr<float>(), M, C);
-    hipLaunchKernelGGL((nrm::bn_bwd_dx_kernel<scalar_t>),
-                       dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
-                       dyp, xp, reinterpret_cast<scalar_t*>(dx.data_ptr()),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(), gp,
-                       dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), M, C);
+    const auto* yp =
+        relu ? reinterpret_cast<const scalar_t*>(y_relu->data_ptr()) : nullptr;
+    if (relu)
+      hipLaunchKernelGGL((nrm::bn_bwd_stats_kernel<scalar_t, true>), sgrid,
+                         dim3(256), 0, stream, dyp, xp, yp,
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         ws.data_ptr<float>(), M, C);
+    else
+      hipLaunchKernelGGL((nrm::bn_bwd_stats_kernel<scalar_t, false>), sgrid,
+                         dim3(256), 0, stream, dyp, xp, yp,
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         ws.data_ptr<float>(), M, C);
+    hipLaunchKernelGGL(nrm::bn_bwd_finalize_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream, ws.data_ptr<float>(),
+                       (int)sgrid.x, dgamma.data_ptr<float>(),
+                       dbeta.data_ptr<float>(), C);
+    if (relu)
+      hipLaunchKernelGGL((nrm::bn_bwd_dx_kernel<scalar_t, true>),
+                         dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
+                         dyp, xp, yp, reinterpret_cast<scalar_t*>(dx.data_ptr()),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(), gp,
+                         dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), M,
+                         C);
+    else
+      hipLaunchKernelGGL((nrm::bn_bwd_dx_kernel<scalar_t, false>),
+                         dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
+                         dyp, xp, yp, reinterpret_cast<scalar_t*>(dx.data_ptr()),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(), gp,
+                         dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), M,
+                         C);
   });
   return {dx, dgamma, dbeta};
 }

@@ -64,7 +64,8 @@ torch::Tensor bn_infer(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
                        bool relu);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor gamma, torch::Tensor mean,
-                                  torch::Tensor rstd);
+                                  torch::Tensor rstd,
+                                  c10::optional<torch::Tensor> y_relu);
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
                                          torch::Tensor beta, double eps);
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
@@ -259,7 +260,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2d_bwd", &maxpool2d_bwd);
   m.def("bn_fwd", &bn_fwd);
   m.def("bn_infer", &bn_infer);
-  m.def("bn_bwd", &bn_bwd);
+  m.def("bn_bwd", &bn_bwd, py::arg("dy"), py::arg("x"), py::arg("gamma"),
+        py::arg("mean"), py::arg("rstd"), py::arg("y_relu") = c10::nullopt);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_fwd", &softmax_fwd);

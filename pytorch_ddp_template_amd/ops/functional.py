@@ -273,9 +273,12 @@ class _BatchNorm2dFn(torch.autograd.Function):
         xs = x.reshape(-1, C)
         if use_native(dy, gamma):
             ext = native()
-            if act == "relu":
-                dys = ext.relu_bwd(dys, y_relu.reshape(-1, C))
-            dx, dgamma, dbeta = ext.bn_bwd(dys, xs.contiguous(), gamma, mean, rstd)
+            # fused relu mask: bn_bwd applies (y_relu > 0) inline — no
+            # separate relu_bwd pass over the tensor
+            yr = y_relu.reshape(-1, C) if act == "relu" else None
+            dx, dgamma, dbeta = ext.bn_bwd(
+                dys, xs.contiguous(), gamma, mean, rstd, yr
+            )
             dgamma = dgamma.to(gamma.dtype)
             dbeta = dbeta.to(gamma.dtype)
         else:
