@@ -74,13 +74,17 @@ struct ConvMeta {
 enum { MODE_PLAIN = 0, MODE_CONV = 1 };
 
 // ---------------------------------------------------------------- NT -----
-template <typename T16, int MODE, bool RELU, bool HAS_BIAS>
+// BNT: the N tile (128, or 64 when N <= 64 — ResNet layer1 Kout, C=64
+// dgrads — so half the MFMAs aren't wasted on zero-page columns).
+template <typename T16, int MODE, bool RELU, bool HAS_BIAS, int BNT = BN>
 __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     const T16* __restrict__ A, const T16* __restrict__ B,
     T16* __restrict__ C, const T16* __restrict__ bias,
     const T16* __restrict__ zpad, int M, int N, int K, long long strideA,
     long long strideB, long long strideC, ConvMeta cm) {
-  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * TILE_BYTES];  // [buf][A|B]
+  constexpr int NI = BNT / 32;              // B fragments per wave
+  constexpr int BUF_BYTES = TILE_BYTES + BNT * 64;  // A tile + B tile
+  __shared__ __attribute__((aligned(16))) char smem[2 * BUF_BYTES];  // [buf][A|B]
 
   // ----- block swizzle (bijective XCD remap over the x*y grid) -----
   int nwg = gridDim.x * gridDim.y;
@@ -92,7 +96,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   }
   const int tx = id % gridDim.x;  // N tile
   const int ty = id / gridDim.x;  // M tile
-  const int m0 = ty * BM, n0 = tx * BN;
+  const int m0 = ty * BM, n0 = tx * BNT;
 
   const long long batch = blockIdx.z;
   A += batch * strideA;
@@ -166,28 +170,32 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
       } else {
         if (a_ok[i] && gk < K) src = A + (long long)(m0 + rl_a[i]) * K + gk;
       }
-      char* ldsA = &smem[buf * 2 * TILE_BYTES + (wave * 2 + i) * 1024];
+      char* ldsA = &smem[buf * BUF_BYTES + (wave * 2 + i) * 1024];
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)ldsA, 16, 0, 0);
-      // ---- B tile (always plain rows of [N,K]) ----
-      const int rlb = rl_a[i];
-      const int gn = n0 + rlb;
-      const int swzb = kswz(rlb, kp);
-      const int gkb = k_base + swzb * 8;
-      const T16* srcb =
-          (gn < N && gkb < K) ? B + (long long)gn * K + gkb : zpad;
-      char* ldsB =
-          &smem[buf * 2 * TILE_BYTES + TILE_BYTES + (wave * 2 + i) * 1024];
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)srcb,
-          (__attribute__((address_space(3))) unsigned int*)ldsB, 16, 0, 0);
+      // ---- B tile (always plain rows of [N,K]; only BNT rows) ----
+      // glds needs a wave-uniform LDS base (writes are lane-linear), so the
+      // guard is per 16-row chunk: chunk (wave*2+i) exists iff < BNT/16.
+      if ((wave * 2 + i) * 16 < BNT) {
+        const int rlb = rl_a[i];
+        const int gn = n0 + rlb;
+        const int swzb = kswz(rlb, kp);
+        const int gkb = k_base + swzb * 8;
+        const T16* srcb =
+            (gn < N && gkb < K) ? B + (long long)gn * K + gkb : zpad;
+        char* ldsB =
+            &smem[buf * BUF_BYTES + TILE_BYTES + (wave * 2 + i) * 1024];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)srcb,
+            (__attribute__((address_space(3))) unsigned int*)ldsB, 16, 0, 0);
+      }
     }
   };
 
   // ----- main loop -----
-  f32x4 acc[4][4] = {};
-  const int wm = (wave >> 1) * 64, wn = (wave & 1) * 64;
+  f32x4 acc[4][NI] = {};
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BNT / 2);
   const int fr = lane & 15;   // fragment row (A) / col (B, D)
   const int fs = lane >> 4;   // k-slot
 
@@ -196,15 +204,18 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   int buf = 0;
   for (int kt = 0; kt < KT; ++kt) {
     if (kt + 1 < KT) stage(buf ^ 1, kt + 1);
-    const char* baseA = &smem[buf * 2 * TILE_BYTES];
+    const char* baseA = &smem[buf * BUF_BYTES];
     const char* baseB = baseA + TILE_BYTES;
     using vec16 = typename M16<T16>::vec;
-    vec16 af[4], bf[4];
+    vec16 af[4], bf[NI];
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int ra = wm + i * 16 + fr;
       af[i] = *reinterpret_cast<const vec16*>(
           baseA + ra * 64 + kswz(ra, fs) * 16);
+    }
+#pragma unroll
+    for (int i = 0; i < NI; ++i) {
       const int rb = wn + i * 16 + fr;
       bf[i] = *reinterpret_cast<const vec16*>(
           baseB + rb * 64 + kswz(rb, fs) * 16);
@@ -212,7 +223,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
+      for (int ni = 0; ni < NI; ++ni)
         acc[mi][ni] = M16<T16>::mma(af[mi], bf[ni], acc[mi][ni]);
     __syncthreads();
     buf ^= 1;
@@ -220,7 +231,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
 
   // ----- epilogue: bias + relu + bf16 store -----
 #pragma unroll
-  for (int ni = 0; ni < 4; ++ni) {
+  for (int ni = 0; ni < NI; ++ni) {
     const int col = n0 + wn + ni * 16 + fr;
     if (col >= N) continue;
     float bv = HAS_BIAS ? to_f(bias[col]) : 0.f;
@@ -396,145 +407,168 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
   }
 }
 
-// ------------------------------------------------------------ TN conv ----
-// Conv wgrad with the S kernel columns of one kernel row fused into a single
-// launch, software-pipelined: per 32-m chunk the dy tile and all NS shifted
-// x tiles are transpose-staged into one of two LDS buffers, the next chunk's
-// global loads are issued before the MFMAs of the current chunk (their
-// latency hides under the matrix work), and there is exactly ONE barrier per
-// chunk.  The x windows of adjacent s overlap so their re-reads hit L1.
-// Host keeps total blocks ~768 so the final f32 atomic fan-in stays small.
-template <typename T16, int NS>
-__global__ __launch_bounds__(THREADS) void gemm_tn_conv_kernel(
-    const T16* __restrict__ A /*dy*/, const T16* __restrict__ B /*x*/,
-    float* __restrict__ C, int Mtot, int I /*Kout*/, int r, long long ldc,
-    long long coff_base /*offset of (r, s=0) slice*/, ConvMeta cm) {
+// ------------------------------------------------- TN conv wgrad (tr) ----
+// dw accumulation, all R*S taps fused in ONE launch, built on two gfx950
+// hardware paths instead of VALU shuffle-transposes:
+//   * global_load_lds stages each operand tile m-major: one [32 m][16 ch]
+//     image (1 KiB) per 16-channel group, written lane-linearly (lane l
+//     covers m = l>>1, 8-ch half l&1) — zero VALU, zero register traffic;
+//   * ds_read_b64_tr_b16 reads MFMA fragments from those images with a free
+//     hardware transpose.  Per-lane address
+//       img + (fs*8 + ((lane&15)>>2))*32B + (lane&3)*8B     (fs = lane>>4)
+//     and a second read at +128B deliver img[fs*8 .. fs*8+7][lane&15] in
+//     natural k order (semantics established empirically by
+//     tools/tr_probe.hip: out[4a+d][j] = elem d at the 8B-aligned address
+//     of lane 4j+a within each 16-lane group).
+// Per 32-m chunk a workgroup stages 1+TAPS tiles (4*(1+TAPS) glds) and runs
+// 4*TAPS MFMAs per thread between barriers; the old shuffle kernel ran 12
+// MFMAs per ~100 staging instructions and was issue-bound (SQ_WAIT_INST_ANY
+// 49%).  Output: f32 atomicAdd, host keeps total blocks small.
+template <typename T16, int TAPS>
+__global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
+    const T16* __restrict__ dy, const T16* __restrict__ x,
+    float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
+    int I /*Kout*/, long long ldc, ConvMeta cm) {
   constexpr int BI = 64, BJ = 64, BMC = 32;
-  constexpr int ROW = BMC + 8;
-  constexpr int TILE = BI * ROW;  // elements per staged operand tile
+  constexpr int IMG = 32 * 16;   // elements per [32 m][16 ch] image
+  constexpr int TILE = 4 * IMG;  // 4 images = one 64-channel tile
+  constexpr int NOPS = 1 + TAPS;
+  typedef short v4s __attribute__((ext_vector_type(4)));
   using vec16 = typename M16<T16>::vec;
-  // [2 pipeline buffers][A | B shift 0..NS-1]
-  __shared__ __attribute__((aligned(16))) T16 lds[2 * (1 + NS) * TILE];
+  __shared__ __attribute__((aligned(16))) T16 lds[2 * NOPS * TILE];
 
   const int Cin = 1 << cm.C_log2;
   const int i0 = blockIdx.y * BI;
-  const int j0 = blockIdx.x * BJ;  // channel tile
+  const int j0 = blockIdx.x * BJ;
   const int n_chunks = (Mtot + BMC - 1) / BMC;
   const int per_z = (n_chunks + gridDim.z - 1) / gridDim.z;
   const int ch0 = blockIdx.z * per_z;
   const int ch1 = min(n_chunks, ch0 + per_z);
-  if (ch0 >= ch1) return;  // uniform per block: no barrier divergence
+  if (ch0 >= ch1) return;  // uniform per block
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
+  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;  // wave tile 32x32
   const int fr = lane & 15;
   const int fs = lane >> 4;
-  const int sm = threadIdx.x >> 3;        // m row in chunk (0..31)
-  const int sc0 = (threadIdx.x & 7) * 8;  // first col of this 16B piece
+  const int sm = lane >> 1;        // staged m row this lane covers
+  const int sh8 = (lane & 1) * 8;  // 8-channel half within the 32-B row
 
-  f32x4 acc[2][2][NS] = {};
-  vec16 va, vb[NS];  // in-flight global data for the chunk being staged
+  f32x4 acc[2][2][TAPS] = {};
 
-  // global fetch of chunk ch into registers (dy tile + NS shifted x tiles)
-  auto fetch = [&](int ch) {
+  // ---- stage chunk ch into LDS buffer buf: 4*NOPS glds, no VALU pack ----
+  auto stage = [&](int buf, int ch) {
     const int gm = ch * BMC + sm;
-    va = vec16{};
-#pragma unroll
-    for (int si = 0; si < NS; ++si) vb[si] = vec16{};
-    if (gm < Mtot) {
-      if (i0 + sc0 < I) {
-        const long long off = (long long)gm * I + i0 + sc0;
-        if (off + 8 <= (long long)Mtot * I) {
-          va = *reinterpret_cast<const vec16*>(A + off);
-        } else {  // last-row partial chunk: element-wise guarded load
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (off + j < (long long)Mtot * I) va[j] = A[off + j];
-        }
-      }
+    const bool mok = gm < Mtot;
+    int n = 0, hb = 0, wb = 0;
+    if (mok) {
       int t = gm;
       const int wo = t % cm.WO;
       t /= cm.WO;
       const int ho = t % cm.HO;
-      const int n = t / cm.HO;
-      const int hi = ho * cm.stride - cm.pad + r;
-      const int wb = wo * cm.stride - cm.pad;
-      if (hi >= 0 && hi < cm.H && j0 + sc0 < Cin) {
-        const long long rowoff =
-            ((long long)n * cm.H + hi) * cm.W * Cin + j0 + sc0;
-#pragma unroll
-        for (int si = 0; si < NS; ++si) {
-          const int wi = wb + si;
-          if (wi >= 0 && wi < cm.W)
-            vb[si] = *reinterpret_cast<const vec16*>(
-                B + rowoff + (long long)wi * Cin);
-        }
+      n = t / cm.HO;
+      hb = ho * cm.stride - cm.pad;
+      wb = wo * cm.stride - cm.pad;
+    }
+    T16* base = lds + buf * NOPS * TILE;
+    // wave w issues units u = w, w+4, ... of the 4*NOPS image stages
+    for (int u = wave; u < 4 * NOPS; u += 4) {
+      const int tile = u >> 2;  // 0 = dy, 1+tap = x
+      const int ig = u & 3;     // 16-channel group
+      const T16* src = zpad;
+      if (tile == 0) {
+        const int ii = i0 + ig * 16 + sh8;
+        if (mok && ii < I) src = dy + (long long)gm * I + ii;
+      } else {
+        const int tap = tile - 1;
+        const int r = TAPS == 1 ? 0 : tap / cm.S;
+        const int s = TAPS == 1 ? 0 : tap % cm.S;
+        const int hi = hb + r, wi = wb + s;
+        const int jj = j0 + ig * 16 + sh8;
+        if (mok && hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W && jj < Cin)
+          src = x + (((long long)n * cm.H + hi) * cm.W + wi) * Cin + jj;
       }
+      T16* dst = base + tile * TILE + ig * IMG;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
     }
   };
 
-  // transpose-pack registers -> LDS buffer `buf` (lane pairs 8 apart pack
-  // two adjacent m rows into one b32; image layout [col][m])
-  auto stage = [&](int buf) {
-    T16* base = lds + buf * (1 + NS) * TILE;
-    const bool writer = ((threadIdx.x >> 3) & 1) == 0;
-#pragma unroll
-    for (int op = 0; op < 1 + NS; ++op) {
-      vec16 v = (op == 0) ? va : vb[op - 1];
-      short8 mine = *reinterpret_cast<short8*>(&v);
-      short8 other;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) other[j] = __shfl_xor((int)mine[j], 8);
-      if (writer) {
-        T16* dst = base + op * TILE;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          unsigned int pack = ((unsigned short)mine[j]) |
-                              (((unsigned int)(unsigned short)other[j]) << 16);
-          *reinterpret_cast<unsigned int*>(&dst[(sc0 + j) * ROW + sm]) = pack;
-        }
-      }
-    }
-  };
+  // per-lane byte offset of this lane's tr-read address within an image
+  const unsigned tr_lane_off =
+      (unsigned)((fs * 8 + ((lane & 15) >> 2)) * 32 + (lane & 3) * 8);
+#define LDS_BYTE(p)                                           \
+  ((unsigned)(unsigned long long)(__attribute__((            \
+      address_space(3))) const T16*)(p))
 
-  fetch(ch0);
-  stage(0);
+  stage(0, ch0);
   __syncthreads();
 
   for (int ch = ch0; ch < ch1; ++ch) {
     const int buf = (ch - ch0) & 1;
     const bool more = ch + 1 < ch1;
-    if (more) fetch(ch + 1);  // latency hides under the MFMAs below
-    const T16* base = lds + buf * (1 + NS) * TILE;
+    if (more) stage(buf ^ 1, ch + 1);  // glds latency hides under the MFMAs
+    const T16* base = lds + buf * NOPS * TILE;
+
+    // A fragments (dy): images wm/16 + mi
     vec16 af[2];
+    {
+      const unsigned a0 =
+          LDS_BYTE(base + ((wm >> 4) + 0) * IMG) + tr_lane_off;
+      const unsigned a1 =
+          LDS_BYTE(base + ((wm >> 4) + 1) * IMG) + tr_lane_off;
+      v4s l0, h0, l1, h1;
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
+          "ds_read_b64_tr_b16 %2, %5 offset:0\n\t"
+          "ds_read_b64_tr_b16 %3, %5 offset:128\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=v"(l0), "=v"(h0), "=v"(l1), "=v"(h1)
+          : "v"(a0), "v"(a1));
+      reinterpret_cast<v4s*>(&af[0])[0] = l0;
+      reinterpret_cast<v4s*>(&af[0])[1] = h0;
+      reinterpret_cast<v4s*>(&af[1])[0] = l1;
+      reinterpret_cast<v4s*>(&af[1])[1] = h1;
+    }
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
-      af[i] = *reinterpret_cast<const vec16*>(
-          &base[(wm + i * 16 + fr) * ROW + fs * 8]);
-#pragma unroll
-    for (int si = 0; si < NS; ++si) {
-      const T16* bb = base + (1 + si) * TILE;
+    for (int tap = 0; tap < TAPS; ++tap) {
+      const T16* tb = base + (1 + tap) * TILE;
+      const unsigned b0 =
+          LDS_BYTE(tb + ((wn >> 4) + 0) * IMG) + tr_lane_off;
+      const unsigned b1 =
+          LDS_BYTE(tb + ((wn >> 4) + 1) * IMG) + tr_lane_off;
       vec16 bfr[2];
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-        bfr[i] = *reinterpret_cast<const vec16*>(
-            &bb[(wn + i * 16 + fr) * ROW + fs * 8]);
+      v4s l0, h0, l1, h1;
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
+          "ds_read_b64_tr_b16 %2, %5 offset:0\n\t"
+          "ds_read_b64_tr_b16 %3, %5 offset:128\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=v"(l0), "=v"(h0), "=v"(l1), "=v"(h1)
+          : "v"(b0), "v"(b1));
+      reinterpret_cast<v4s*>(&bfr[0])[0] = l0;
+      reinterpret_cast<v4s*>(&bfr[0])[1] = h0;
+      reinterpret_cast<v4s*>(&bfr[1])[0] = l1;
+      reinterpret_cast<v4s*>(&bfr[1])[1] = h1;
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 2; ++ni)
-          acc[mi][ni][si] = M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni][si]);
+          acc[mi][ni][tap] = M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni][tap]);
     }
-    if (more) {
-      stage(buf ^ 1);
-      __syncthreads();  // the ONE barrier per chunk
-    }
+    if (more) __syncthreads();  // drains the glds (vmcnt) + publishes buf^1
   }
+#undef LDS_BYTE
 
+  // ---- writeback: f32 atomics into dw[I][taps][Cin] slices ----
+  // (static tap order: a runtime-rotated index into acc would demote the
+  // accumulators to scratch)
 #pragma unroll
-  for (int si = 0; si < NS; ++si) {
-    const long long coff = coff_base + (long long)si * Cin;
+  for (int tap = 0; tap < TAPS; ++tap) {
+    const long long coff = (long long)tap * Cin;
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
       const int col = j0 + wn + ni * 16 + fr;
@@ -545,8 +579,8 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_conv_kernel(
         for (int rr = 0; rr < 4; ++rr) {
           const int row = i0 + wm + mi * 16 + fs * 4 + rr;
           if (row >= I) continue;
-          atomicAdd(&C[coff + (long long)row * ldc + col],
-                    acc[mi][ni][si][rr]);
+          atomicAdd(&dw[(long long)row * ldc + coff + col],
+                    acc[mi][ni][tap][rr]);
         }
     }
   }
@@ -580,14 +614,29 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
   const t16* bias_p =
       bias.has_value() ? reinterpret_cast<const t16*>(bias->data_ptr())
                        : nullptr;
-#define LAUNCH_NT16(RELU, HB)                                                \
-  hipLaunchKernelGGL((g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB>), grid,  \
-                     dim3(g16::THREADS), 0, stream,                          \
-                     reinterpret_cast<const t16*>(A.data_ptr()),             \
-                     reinterpret_cast<const t16*>(B.data_ptr()),             \
-                     reinterpret_cast<t16*>(C.data_ptr()), bias_p,           \
-                     reinterpret_cast<const t16*>(zp.data_ptr()), M, N, K,   \
-                     sA, sB, sC, cm)
+  // narrow-N variant: BNT=64 halves the B tile so no MFMA computes
+  // zero-page columns (ResNet layer1 Kout=64 fwd, C=64 dgrads).
+  const bool narrow = N <= 64;
+  if (narrow) grid.x = (N + 63) / 64;
+#define LAUNCH_NT16(RELU, HB)                                                 \
+  do {                                                                        \
+    if (narrow)                                                               \
+      hipLaunchKernelGGL((g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB, 64>), \
+                         grid, dim3(g16::THREADS), 0, stream,                 \
+                         reinterpret_cast<const t16*>(A.data_ptr()),          \
+                         reinterpret_cast<const t16*>(B.data_ptr()),          \
+                         reinterpret_cast<t16*>(C.data_ptr()), bias_p,        \
+                         reinterpret_cast<const t16*>(zp.data_ptr()), M, N,   \
+                         K, sA, sB, sC, cm);                                  \
+    else                                                                      \
+      hipLaunchKernelGGL((g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB>),     \
+                         grid, dim3(g16::THREADS), 0, stream,                 \
+                         reinterpret_cast<const t16*>(A.data_ptr()),          \
+                         reinterpret_cast<const t16*>(B.data_ptr()),          \
+                         reinterpret_cast<t16*>(C.data_ptr()), bias_p,        \
+                         reinterpret_cast<const t16*>(zp.data_ptr()), M, N,   \
+                         K, sA, sB, sC, cm);                                  \
+  } while (0)
   if (relu) {
     if (bias_p) LAUNCH_NT16(true, true);
     else LAUNCH_NT16(true, false);
@@ -750,34 +799,36 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   int M = N * HO * WO;
   auto dw = torch::zeros({(long long)Kout, R, S, (long long)Cin},
                          x.options().dtype(torch::kFloat32));
+  auto& zp = zero_page(x.device(), x.scalar_type());
   auto stream = c10::hip::getCurrentHIPStream();
   g16::ConvMeta cm{H, W, cl, (int)S, (int)R, (int)stride, (int)pad, HO, WO};
   int n_chunks = (M + 31) / 32;
   int tiles = ((Cin + 63) / 64) * ((Kout + 63) / 64);
-  // ~768 total blocks: enough to fill 256 CUs ~3 deep while keeping the
-  // f32 atomic writeback fan-in (z * tile bytes) small.
-  int z = std::max(1, std::min(n_chunks, (768 + tiles - 1) / std::max(1, tiles)));
+  // ~512 total blocks: fills 256 CUs 2 deep (the tr kernel runs 2 WGs/CU at
+  // its LDS size) while keeping the f32 atomic fan-in small.
+  int z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
+  const long long ldc = (long long)R * S * Cin;
+  dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
-    for (int r = 0; r < (int)R; ++r) {
-      dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
-      // one launch per kernel ROW: the S column shifts are fused in-kernel
-      // (dy staged once, x windows overlap in cache).
-      const long long ldc = (long long)R * S * Cin;
-      const long long coff = (long long)r * (int)S * Cin;
-      if (S == 3)
-        hipLaunchKernelGGL((g16::gemm_tn_conv_kernel<t16, 3>), grid,
-                           dim3(g16::THREADS), 0, stream,
-                           reinterpret_cast<const t16*>(dy.data_ptr()),
-                           reinterpret_cast<const t16*>(x.data_ptr()),
-                           dw.data_ptr<float>(), M, Kout, r, ldc, coff, cm);
-      else if (S == 1)
-        hipLaunchKernelGGL((g16::gemm_tn_conv_kernel<t16, 1>), grid,
-                           dim3(g16::THREADS), 0, stream,
-                           reinterpret_cast<const t16*>(dy.data_ptr()),
-                           reinterpret_cast<const t16*>(x.data_ptr()),
-                           dw.data_ptr<float>(), M, Kout, r, ldc, coff, cm);
-      else
+    if (R == 3 && S == 3) {
+      hipLaunchKernelGGL((g16::gemm_wgrad_tr_kernel<t16, 9>), grid,
+                         dim3(g16::THREADS), 0, stream,
+                         reinterpret_cast<const t16*>(dy.data_ptr()),
+                         reinterpret_cast<const t16*>(x.data_ptr()),
+                         dw.data_ptr<float>(),
+                         reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
+                         ldc, cm);
+    } else if (R == 1 && S == 1) {
+      hipLaunchKernelGGL((g16::gemm_wgrad_tr_kernel<t16, 1>), grid,
+                         dim3(g16::THREADS), 0, stream,
+                         reinterpret_cast<const t16*>(dy.data_ptr()),
+                         reinterpret_cast<const t16*>(x.data_ptr()),
+                         dw.data_ptr<float>(),
+                         reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
+                         ldc, cm);
+    } else {  // generic R x S: per-tap TN launches (cold path)
+      for (int r = 0; r < (int)R; ++r)
         for (int s2 = 0; s2 < (int)S; ++s2)
           hipLaunchKernelGGL(
               (g16::gemm_tn_bf16_kernel<t16, g16::MODE_CONV>), grid,
@@ -785,7 +836,7 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
               reinterpret_cast<const t16*>(dy.data_ptr()),
               reinterpret_cast<const t16*>(x.data_ptr()),
               dw.data_ptr<float>(), M, Kout, Cin, r, s2, ldc,
-              coff + (long long)s2 * Cin, cm);
+              ((long long)r * S + s2) * Cin, cm);
     }
   };
   if (x.scalar_type() == torch::kBFloat16) run(bf16{});

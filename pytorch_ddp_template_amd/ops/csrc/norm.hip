@@ -83,6 +83,9 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ ws,
 }
 
 // ---- BN pass 2: finalize mean/rstd + update running stats ----
+// one block per channel; 256 threads stripe the workspace rows then tree-
+// reduce in LDS (a serial per-thread loop over ~1024 partials sat on the
+// critical path between the stats and norm kernels).
 __global__ void bn_finalize_kernel(const float* __restrict__ ws, int nblocks,
                                    float* __restrict__ mean,
                                    float* __restrict__ rstd,
@@ -90,13 +93,26 @@ __global__ void bn_finalize_kernel(const float* __restrict__ ws, int nblocks,
                                    float* __restrict__ running_var,
                                    float momentum, float eps, long long M,
                                    int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float r1[256], r2[256];
+  const int c = blockIdx.x;
   float s = 0.f, sq = 0.f;
-  for (int b = 0; b < nblocks; ++b) {
+  for (int b = threadIdx.x; b < nblocks; b += blockDim.x) {
     s += ws[(long long)b * C + c];
     sq += ws[((long long)nblocks + b) * C + c];
   }
+  r1[threadIdx.x] = s;
+  r2[threadIdx.x] = sq;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) {
+      r1[threadIdx.x] += r1[threadIdx.x + w];
+      r2[threadIdx.x] += r2[threadIdx.x + w];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x != 0) return;
+  s = r1[0];
+  sq = r2[0];
   const float mu = s / (float)M;
   const float var = fmaxf(sq / (float)M - mu * mu, 0.f);
   mean[c] = mu;
@@ -262,20 +278,40 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ dy,
   }
 }
 
-// reduce the bwd-stats workspace into dgamma/dbeta
+// reduce the bwd-stats workspace into dgamma/dbeta (block-per-channel tree
+// reduce, like bn_finalize_kernel); also emits the grads in the param dtype
+// so no separate f32->bf16 cast kernel runs per BN layer.
+template <typename T>
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ ws,
                                        int nblocks,
                                        float* __restrict__ dgamma,
-                                       float* __restrict__ dbeta, int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+                                       float* __restrict__ dbeta,
+                                       T* __restrict__ dgamma16,
+                                       T* __restrict__ dbeta16, int C) {
+  __shared__ float r1[256], r2[256];
+  const int c = blockIdx.x;
   float dg = 0.f, db = 0.f;
-  for (int b = 0; b < nblocks; ++b) {
+  for (int b = threadIdx.x; b < nblocks; b += blockDim.x) {
     dg += ws[(long long)b * C + c];
     db += ws[((long long)nblocks + b) * C + c];
   }
-  dgamma[c] = dg;
-  dbeta[c] = db;
+  r1[threadIdx.x] = dg;
+  r2[threadIdx.x] = db;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) {
+      r1[threadIdx.x] += r1[threadIdx.x + w];
+      r2[threadIdx.x] += r2[threadIdx.x + w];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x != 0) return;
+  dgamma[c] = r1[0];
+  dbeta[c] = r2[0];
+  if (dgamma16) {
+    dgamma16[c] = to_t<T>(r1[0]);
+    dbeta16[c] = to_t<T>(r2[0]);
+  }
 }
 
 // ---- BN bwd pass 2: dx (vectorized 8-channel slots) ----
@@ -473,7 +509,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     hipLaunchKernelGGL((nrm::bn_stats_kernel<scalar_t>), sgrid, dim3(256), 0,
                        stream, xp, ws.data_ptr<float>(), M, C);
     hipLaunchKernelGGL(
-        nrm::bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
+        nrm::bn_finalize_kernel, dim3(C), dim3(256), 0, stream,
         ws.data_ptr<float>(), (int)sgrid.x, mean.data_ptr<float>(),
         rstd.data_ptr<float>(),
         running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
@@ -538,6 +574,12 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   const bool relu = y_relu.has_value();
   dim3 sgrid(grid_1d(M * C / 8, 256, 1024));
   auto ws = torch::empty({2 * (long long)sgrid.x, C}, f32);
+  // grads returned in the param dtype (cast fused into the finalize)
+  const bool wide = gamma.scalar_type() == torch::kFloat32;
+  auto dgamma_out =
+      wide ? dgamma : torch::empty({C}, x.options().dtype(gamma.scalar_type()));
+  auto dbeta_out =
+      wide ? dbeta : torch::empty({C}, x.options().dtype(gamma.scalar_type()));
   DDP_DISPATCH_FLOAT(x.scalar_type(), "bn_bwd", [&] {
     const auto* dyp = reinterpret_cast<const scalar_t*>(dy.data_ptr());
     const auto* xp = reinterpret_cast<const scalar_t*>(x.data_ptr());
@@ -554,10 +596,15 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                          dim3(256), 0, stream, dyp, xp, yp,
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
                          ws.data_ptr<float>(), M, C);
-    hipLaunchKernelGGL(nrm::bn_bwd_finalize_kernel, dim3((C + 255) / 256),
+    hipLaunchKernelGGL((nrm::bn_bwd_finalize_kernel<scalar_t>), dim3(C),
                        dim3(256), 0, stream, ws.data_ptr<float>(),
                        (int)sgrid.x, dgamma.data_ptr<float>(),
-                       dbeta.data_ptr<float>(), C);
+                       dbeta.data_ptr<float>(),
+                       wide ? nullptr
+                            : reinterpret_cast<scalar_t*>(dgamma_out.data_ptr()),
+                       wide ? nullptr
+                            : reinterpret_cast<scalar_t*>(dbeta_out.data_ptr()),
+                       C);
     if (relu)
       hipLaunchKernelGGL((nrm::bn_bwd_dx_kernel<scalar_t, true>),
                          dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
@@ -573,7 +620,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                          dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), M,
                          C);
   });
-  return {dx, dgamma, dbeta};
+  return {dx, dgamma_out, dbeta_out};
 }
 
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
