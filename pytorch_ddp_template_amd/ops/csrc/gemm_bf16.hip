@@ -424,11 +424,14 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 // 4*TAPS MFMAs per thread between barriers; the old shuffle kernel ran 12
 // MFMAs per ~100 staging instructions and was issue-bound (SQ_WAIT_INST_ANY
 // 49%).  Output: f32 atomicAdd, host keeps total blocks small.
-template <typename T16, int TAPS>
+// MODE_PLAIN reuses the same machinery for the plain TN GEMM
+// C[I,J] += sum_m A[m,I] B[m,J] (linear / ViT weight grads): the B gather is
+// then just rows of [M,J] and TAPS must be 1.  J == Cin for conv mode.
+template <typename T16, int TAPS, int MODE = MODE_CONV>
 __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
     float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
-    int I /*Kout*/, long long ldc, ConvMeta cm) {
+    int I /*Kout*/, int J, long long ldc, ConvMeta cm) {
   constexpr int BI = 64, BJ = 64, BMC = 32;
   constexpr int IMG = 32 * 16;   // elements per [32 m][16 ch] image
   constexpr int TILE = 4 * IMG;  // 4 images = one 64-channel tile
@@ -437,7 +440,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   using vec16 = typename M16<T16>::vec;
   __shared__ __attribute__((aligned(16))) T16 lds[2 * NOPS * TILE];
 
-  const int Cin = 1 << cm.C_log2;
+  const int Cin = J;
   const int i0 = blockIdx.y * BI;
   const int j0 = blockIdx.x * BJ;
   const int n_chunks = (Mtot + BMC - 1) / BMC;
@@ -461,7 +464,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
     const int gm = ch * BMC + sm;
     const bool mok = gm < Mtot;
     int n = 0, hb = 0, wb = 0;
-    if (mok) {
+    if (MODE == MODE_CONV && mok) {
       int t = gm;
       const int wo = t % cm.WO;
       t /= cm.WO;
@@ -479,6 +482,9 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
       if (tile == 0) {
         const int ii = i0 + ig * 16 + sh8;
         if (mok && ii < I) src = dy + (long long)gm * I + ii;
+      } else if (MODE == MODE_PLAIN) {
+        const int jj = j0 + ig * 16 + sh8;
+        if (mok && jj < J) src = x + (long long)gm * J + jj;
       } else {
         const int tap = tile - 1;
         const int r = TAPS == 1 ? 0 : tap / cm.S;
@@ -769,17 +775,20 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   // f32 atomic writeback fan-in (z * tile bytes) small.
   int z = std::max(1, std::min(n_chunks, (768 + tiles - 1) / std::max(1, tiles)));
   g16::ConvMeta cm{};
+  auto& zp = zero_page(A.device(), A.scalar_type());
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
     for (long long b = 0; b < bsz; ++b) {
       dim3 grid((J + 63) / 64, (I + 63) / 64, z);
+      // glds + tr-read TN path (same kernel as conv wgrad, plain B gather)
       hipLaunchKernelGGL(
-          (g16::gemm_tn_bf16_kernel<t16, g16::MODE_PLAIN>), grid,
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN>), grid,
           dim3(g16::THREADS), 0, stream,
           reinterpret_cast<const t16*>(A.data_ptr()) + b * (long long)M * I,
           reinterpret_cast<const t16*>(B.data_ptr()) + b * (long long)M * J,
-          C.data_ptr<float>() + b * (long long)I * J, M, I, J, 0, 0,
-          /*ldc=*/J, /*coff=*/0, cm);
+          C.data_ptr<float>() + b * (long long)I * J,
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J,
+          /*ldc=*/J, cm);
     }
   };
   if (A.scalar_type() == torch::kBFloat16) run(bf16{});
@@ -818,7 +827,7 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                          reinterpret_cast<const t16*>(x.data_ptr()),
                          dw.data_ptr<float>(),
                          reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
-                         ldc, cm);
+                         Cin, ldc, cm);
     } else if (R == 1 && S == 1) {
       hipLaunchKernelGGL((g16::gemm_wgrad_tr_kernel<t16, 1>), grid,
                          dim3(g16::THREADS), 0, stream,
@@ -826,7 +835,7 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                          reinterpret_cast<const t16*>(x.data_ptr()),
                          dw.data_ptr<float>(),
                          reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
-                         ldc, cm);
+                         Cin, ldc, cm);
     } else {  // generic R x S: per-tap TN launches (cold path)
       for (int r = 0; r < (int)R; ++r)
         for (int s2 = 0; s2 < (int)S; ++s2)

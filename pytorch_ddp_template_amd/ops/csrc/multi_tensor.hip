@@ -18,7 +18,9 @@
 
 namespace mt {
 
-constexpr int kChunk = 1 << 16;  // elements per block chunk
+constexpr int kChunk = 1 << 13;  // elements per block chunk (8K: ResNet-18's
+                                 // 11.7M params -> ~1.4K blocks, vs 64K chunks
+                                 // which left the GPU <1 block/CU)
 constexpr int kMaxTensors = 512;
 
 struct ChunkDesc {
@@ -35,6 +37,16 @@ struct PtrTable {
 };
 
 // ---- fused SGD (momentum / weight decay / dampening / nesterov) ----
+// 4-wide aligned vector views (chunk offsets are kChunk-aligned and torch
+// allocations are 256-B aligned, so only the final partial quad is scalar)
+template <typename T>
+struct alignas(8) Q4 {
+  T v[4];
+};
+struct alignas(16) F4 {
+  float v[4];
+};
+
 template <typename T>
 __global__ void sgd_kernel(const ChunkDesc* __restrict__ chunks,
                            const PtrTable* __restrict__ tab, float lr,
@@ -48,7 +60,34 @@ __global__ void sgd_kernel(const ChunkDesc* __restrict__ chunks,
   T* p = reinterpret_cast<T*>(tab->param[d.tensor]);
   float* m = tab->mom[d.tensor];
   float* mw = tab->master[d.tensor];
-  for (long long i = base + threadIdx.x; i < end; i += blockDim.x) {
+  const long long vend = base + ((end - base) & ~3LL);
+  for (long long i = base + (long long)threadIdx.x * 4; i < vend;
+       i += (long long)blockDim.x * 4) {
+    Q4<T> g4 = *reinterpret_cast<const Q4<T>*>(g + i);
+    F4 w4, m4;
+    if (mw) w4 = *reinterpret_cast<const F4*>(mw + i);
+    Q4<T> p4;
+    if (!mw) p4 = *reinterpret_cast<const Q4<T>*>(p + i);
+    if (use_mom) m4 = *reinterpret_cast<const F4*>(m + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gf = to_f(g4.v[j]);
+      float w = mw ? w4.v[j] : to_f(p4.v[j]);
+      if (wd != 0.f) gf += wd * w;
+      if (use_mom) {
+        float mv = m4.v[j] * momentum + (1.f - damp) * gf;
+        m4.v[j] = mv;
+        gf = nesterov ? gf + momentum * mv : mv;
+      }
+      w -= lr * gf;
+      w4.v[j] = w;
+      p4.v[j] = to_t<T>(w);
+    }
+    if (use_mom) *reinterpret_cast<F4*>(m + i) = m4;
+    if (mw) *reinterpret_cast<F4*>(mw + i) = w4;
+    *reinterpret_cast<Q4<T>*>(p + i) = p4;
+  }
+  for (long long i = vend + threadIdx.x; i < end; i += blockDim.x) {
     float gf = to_f(g[i]);
     float w = mw ? mw[i] : to_f(p[i]);
     if (wd != 0.f) gf += wd * w;
@@ -75,7 +114,17 @@ __global__ void l2norm_sq_kernel(const ChunkDesc* __restrict__ chunks,
   const long long end = min(n, base + (long long)kChunk);
   const T* g = reinterpret_cast<const T*>(tab->grad[d.tensor]);
   float acc = 0.f;
-  for (long long i = base + threadIdx.x; i < end; i += blockDim.x) {
+  const long long vend = base + ((end - base) & ~3LL);
+  for (long long i = base + (long long)threadIdx.x * 4; i < vend;
+       i += (long long)blockDim.x * 4) {
+    Q4<T> g4 = *reinterpret_cast<const Q4<T>*>(g + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float v = to_f(g4.v[j]);
+      acc += v * v;
+    }
+  }
+  for (long long i = vend + threadIdx.x; i < end; i += blockDim.x) {
     float v = to_f(g[i]);
     acc += v * v;
   }
@@ -96,7 +145,15 @@ __global__ void scale_kernel(const ChunkDesc* __restrict__ chunks,
   T* p = reinterpret_cast<T*>(tab->param[d.tensor]);
   float s = dev_scale ? *dev_scale : scale;
   if (clamp_to_one) s = fminf(s, 1.f);
-  for (long long i = base + threadIdx.x; i < end; i += blockDim.x)
+  const long long vend = base + ((end - base) & ~3LL);
+  for (long long i = base + (long long)threadIdx.x * 4; i < vend;
+       i += (long long)blockDim.x * 4) {
+    Q4<T> p4 = *reinterpret_cast<const Q4<T>*>(p + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) p4.v[j] = to_t<T>(to_f(p4.v[j]) * s);
+    *reinterpret_cast<Q4<T>*>(p + i) = p4;
+  }
+  for (long long i = vend + threadIdx.x; i < end; i += blockDim.x)
     p[i] = to_t<T>(to_f(p[i]) * s);
 }
 
