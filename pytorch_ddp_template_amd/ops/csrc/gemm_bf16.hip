@@ -84,7 +84,11 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     long long strideB, long long strideC, ConvMeta cm) {
   constexpr int NI = BNT / 32;              // B fragments per wave
   constexpr int BUF_BYTES = TILE_BYTES + BNT * 64;  // A tile + B tile
-  __shared__ __attribute__((aligned(16))) char smem[2 * BUF_BYTES];  // [buf][A|B]
+  // 3 LDS buffers + raw s_barrier with counted vmcnt waits: __syncthreads()
+  // with glds in flight emits vmcnt(0) and drains the NEXT k-step's loads,
+  // collapsing the pipeline (guide §6: -16..20%% at GEMM scale).  With 3
+  // buffers each barrier only waits for loads issued TWO steps back.
+  __shared__ __attribute__((aligned(16))) char smem[3 * BUF_BYTES];
 
   // ----- block swizzle (bijective XCD remap over the x*y grid) -----
   int nwg = gridDim.x * gridDim.y;
@@ -136,6 +140,9 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   }
 
   const int KT = (K + BK - 1) / BK;
+  // magic reciprocal for the per-k-step rs/S decode: a runtime integer
+  // divide is a ~40-cycle VALU sequence and sat inside the staging loop
+  const int rcpS = MODE == MODE_CONV ? 65536 / cm.S + 1 : 0;
 
   // ----- staging -----
   auto stage = [&](int buf, int kt) {
@@ -150,7 +157,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
         if (a_ok[i] && gk < K) {
           const int c0 = gk & ((1 << cm.C_log2) - 1);
           const int rs = gk >> cm.C_log2;
-          const int r = rs / cm.S, s = rs % cm.S;
+          const int r = (rs * rcpS) >> 16, s = rs - r * cm.S;
           const int hi = a_hb[i] + r, wi = a_wb[i] + s;
           if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W) {
             if (cm.ss <= 1) {
@@ -199,11 +206,29 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   const int fr = lane & 15;   // fragment row (A) / col (B, D)
   const int fs = lane >> 4;   // k-slot
 
+  // per-wave glds count of one stage() call (the vmcnt budget at barriers)
+  const bool light = (BNT == 64) && (wave >= 2);  // no B chunks to stage
+#define NT_WAIT_STAGE()                                        \
+  do {                                                         \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");         \
+    if (light)                                                 \
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");         \
+    else                                                       \
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");         \
+  } while (0)
+#define NT_WAIT_ALL()                                          \
+  do {                                                         \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");         \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");           \
+  } while (0)
+
   stage(0, 0);
-  __syncthreads();
-  int buf = 0;
+  if (KT > 1) stage(1, 1);
+  if (KT > 1) NT_WAIT_STAGE(); else NT_WAIT_ALL();
+  asm volatile("s_barrier" ::: "memory");
   for (int kt = 0; kt < KT; ++kt) {
-    if (kt + 1 < KT) stage(buf ^ 1, kt + 1);
+    const int buf = kt % 3;
+    if (kt + 2 < KT) stage((kt + 2) % 3, kt + 2);
     const char* baseA = &smem[buf * BUF_BYTES];
     const char* baseB = baseA + TILE_BYTES;
     using vec16 = typename M16<T16>::vec;
@@ -225,9 +250,13 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
 #pragma unroll
       for (int ni = 0; ni < NI; ++ni)
         acc[mi][ni] = M16<T16>::mma(af[mi], bf[ni], acc[mi][ni]);
-    __syncthreads();
-    buf ^= 1;
+    if (kt + 1 < KT) {
+      if (kt + 2 < KT) NT_WAIT_STAGE(); else NT_WAIT_ALL();
+      asm volatile("s_barrier" ::: "memory");
+    }
   }
+#undef NT_WAIT_STAGE
+#undef NT_WAIT_ALL
 
   // ----- epilogue: bias + relu + bf16 store -----
 #pragma unroll
@@ -487,8 +516,10 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
         if (mok && jj < J) src = x + (long long)gm * J + jj;
       } else {
         const int tap = tile - 1;
-        const int r = TAPS == 1 ? 0 : tap / cm.S;
-        const int s = TAPS == 1 ? 0 : tap % cm.S;
+        // TAPS==9 is only launched for 3x3 kernels, so the decode is
+        // compile-time (tap is a constant in the unrolled loop)
+        const int r = TAPS == 1 ? 0 : tap / 3;
+        const int s = TAPS == 1 ? 0 : tap % 3;
         const int hi = hb + r, wi = wb + s;
         const int jj = j0 + ig * 16 + sh8;
         if (mok && hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W && jj < Cin)

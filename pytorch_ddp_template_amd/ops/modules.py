@@ -99,7 +99,22 @@ class BatchNorm2dNHWC(nn.Module):
         self.bias = nn.Parameter(torch.zeros(num_features))
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
-        self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+        # tracked count lives host-side (a GPU scalar += here cost a kernel
+        # launch per BN layer per step); it enters state_dict via the
+        # _save/_load hooks below so checkpoints still carry it.
+        self._batches_tracked = 0
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+        destination[prefix + "num_batches_tracked"] = torch.tensor(
+            self._batches_tracked, dtype=torch.long
+        )
+
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+        key = prefix + "num_batches_tracked"
+        if key in state_dict:
+            self._batches_tracked = int(state_dict.pop(key))
+        super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
 
     def _apply(self, fn, recurse=True):
         # Running stats stay fp32 regardless of model dtype: the BN kernels
@@ -112,7 +127,7 @@ class BatchNorm2dNHWC(nn.Module):
 
     def forward(self, x):
         if self.training:
-            self.num_batches_tracked += 1
+            self._batches_tracked += 1
         return X.batch_norm2d_nhwc(
             x,
             self.weight,
