@@ -25,6 +25,8 @@ from ..ops import (
     MaxPool2dNHWC,
     add_relu,
 )
+from ..ops.fused_block import fused_basic_block
+from ..ops.native import native_available
 
 
 def _conv_bn(in_ch, out_ch, k, stride, pad, act=None):
@@ -39,6 +41,7 @@ class BasicBlock(nn.Module):
 
     def __init__(self, in_ch, ch, stride=1):
         super().__init__()
+        self.stride = stride
         self.conv1 = _conv_bn(in_ch, ch, 3, stride, 1, act="relu")
         self.conv2 = _conv_bn(ch, ch, 3, 1, 1)
         self.down = (
@@ -47,7 +50,29 @@ class BasicBlock(nn.Module):
             else None
         )
 
+    def _can_fuse(self, x):
+        # single-autograd-node fused path (ops/fused_block.py): training-mode
+        # GPU bf16/f16 with pow2 channels (the fast conv kernels' domain)
+        if not (self.training and x.is_cuda and torch.is_grad_enabled()):
+            return False
+        if x.dtype not in (torch.bfloat16, torch.float16):
+            return False
+        c_in, c_out = x.shape[-1], self.conv1[0].weight.shape[0]
+        pow2 = lambda v: v >= 8 and (v & (v - 1)) == 0  # noqa: E731
+        return pow2(c_in) and pow2(c_out) and native_available()
+
     def forward(self, x):
+        if self._can_fuse(x):
+            for m in (self.conv1[1], self.conv2[1]) + (
+                (self.down[1],) if self.down is not None else ()
+            ):
+                m._batches_tracked += 1
+            return fused_basic_block(
+                x, self.conv1[0], self.conv1[1], self.conv2[0], self.conv2[1],
+                self.down[0] if self.down is not None else None,
+                self.down[1] if self.down is not None else None,
+                self.stride,
+            )
         idt = x if self.down is None else self.down(x)
         out = self.conv2(self.conv1(x))
         return add_relu(out, idt)

@@ -76,12 +76,19 @@ enum { MODE_PLAIN = 0, MODE_CONV = 1 };
 // ---------------------------------------------------------------- NT -----
 // BNT: the N tile (128, or 64 when N <= 64 — ResNet layer1 Kout, C=64
 // dgrads — so half the MFMAs aren't wasted on zero-page columns).
+// Optional epilogue extras (runtime pointers, null = off):
+//  * stats_ws [2*nblocks][N]: per-block BN partial sums/sumsqs of the
+//    written outputs (the BN stats pass then never re-reads the tensor);
+//  * addend (+addend_mask): C = acc + (mask>0 ? addend : 0) — fuses the
+//    residual skip-gradient add (and its ReLU mask) into conv dgrad.
 template <typename T16, int MODE, bool RELU, bool HAS_BIAS, int BNT = BN>
 __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     const T16* __restrict__ A, const T16* __restrict__ B,
     T16* __restrict__ C, const T16* __restrict__ bias,
     const T16* __restrict__ zpad, int M, int N, int K, long long strideA,
-    long long strideB, long long strideC, ConvMeta cm) {
+    long long strideB, long long strideC, ConvMeta cm,
+    float* __restrict__ stats_ws, int ws_nblocks,
+    const T16* __restrict__ addend, const T16* __restrict__ addend_mask) {
   constexpr int NI = BNT / 32;              // B fragments per wave
   constexpr int BUF_BYTES = TILE_BYTES + BNT * 64;  // A tile + B tile
   // 3 LDS buffers + raw s_barrier with counted vmcnt waits: __syncthreads()
@@ -93,6 +100,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   // ----- block swizzle (bijective XCD remap over the x*y grid) -----
   int nwg = gridDim.x * gridDim.y;
   int id = blockIdx.y * gridDim.x + blockIdx.x;
+  const int flat_id = id;  // stats_ws row (pre-swizzle: any bijection works)
   if (nwg >= 16) {
     int q = nwg >> 3, r = nwg & 7;
     int xcd = id & 7, idx = id >> 3;
@@ -258,7 +266,8 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
 #undef NT_WAIT_STAGE
 #undef NT_WAIT_ALL
 
-  // ----- epilogue: bias + relu + bf16 store -----
+  // ----- epilogue: bias + relu (+addend) + bf16 store (+BN stat partials) --
+  float col_sum[NI] = {}, col_sq[NI] = {};
 #pragma unroll
   for (int ni = 0; ni < NI; ++ni) {
     const int col = n0 + wn + ni * 16 + fr;
@@ -273,7 +282,48 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
         if (row >= M) continue;
         float v = acc[mi][ni][r] + bv;
         if (RELU) v = fmaxf(v, 0.f);
+        if (addend) {
+          const long long idx = (long long)row * N + col;
+          float a = to_f(addend[idx]);
+          if (addend_mask && !(to_f(addend_mask[idx]) > 0.f)) a = 0.f;
+          v += a;
+        }
         C[(long long)row * N + col] = to_t<T16>(v);
+        if (stats_ws) {
+          col_sum[ni] += v;
+          col_sq[ni] += v * v;
+        }
+      }
+    }
+  }
+  if (stats_ws) {
+    // reduce over fs lanes (same col, different rows), then cross-wave via
+    // LDS (waves with equal wn cover the same columns), then one plain
+    // store per column into this block's workspace rows.
+    float* lsum = reinterpret_cast<float*>(smem);         // [BNT]
+    float* lsq = lsum + BNT;                              // [BNT]
+    __syncthreads();  // everyone done with the k-loop LDS buffers
+    for (int c = threadIdx.x; c < 2 * BNT; c += blockDim.x) lsum[c] = 0.f;
+    __syncthreads();
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) {
+      float sv = col_sum[ni], qv = col_sq[ni];
+      sv += __shfl_xor(sv, 16);
+      sv += __shfl_xor(sv, 32);
+      qv += __shfl_xor(qv, 16);
+      qv += __shfl_xor(qv, 32);
+      const int col = n0 + wn + ni * 16 + fr;
+      if (fs == 0 && col < N) {
+        atomicAdd(&lsum[wn + ni * 16 + fr], sv);
+        atomicAdd(&lsq[wn + ni * 16 + fr], qv);
+      }
+    }
+    __syncthreads();
+    for (int c = threadIdx.x; c < BNT; c += blockDim.x) {
+      const int col = n0 + c;
+      if (col < N) {
+        stats_ws[(long long)flat_id * N + col] = lsum[c];
+        stats_ws[((long long)ws_nblocks + flat_id) * N + col] = lsq[c];
       }
     }
   }
@@ -641,12 +691,19 @@ torch::Tensor& zero_page(const torch::Device& dev, torch::ScalarType st) {
   return it->second;
 }
 
+struct NtExtras {
+  float* stats_ws = nullptr;  // [2*nblocks][N], pre-zeroed
+  int ws_nblocks = 0;
+  const void* addend = nullptr;
+  const void* addend_mask = nullptr;
+};
+
 template <typename t16, int MODE>
 void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
                  torch::Tensor& C, const c10::optional<torch::Tensor>& bias,
                  bool relu, const torch::Tensor& zp, dim3 grid, int M, int N,
                  int K, long long sA, long long sB, long long sC,
-                 g16::ConvMeta cm) {
+                 g16::ConvMeta cm, NtExtras ex = {}) {
   auto stream = c10::hip::getCurrentHIPStream();
   const t16* bias_p =
       bias.has_value() ? reinterpret_cast<const t16*>(bias->data_ptr())
@@ -664,7 +721,9 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
                          reinterpret_cast<const t16*>(B.data_ptr()),          \
                          reinterpret_cast<t16*>(C.data_ptr()), bias_p,        \
                          reinterpret_cast<const t16*>(zp.data_ptr()), M, N,   \
-                         K, sA, sB, sC, cm);                                  \
+                         K, sA, sB, sC, cm, ex.stats_ws, ex.ws_nblocks,       \
+                         reinterpret_cast<const t16*>(ex.addend),             \
+                         reinterpret_cast<const t16*>(ex.addend_mask));       \
     else                                                                      \
       hipLaunchKernelGGL((g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB>),     \
                          grid, dim3(g16::THREADS), 0, stream,                 \
@@ -672,7 +731,9 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
                          reinterpret_cast<const t16*>(B.data_ptr()),          \
                          reinterpret_cast<t16*>(C.data_ptr()), bias_p,        \
                          reinterpret_cast<const t16*>(zp.data_ptr()), M, N,   \
-                         K, sA, sB, sC, cm);                                  \
+                         K, sA, sB, sC, cm, ex.stats_ws, ex.ws_nblocks,       \
+                         reinterpret_cast<const t16*>(ex.addend),             \
+                         reinterpret_cast<const t16*>(ex.addend_mask));       \
   } while (0)
   if (relu) {
     if (bias_p) LAUNCH_NT16(true, true);
@@ -722,9 +783,12 @@ torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
 }
 
 // conv2d forward, NHWC x[N,H,W,C] * w[Kout,R,S,C] -> y[N,HO,WO,Kout]
-torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
-                              c10::optional<torch::Tensor> bias, int64_t stride,
-                              int64_t pad, bool relu) {
+// stats=true additionally returns the BN partial workspace ws
+// ([2*nblocks][Kout], block-level sums/sumsqs of y) so the following
+// BatchNorm needs no stats pass over y (bn_fwd_ws consumes it).
+std::vector<torch::Tensor> conv2d_fwd_bf16_impl(
+    torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias,
+    int64_t stride, int64_t pad, bool relu, bool stats) {
   int N = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
       Cin = (int)x.size(3);
   int Kout = (int)w.size(0), R = (int)w.size(1), S = (int)w.size(2);
@@ -737,14 +801,37 @@ torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
   auto y = torch::empty({N, HO, WO, Kout}, x.options());
   auto& zp = zero_page(x.device(), x.scalar_type());
   dim3 grid((Kout + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
+  const int nblocks =
+      (int)(((Kout <= 64 ? (Kout + 63) / 64 : grid.x)) * grid.y);
   g16::ConvMeta cm{H, W, cl, S, R, (int)stride, (int)pad, HO, WO};
+  NtExtras ex{};
+  torch::Tensor ws;
+  if (stats) {
+    ws = torch::zeros({2LL * nblocks, Kout},
+                      x.options().dtype(torch::kFloat32));
+    ex.stats_ws = ws.data_ptr<float>();
+    ex.ws_nblocks = nblocks;
+  }
   if (x.scalar_type() == torch::kBFloat16)
     launch_nt16<bf16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M, Kout,
-                                      K, 0, 0, 0, cm);
+                                      K, 0, 0, 0, cm, ex);
   else
     launch_nt16<_Float16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M,
-                                          Kout, K, 0, 0, 0, cm);
-  return y;
+                                          Kout, K, 0, 0, 0, cm, ex);
+  if (stats) return {y, ws};
+  return {y};
+}
+
+torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
+                              c10::optional<torch::Tensor> bias, int64_t stride,
+                              int64_t pad, bool relu) {
+  return conv2d_fwd_bf16_impl(x, w, bias, stride, pad, relu, false)[0];
+}
+
+std::vector<torch::Tensor> conv2d_fwd_stats_bf16(
+    torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias,
+    int64_t stride, int64_t pad, bool relu) {
+  return conv2d_fwd_bf16_impl(x, w, bias, stride, pad, relu, true);
 }
 
 // conv2d dgrad, direct: dx = conv(zero_stuffed(dy), wr, 1, R-1-pad) with the
@@ -753,7 +840,9 @@ torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
 // 180°-rotated weight), stride/pad are the ORIGINAL forward conv's.
 torch::Tensor conv2d_dgrad_bf16(torch::Tensor dy, torch::Tensor wr,
                                 int64_t stride, int64_t pad, int64_t H,
-                                int64_t W) {
+                                int64_t W,
+                                c10::optional<torch::Tensor> addend,
+                                c10::optional<torch::Tensor> addend_mask) {
   int N = (int)dy.size(0), HOs = (int)dy.size(1), WOs = (int)dy.size(2),
       Kout = (int)dy.size(3);
   int Cin = (int)wr.size(0), R = (int)wr.size(1), S = (int)wr.size(2);
@@ -775,12 +864,24 @@ torch::Tensor conv2d_dgrad_bf16(torch::Tensor dy, torch::Tensor wr,
   dim3 grid((Cin + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
   g16::ConvMeta cm{HS,   WS,   cl,   S,  R, 1, dpad, HOut, WOut,
                    (int)stride, HOs, WOs};
+  NtExtras ex{};
+  if (addend.has_value()) {
+    TORCH_CHECK(addend->is_contiguous() && addend->numel() == dx.numel() &&
+                addend->scalar_type() == dy.scalar_type());
+    ex.addend = addend->data_ptr();
+    if (addend_mask.has_value()) {
+      TORCH_CHECK(addend_mask->is_contiguous() &&
+                  addend_mask->numel() == dx.numel() &&
+                  addend_mask->scalar_type() == dy.scalar_type());
+      ex.addend_mask = addend_mask->data_ptr();
+    }
+  }
   if (dy.scalar_type() == torch::kBFloat16)
     launch_nt16<bf16, g16::MODE_CONV>(dy, wr, dx, {}, false, zp, grid, M, Cin,
-                                      K, 0, 0, 0, cm);
+                                      K, 0, 0, 0, cm, ex);
   else
     launch_nt16<_Float16, g16::MODE_CONV>(dy, wr, dx, {}, false, zp, grid, M,
-                                          Cin, K, 0, 0, 0, cm);
+                                          Cin, K, 0, 0, 0, cm, ex);
   return dx;
 }
 

@@ -532,6 +532,51 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   return {y, mean, rstd};
 }
 
+// BN forward from a precomputed partial workspace (the producing conv's
+// epilogue wrote per-block sums/sumsqs of its output into ws [2*nb][C]) —
+// the stats pass over x disappears entirely.
+std::vector<torch::Tensor> bn_fwd_ws(torch::Tensor x, torch::Tensor gamma,
+                                     torch::Tensor beta, torch::Tensor ws,
+                                     c10::optional<torch::Tensor> running_mean,
+                                     c10::optional<torch::Tensor> running_var,
+                                     double momentum, double eps, bool relu) {
+  TORCH_CHECK(x.dim() == 2 && x.is_contiguous());
+  long long M = x.size(0);
+  int C = (int)x.size(1);
+  TORCH_CHECK(ws.size(1) == C && ws.size(0) % 2 == 0);
+  const int nb = (int)(ws.size(0) / 2);
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({C}, f32);
+  auto rstd = torch::empty({C}, f32);
+  auto y = torch::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStream();
+  TORCH_CHECK(C % 8 == 0 && C <= 2048);
+  DDP_DISPATCH_FLOAT(x.scalar_type(), "bn_fwd_ws", [&] {
+    hipLaunchKernelGGL(
+        nrm::bn_finalize_kernel, dim3(C), dim3(256), 0, stream,
+        ws.data_ptr<float>(), nb, mean.data_ptr<float>(),
+        rstd.data_ptr<float>(),
+        running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
+        running_var.has_value() ? running_var->data_ptr<float>() : nullptr,
+        (float)momentum, (float)eps, M, C);
+    const auto* xp = reinterpret_cast<const scalar_t*>(x.data_ptr());
+    const auto* gp = reinterpret_cast<const scalar_t*>(gamma.data_ptr());
+    const auto* bp = reinterpret_cast<const scalar_t*>(beta.data_ptr());
+    auto* yp = reinterpret_cast<scalar_t*>(y.data_ptr());
+    if (relu)
+      hipLaunchKernelGGL((nrm::bn_norm_kernel<scalar_t, true>),
+                         dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
+                         xp, yp, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         gp, bp, M, C);
+    else
+      hipLaunchKernelGGL((nrm::bn_norm_kernel<scalar_t, false>),
+                         dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
+                         xp, yp, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         gp, bp, M, C);
+  });
+  return {y, mean, rstd};
+}
+
 torch::Tensor bn_infer(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
                        torch::Tensor rmean, torch::Tensor rvar, double eps,
                        bool relu) {

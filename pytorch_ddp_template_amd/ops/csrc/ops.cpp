@@ -42,7 +42,12 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                                 int64_t S);
 torch::Tensor conv2d_dgrad_bf16(torch::Tensor dy, torch::Tensor wr,
                                 int64_t stride, int64_t pad, int64_t H,
-                                int64_t W);
+                                int64_t W,
+                                c10::optional<torch::Tensor> addend,
+                                c10::optional<torch::Tensor> addend_mask);
+std::vector<torch::Tensor> conv2d_fwd_stats_bf16(
+    torch::Tensor x, torch::Tensor w, c10::optional<torch::Tensor> bias,
+    int64_t stride, int64_t pad, bool relu);
 // gemm_f32.hip
 torch::Tensor bmm_nt_f32(torch::Tensor A, torch::Tensor B,
                          c10::optional<torch::Tensor> bias, bool relu);
@@ -59,6 +64,11 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   c10::optional<torch::Tensor> running_mean,
                                   c10::optional<torch::Tensor> running_var,
                                   double momentum, double eps, bool relu);
+std::vector<torch::Tensor> bn_fwd_ws(torch::Tensor x, torch::Tensor gamma,
+                                     torch::Tensor beta, torch::Tensor ws,
+                                     c10::optional<torch::Tensor> running_mean,
+                                     c10::optional<torch::Tensor> running_var,
+                                     double momentum, double eps, bool relu);
 torch::Tensor bn_infer(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
                        torch::Tensor rmean, torch::Tensor rvar, double eps,
                        bool relu);
@@ -188,7 +198,9 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
 }
 
 torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
-                           int64_t pad, int64_t H, int64_t W) {
+                           int64_t pad, int64_t H, int64_t W,
+                           c10::optional<torch::Tensor> addend,
+                           c10::optional<torch::Tensor> addend_mask) {
   // dx = conv(zero_stuffed(dy), rot(w), stride=1, pad=R-1-pad); output
   // padding covers (H+2p-R) % stride != 0 (inputs the fwd never reached).
   int R = (int)w.size(1), S = (int)w.size(2);
@@ -198,13 +210,20 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
   if (is_bf16(dy) ? pow2_ge(Kout, 8) : false) {
     // bf16/f16 fast path: zero-stuffing folded into the GEMM's im2col
     // gather (never materialized) — gemm_bf16.hip conv2d_dgrad_bf16.
-    return conv2d_dgrad_bf16(dy.contiguous(), wr, stride, pad, H, W);
+    return conv2d_dgrad_bf16(dy.contiguous(), wr, stride, pad, H, W, addend,
+                             addend_mask);
   }
   int64_t opad_h = (H + 2 * pad - R) % stride;
   int64_t opad_w = (W + 2 * pad - S) % stride;
   auto dys = stride > 1 ? zero_stuff(dy.contiguous(), stride, opad_h, opad_w)
                         : dy.contiguous();
   auto dx = conv2d_fwd(dys, wr, {}, 1, R - 1 - pad, false);
+  if (addend.has_value()) {
+    auto a = *addend;
+    if (addend_mask.has_value())
+      a = a * (addend_mask->reshape(a.sizes()) > 0).to(a.scalar_type());
+    dx = dx + a.reshape(dx.sizes());
+  }
   TORCH_CHECK(dx.size(1) == H && dx.size(2) == W,
               "dgrad shape mismatch: got ", dx.sizes(), " want H=", H, " W=",
               W, " (input H+2p-R must be divisible by stride)");
@@ -246,7 +265,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias") = c10::nullopt, py::arg("stride") = 1,
         py::arg("pad") = 0, py::arg("relu") = false);
-  m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_dgrad", &conv2d_dgrad, py::arg("dy"), py::arg("w"),
+        py::arg("stride"), py::arg("pad"), py::arg("H"), py::arg("W"),
+        py::arg("addend") = c10::nullopt,
+        py::arg("addend_mask") = c10::nullopt);
+  m.def("conv2d_fwd_stats", &conv2d_fwd_stats_bf16, py::arg("x"),
+        py::arg("w"), py::arg("bias") = c10::nullopt, py::arg("stride") = 1,
+        py::arg("pad") = 0, py::arg("relu") = false);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("relu_fwd", &relu_fwd);
   m.def("relu_bwd", &relu_bwd);
@@ -259,6 +284,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2d_fwd", &maxpool2d_fwd);
   m.def("maxpool2d_bwd", &maxpool2d_bwd);
   m.def("bn_fwd", &bn_fwd);
+  m.def("bn_fwd_ws", &bn_fwd_ws);
   m.def("bn_infer", &bn_infer);
   m.def("bn_bwd", &bn_bwd, py::arg("dy"), py::arg("x"), py::arg("gamma"),
         py::arg("mean"), py::arg("rstd"), py::arg("y_relu") = c10::nullopt);

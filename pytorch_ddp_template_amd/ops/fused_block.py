@@ -1,0 +1,117 @@
+"""Fused ResNet BasicBlock — one autograd node for conv1→bn1(relu)→conv2→bn2
+(+ optional 1x1 downsample) → add+relu.
+
+Why (measured on MI355X, ResNet-18/CIFAR bf16 step profile):
+* the conv epilogue already holds its output tile in registers, so it emits
+  the BN per-block partial sums for free (``conv2d_fwd_stats``) — the BN
+  stats pass that re-read every conv output disappears;
+* BN backward applies the downstream ReLU mask inline (``bn_bwd(y_relu=·)``)
+  — no relu_bwd pass;
+* the residual skip gradient (ReLU-masked dy) is accumulated inside the
+  conv1/downsample dgrad epilogue (``conv2d_dgrad(addend=·)``) — the
+  autograd-engine at::add of two full-size grad tensors disappears.
+
+The unfused module path (ops/modules.py) remains the CPU/eval/odd-shape
+reference; tests compare the two.  Reference scope note: the reference
+template has no conv/BN at all (these ops come from the
+BASELINE.json ResNet configs).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .native import native
+
+
+def _flat(t):
+    return t.reshape(-1, t.shape[-1])
+
+
+class _FusedBasicBlockFn(torch.autograd.Function):
+    @staticmethod
+    def forward(
+        ctx, x, w1, g1, b1, w2, g2, b2, wd, gd, bd, bn1, bn2, bnd, stride
+    ):
+        ext = native()
+        x = x.contiguous()
+        mom, eps = bn1.momentum, bn1.eps
+        t1, ws1 = ext.conv2d_fwd_stats(x, w1, None, stride, 1, False)
+        z2, mean1, rstd1 = ext.bn_fwd_ws(
+            _flat(t1), g1, b1, ws1, bn1.running_mean, bn1.running_var,
+            mom, eps, True,
+        )
+        z = z2.reshape(t1.shape)
+        t2, ws2 = ext.conv2d_fwd_stats(z, w2, None, 1, 1, False)
+        y2, mean2, rstd2 = ext.bn_fwd_ws(
+            _flat(t2), g2, b2, ws2, bn2.running_mean, bn2.running_var,
+            mom, eps, False,
+        )
+        if wd is not None:
+            td, wsd = ext.conv2d_fwd_stats(x, wd, None, stride, 0, False)
+            idn2, meand, rstdd = ext.bn_fwd_ws(
+                _flat(td), gd, bd, wsd, bnd.running_mean, bnd.running_var,
+                mom, eps, False,
+            )
+            idn = idn2.reshape(td.shape)
+        else:
+            td = meand = rstdd = None
+            idn = x
+        out = ext.add_relu_fwd(y2.reshape(t2.shape), idn)
+        ctx.save_for_backward(
+            x, w1, g1, t1, z, mean1, rstd1, w2, g2, t2, mean2, rstd2,
+            wd, gd, td, meand, rstdd, out,
+        )
+        ctx.stride = stride
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x, w1, g1, t1, z, mean1, rstd1, w2, g2, t2, mean2, rstd2,
+         wd, gd, td, meand, rstdd, out) = ctx.saved_tensors
+        stride = ctx.stride
+        ext = native()
+        dy = dy.contiguous()
+        dy2, out2 = _flat(dy), _flat(out)
+        H, W = x.shape[1], x.shape[2]
+        Hz, Wz = z.shape[1], z.shape[2]
+
+        # bn2 backward, ReLU mask (from the add_relu output) fused in
+        dx2, dg2, db2 = ext.bn_bwd(dy2, _flat(t2), g2, mean2, rstd2, out2)
+        dx2 = dx2.reshape(t2.shape)
+        # conv2 backward
+        dz = ext.conv2d_dgrad(dx2, w2, 1, 1, Hz, Wz)
+        dw2 = ext.conv2d_wgrad(dx2, z, 1, 1, 3, 3).to(w2.dtype)
+        # bn1 backward (its own fused ReLU: z is bn1's relu output)
+        dt1, dg1, db1 = ext.bn_bwd(
+            _flat(dz), _flat(t1), g1, mean1, rstd1, _flat(z)
+        )
+        dt1 = dt1.reshape(t1.shape)
+        dw1 = ext.conv2d_wgrad(dt1, x, stride, 1, 3, 3).to(w1.dtype)
+
+        if wd is None:
+            # identity skip: masked dy accumulates inside conv1's dgrad
+            dx = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W, dy, out)
+            dwd = dgd = dbd = None
+        else:
+            dxa = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W)
+            dtd, dgd, dbd = ext.bn_bwd(
+                dy2, _flat(td), gd, meand, rstdd, out2
+            )
+            dtd = dtd.reshape(td.shape)
+            dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
+            dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W, dxa)
+        return (dx, dw1, dg1, db1, dw2, dg2, db2, dwd, dgd, dbd,
+                None, None, None, None)
+
+
+def fused_basic_block(x, conv1, bn1, conv2, bn2, convd, bnd, stride):
+    """Run a BasicBlock through the fused Function (training, native path)."""
+    return _FusedBasicBlockFn.apply(
+        x, conv1.weight, bn1.weight, bn1.bias,
+        conv2.weight, bn2.weight, bn2.bias,
+        convd.weight if convd is not None else None,
+        bnd.weight if bnd is not None else None,
+        bnd.bias if bnd is not None else None,
+        bn1, bn2, bnd, stride,
+    )
