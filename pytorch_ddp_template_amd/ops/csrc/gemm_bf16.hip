@@ -606,13 +606,17 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
       const unsigned a1 =
           LDS_BYTE(base + ((wm >> 4) + 1) * IMG) + tr_lane_off;
       v4s l0, h0, l1, h1;
+      // "=&v" (early-clobber) is load-bearing: without it LLVM may alias an
+      // output pair with an address input, and since ds_read results land
+      // asynchronously (lgkmcnt), a later read in the block can consume a
+      // clobbered address — a cold-launch-timing-dependent corruption.
       asm volatile(
           "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
           "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
           "ds_read_b64_tr_b16 %2, %5 offset:0\n\t"
           "ds_read_b64_tr_b16 %3, %5 offset:128\n\t"
           "s_waitcnt lgkmcnt(0)"
-          : "=v"(l0), "=v"(h0), "=v"(l1), "=v"(h1)
+          : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
           : "v"(a0), "v"(a1));
       reinterpret_cast<v4s*>(&af[0])[0] = l0;
       reinterpret_cast<v4s*>(&af[0])[1] = h0;
@@ -634,7 +638,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
           "ds_read_b64_tr_b16 %2, %5 offset:0\n\t"
           "ds_read_b64_tr_b16 %3, %5 offset:128\n\t"
           "s_waitcnt lgkmcnt(0)"
-          : "=v"(l0), "=v"(h0), "=v"(l1), "=v"(h1)
+          : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
           : "v"(b0), "v"(b1));
       reinterpret_cast<v4s*>(&bfr[0])[0] = l0;
       reinterpret_cast<v4s*>(&bfr[0])[1] = h0;
