@@ -297,14 +297,13 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     }
   }
   if (stats_ws) {
-    // reduce over fs lanes (same col, different rows), then cross-wave via
-    // LDS (waves with equal wn cover the same columns), then one plain
-    // store per column into this block's workspace rows.
-    float* lsum = reinterpret_cast<float*>(smem);         // [BNT]
-    float* lsq = lsum + BNT;                              // [BNT]
-    __syncthreads();  // everyone done with the k-loop LDS buffers
-    for (int c = threadIdx.x; c < 2 * BNT; c += blockDim.x) lsum[c] = 0.f;
-    __syncthreads();
+    // barrier-free: reduce over the fs lanes (same col, different rows) via
+    // two shuffles, then each wave PAIR (wave>>1 disambiguates waves that
+    // share a column range but cover different row halves) stores its
+    // 32-col slice straight into its own workspace row.  ws therefore has
+    // 2 rows per block (ws_nblocks = 2 * grid blocks); the finalize pass
+    // reduces rows regardless.
+    const long long wsrow = (long long)flat_id * 2 + (wave >> 1);
 #pragma unroll
     for (int ni = 0; ni < NI; ++ni) {
       float sv = col_sum[ni], qv = col_sq[ni];
@@ -314,16 +313,8 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
       qv += __shfl_xor(qv, 32);
       const int col = n0 + wn + ni * 16 + fr;
       if (fs == 0 && col < N) {
-        atomicAdd(&lsum[wn + ni * 16 + fr], sv);
-        atomicAdd(&lsq[wn + ni * 16 + fr], qv);
-      }
-    }
-    __syncthreads();
-    for (int c = threadIdx.x; c < BNT; c += blockDim.x) {
-      const int col = n0 + c;
-      if (col < N) {
-        stats_ws[(long long)flat_id * N + col] = lsum[c];
-        stats_ws[((long long)ws_nblocks + flat_id) * N + col] = lsq[c];
+        stats_ws[wsrow * N + col] = sv;
+        stats_ws[((long long)ws_nblocks + wsrow) * N + col] = qv;
       }
     }
   }
@@ -805,8 +796,9 @@ std::vector<torch::Tensor> conv2d_fwd_bf16_impl(
   auto y = torch::empty({N, HO, WO, Kout}, x.options());
   auto& zp = zero_page(x.device(), x.scalar_type());
   dim3 grid((Kout + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
+  // 2 workspace rows per block (per wave-pair; see the epilogue)
   const int nblocks =
-      (int)(((Kout <= 64 ? (Kout + 63) / 64 : grid.x)) * grid.y);
+      2 * (int)(((Kout <= 64 ? (Kout + 63) / 64 : grid.x)) * grid.y);
   g16::ConvMeta cm{H, W, cl, S, R, (int)stride, (int)pad, HO, WO};
   NtExtras ex{};
   torch::Tensor ws;
