@@ -25,7 +25,7 @@ from ..ops import (
     MaxPool2dNHWC,
     add_relu,
 )
-from ..ops.fused_block import fused_basic_block
+from ..ops.fused_block import fused_basic_block, fused_bottleneck
 from ..ops.native import native_available
 
 
@@ -83,6 +83,7 @@ class Bottleneck(nn.Module):
 
     def __init__(self, in_ch, ch, stride=1):
         super().__init__()
+        self.stride = stride
         self.conv1 = _conv_bn(in_ch, ch, 1, 1, 0, act="relu")
         self.conv2 = _conv_bn(ch, ch, 3, stride, 1, act="relu")
         self.conv3 = _conv_bn(ch, ch * self.expansion, 1, 1, 0)
@@ -92,7 +93,22 @@ class Bottleneck(nn.Module):
             else None
         )
 
+    _can_fuse = BasicBlock._can_fuse
+
     def forward(self, x):
+        if self._can_fuse(x):
+            bns = [self.conv1[1], self.conv2[1], self.conv3[1]] + (
+                [self.down[1]] if self.down is not None else []
+            )
+            for m in bns:
+                m._batches_tracked += 1
+            return fused_bottleneck(
+                x, self.conv1[0], self.conv1[1], self.conv2[0], self.conv2[1],
+                self.conv3[0], self.conv3[1],
+                self.down[0] if self.down is not None else None,
+                self.down[1] if self.down is not None else None,
+                self.stride,
+            )
         idt = x if self.down is None else self.down(x)
         out = self.conv3(self.conv2(self.conv1(x)))
         return add_relu(out, idt)

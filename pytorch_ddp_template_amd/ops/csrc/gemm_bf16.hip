@@ -273,6 +273,22 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     const int col = n0 + wn + ni * 16 + fr;
     if (col >= N) continue;
     float bv = HAS_BIAS ? to_f(bias[col]) : 0.f;
+    // gather the residual addend (+mask) up front: 16 independent loads
+    // issue back-to-back and overlap one memory latency, instead of
+    // load->use->store serializing per element
+    float av[4][4];
+    if (addend) {
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = m0 + wm + mi * 16 + fs * 4 + r;
+          const long long idx = (long long)row * N + col;
+          av[mi][r] = row < M ? to_f(addend[idx]) : 0.f;
+          if (addend_mask && row < M && !(to_f(addend_mask[idx]) > 0.f))
+            av[mi][r] = 0.f;
+        }
+    }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const int row_base = m0 + wm + mi * 16 + fs * 4;
@@ -282,12 +298,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
         if (row >= M) continue;
         float v = acc[mi][ni][r] + bv;
         if (RELU) v = fmaxf(v, 0.f);
-        if (addend) {
-          const long long idx = (long long)row * N + col;
-          float a = to_f(addend[idx]);
-          if (addend_mask && !(to_f(addend_mask[idx]) > 0.f)) a = 0.f;
-          v += a;
-        }
+        if (addend) v += av[mi][r];
         C[(long long)row * N + col] = to_t<T16>(v);
         if (stats_ws) {
           col_sum[ni] += v;

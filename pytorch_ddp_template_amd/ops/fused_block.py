@@ -115,3 +115,98 @@ def fused_basic_block(x, conv1, bn1, conv2, bn2, convd, bnd, stride):
         bnd.bias if bnd is not None else None,
         bn1, bn2, bnd, stride,
     )
+
+
+class _FusedBottleneckFn(torch.autograd.Function):
+    """ResNet Bottleneck (1x1 -> 3x3(s) -> 1x1, + optional 1x1(s) downsample)
+    as one autograd node, same fusions as the BasicBlock variant."""
+
+    @staticmethod
+    def forward(ctx, x, w1, g1, b1, w2, g2, b2, w3, g3, b3, wd, gd, bd,
+                bn1, bn2, bn3, bnd, stride):
+        ext = native()
+        x = x.contiguous()
+        mom, eps = bn1.momentum, bn1.eps
+        t1, ws1 = ext.conv2d_fwd_stats(x, w1, None, 1, 0, False)
+        z1f, mean1, rstd1 = ext.bn_fwd_ws(
+            _flat(t1), g1, b1, ws1, bn1.running_mean, bn1.running_var,
+            mom, eps, True)
+        z1 = z1f.reshape(t1.shape)
+        t2, ws2 = ext.conv2d_fwd_stats(z1, w2, None, stride, 1, False)
+        z2f, mean2, rstd2 = ext.bn_fwd_ws(
+            _flat(t2), g2, b2, ws2, bn2.running_mean, bn2.running_var,
+            mom, eps, True)
+        z2 = z2f.reshape(t2.shape)
+        t3, ws3 = ext.conv2d_fwd_stats(z2, w3, None, 1, 0, False)
+        y3, mean3, rstd3 = ext.bn_fwd_ws(
+            _flat(t3), g3, b3, ws3, bn3.running_mean, bn3.running_var,
+            mom, eps, False)
+        if wd is not None:
+            td, wsd = ext.conv2d_fwd_stats(x, wd, None, stride, 0, False)
+            idnf, meand, rstdd = ext.bn_fwd_ws(
+                _flat(td), gd, bd, wsd, bnd.running_mean, bnd.running_var,
+                mom, eps, False)
+            idn = idnf.reshape(td.shape)
+        else:
+            td = meand = rstdd = None
+            idn = x
+        out = ext.add_relu_fwd(y3.reshape(t3.shape), idn)
+        ctx.save_for_backward(
+            x, w1, g1, t1, z1, mean1, rstd1, w2, g2, t2, z2, mean2, rstd2,
+            w3, g3, t3, mean3, rstd3, wd, gd, td, meand, rstdd, out)
+        ctx.stride = stride
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x, w1, g1, t1, z1, mean1, rstd1, w2, g2, t2, z2, mean2, rstd2,
+         w3, g3, t3, mean3, rstd3, wd, gd, td, meand, rstdd, out) = (
+            ctx.saved_tensors)
+        stride = ctx.stride
+        ext = native()
+        dy = dy.contiguous()
+        dy2, out2 = _flat(dy), _flat(out)
+        H, W = x.shape[1], x.shape[2]
+        H1, W1 = z1.shape[1], z1.shape[2]
+        H2, W2 = z2.shape[1], z2.shape[2]
+
+        dx3, dg3, db3 = ext.bn_bwd(dy2, _flat(t3), g3, mean3, rstd3, out2)
+        dx3 = dx3.reshape(t3.shape)
+        dz2 = ext.conv2d_dgrad(dx3, w3, 1, 0, H2, W2)
+        dw3 = ext.conv2d_wgrad(dx3, z2, 1, 0, 1, 1).to(w3.dtype)
+
+        dt2, dg2, db2 = ext.bn_bwd(
+            _flat(dz2), _flat(t2), g2, mean2, rstd2, _flat(z2))
+        dt2 = dt2.reshape(t2.shape)
+        dz1 = ext.conv2d_dgrad(dt2, w2, stride, 1, H1, W1)
+        dw2 = ext.conv2d_wgrad(dt2, z1, stride, 1, 3, 3).to(w2.dtype)
+
+        dt1, dg1, db1 = ext.bn_bwd(
+            _flat(dz1), _flat(t1), g1, mean1, rstd1, _flat(z1))
+        dt1 = dt1.reshape(t1.shape)
+        dw1 = ext.conv2d_wgrad(dt1, x, 1, 0, 1, 1).to(w1.dtype)
+
+        if wd is None:
+            dx = ext.conv2d_dgrad(dt1, w1, 1, 0, H, W, dy, out)
+            dwd = dgd = dbd = None
+        else:
+            dxa = ext.conv2d_dgrad(dt1, w1, 1, 0, H, W)
+            dtd, dgd, dbd = ext.bn_bwd(dy2, _flat(td), gd, meand, rstdd, out2)
+            dtd = dtd.reshape(td.shape)
+            dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
+            dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W, dxa)
+        return (dx, dw1, dg1, db1, dw2, dg2, db2, dw3, dg3, db3,
+                dwd, dgd, dbd, None, None, None, None, None)
+
+
+def fused_bottleneck(x, conv1, bn1, conv2, bn2, conv3, bn3, convd, bnd,
+                     stride):
+    return _FusedBottleneckFn.apply(
+        x, conv1.weight, bn1.weight, bn1.bias,
+        conv2.weight, bn2.weight, bn2.bias,
+        conv3.weight, bn3.weight, bn3.bias,
+        convd.weight if convd is not None else None,
+        bnd.weight if bnd is not None else None,
+        bnd.bias if bnd is not None else None,
+        bn1, bn2, bn3, bnd, stride,
+    )

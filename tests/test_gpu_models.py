@@ -182,3 +182,31 @@ def test_fused_basic_block_matches_unfused():
         assert torch.allclose(
             rm_f, blk.conv1[1].running_mean, atol=1e-3, rtol=1e-3
         )
+
+
+@pytest.mark.gpu
+def test_fused_bottleneck_matches_unfused():
+    from pytorch_ddp_template_amd.models.resnet import Bottleneck
+
+    torch.manual_seed(0)
+    for in_ch, ch, stride in [(64, 16, 1), (64, 32, 2)]:
+        blk = Bottleneck(in_ch, ch, stride).to(torch.bfloat16).cuda()
+        blk.train()
+        x = torch.randn(8, 8, 8, in_ch, dtype=torch.bfloat16, device="cuda")
+        x1 = x.clone().requires_grad_(True)
+        x2 = x.clone().requires_grad_(True)
+        out_f = blk(x1)
+        out_f.float().square().mean().backward()
+        gf = [p.grad.clone() for p in blk.parameters()]
+        xf = x1.grad.clone()
+        for p in blk.parameters():
+            p.grad = None
+        for seq in [blk.conv1, blk.conv2, blk.conv3] + ([blk.down] if blk.down is not None else []):
+            seq[1].running_mean.zero_(); seq[1].running_var.fill_(1)
+        blk._can_fuse = lambda _x: False
+        out_u = blk(x2)
+        out_u.float().square().mean().backward()
+        assert torch.allclose(out_f.float(), out_u.float(), atol=3e-2, rtol=3e-2)
+        assert torch.allclose(xf.float(), x2.grad.float(), atol=3e-2, rtol=3e-2)
+        for a, p in zip(gf, blk.parameters()):
+            assert torch.allclose(a.float(), p.grad.float(), atol=5e-2, rtol=5e-2)
