@@ -81,7 +81,10 @@ enum { MODE_PLAIN = 0, MODE_CONV = 1 };
 //    written outputs (the BN stats pass then never re-reads the tensor);
 //  * addend (+addend_mask): C = acc + (mask>0 ? addend : 0) — fuses the
 //    residual skip-gradient add (and its ReLU mask) into conv dgrad.
-template <typename T16, int MODE, bool RELU, bool HAS_BIAS, int BNT = BN>
+// EXTRAS=false compiles the epilogue extras out entirely (they cost
+// registers/codegen even when the pointers are null).
+template <typename T16, int MODE, bool RELU, bool HAS_BIAS, int BNT = BN,
+          bool EXTRAS = false>
 __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     const T16* __restrict__ A, const T16* __restrict__ B,
     T16* __restrict__ C, const T16* __restrict__ bias,
@@ -277,7 +280,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     // issue back-to-back and overlap one memory latency, instead of
     // load->use->store serializing per element
     float av[4][4];
-    if (addend) {
+    if (EXTRAS && addend) {
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -298,16 +301,16 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
         if (row >= M) continue;
         float v = acc[mi][ni][r] + bv;
         if (RELU) v = fmaxf(v, 0.f);
-        if (addend) v += av[mi][r];
+        if (EXTRAS && addend) v += av[mi][r];
         C[(long long)row * N + col] = to_t<T16>(v);
-        if (stats_ws) {
+        if (EXTRAS && stats_ws) {
           col_sum[ni] += v;
           col_sq[ni] += v * v;
         }
       }
     }
   }
-  if (stats_ws) {
+  if (EXTRAS && stats_ws) {
     // barrier-free: reduce over the fs lanes (same col, different rows) via
     // two shuffles, then each wave PAIR (wave>>1 disambiguates waves that
     // share a column range but cover different row halves) stores its
@@ -718,6 +721,34 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
   // zero-page columns (ResNet layer1 Kout=64 fwd, C=64 dgrads).
   const bool narrow = N <= 64;
   if (narrow) grid.x = (N + 63) / 64;
+  const bool extras = ex.stats_ws != nullptr || ex.addend != nullptr;
+  if (extras) {
+    TORCH_CHECK(!relu && !bias_p,
+                "epilogue extras only instantiated for relu=false, bias=none");
+    if (narrow)
+      hipLaunchKernelGGL(
+          (g16::gemm_nt_bf16_kernel<t16, MODE, false, false, 64, true>), grid,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()),
+          reinterpret_cast<const t16*>(B.data_ptr()),
+          reinterpret_cast<t16*>(C.data_ptr()), nullptr,
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, N, K, sA, sB, sC,
+          cm, ex.stats_ws, ex.ws_nblocks,
+          reinterpret_cast<const t16*>(ex.addend),
+          reinterpret_cast<const t16*>(ex.addend_mask));
+    else
+      hipLaunchKernelGGL(
+          (g16::gemm_nt_bf16_kernel<t16, MODE, false, false, g16::BN, true>),
+          grid, dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()),
+          reinterpret_cast<const t16*>(B.data_ptr()),
+          reinterpret_cast<t16*>(C.data_ptr()), nullptr,
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, N, K, sA, sB, sC,
+          cm, ex.stats_ws, ex.ws_nblocks,
+          reinterpret_cast<const t16*>(ex.addend),
+          reinterpret_cast<const t16*>(ex.addend_mask));
+    return;
+  }
 #define LAUNCH_NT16(RELU, HB)                                                 \
   do {                                                                        \
     if (narrow)                                                               \

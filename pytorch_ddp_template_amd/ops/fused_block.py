@@ -72,12 +72,14 @@ class _FusedBasicBlockFn(torch.autograd.Function):
         stride = ctx.stride
         ext = native()
         dy = dy.contiguous()
-        dy2, out2 = _flat(dy), _flat(out)
         H, W = x.shape[1], x.shape[2]
         Hz, Wz = z.shape[1], z.shape[2]
 
-        # bn2 backward, ReLU mask (from the add_relu output) fused in
-        dx2, dg2, db2 = ext.bn_bwd(dy2, _flat(t2), g2, mean2, rstd2, out2)
+        # apply the add_relu mask ONCE; bn2/bnd backwards then read two
+        # tensors instead of three, and the skip grad is g itself
+        g = ext.relu_bwd(dy, out)
+        g2d = _flat(g)
+        dx2, dg2, db2 = ext.bn_bwd(g2d, _flat(t2), g2, mean2, rstd2, None)
         dx2 = dx2.reshape(t2.shape)
         # conv2 backward
         dz = ext.conv2d_dgrad(dx2, w2, 1, 1, Hz, Wz)
@@ -90,17 +92,14 @@ class _FusedBasicBlockFn(torch.autograd.Function):
         dw1 = ext.conv2d_wgrad(dt1, x, stride, 1, 3, 3).to(w1.dtype)
 
         if wd is None:
-            # identity skip: masked dy accumulates inside conv1's dgrad
-            dx = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W, dy, out)
+            dx = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W) + g
             dwd = dgd = dbd = None
         else:
             dxa = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W)
-            dtd, dgd, dbd = ext.bn_bwd(
-                dy2, _flat(td), gd, meand, rstdd, out2
-            )
+            dtd, dgd, dbd = ext.bn_bwd(g2d, _flat(td), gd, meand, rstdd, None)
             dtd = dtd.reshape(td.shape)
             dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
-            dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W, dxa)
+            dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W) + dxa
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dwd, dgd, dbd,
                 None, None, None, None)
 
@@ -165,12 +164,13 @@ class _FusedBottleneckFn(torch.autograd.Function):
         stride = ctx.stride
         ext = native()
         dy = dy.contiguous()
-        dy2, out2 = _flat(dy), _flat(out)
         H, W = x.shape[1], x.shape[2]
         H1, W1 = z1.shape[1], z1.shape[2]
         H2, W2 = z2.shape[1], z2.shape[2]
 
-        dx3, dg3, db3 = ext.bn_bwd(dy2, _flat(t3), g3, mean3, rstd3, out2)
+        g = ext.relu_bwd(dy, out)
+        g2d = _flat(g)
+        dx3, dg3, db3 = ext.bn_bwd(g2d, _flat(t3), g3, mean3, rstd3, None)
         dx3 = dx3.reshape(t3.shape)
         dz2 = ext.conv2d_dgrad(dx3, w3, 1, 0, H2, W2)
         dw3 = ext.conv2d_wgrad(dx3, z2, 1, 0, 1, 1).to(w3.dtype)
@@ -187,14 +187,14 @@ class _FusedBottleneckFn(torch.autograd.Function):
         dw1 = ext.conv2d_wgrad(dt1, x, 1, 0, 1, 1).to(w1.dtype)
 
         if wd is None:
-            dx = ext.conv2d_dgrad(dt1, w1, 1, 0, H, W, dy, out)
+            dx = ext.conv2d_dgrad(dt1, w1, 1, 0, H, W) + g
             dwd = dgd = dbd = None
         else:
             dxa = ext.conv2d_dgrad(dt1, w1, 1, 0, H, W)
-            dtd, dgd, dbd = ext.bn_bwd(dy2, _flat(td), gd, meand, rstdd, out2)
+            dtd, dgd, dbd = ext.bn_bwd(g2d, _flat(td), gd, meand, rstdd, None)
             dtd = dtd.reshape(td.shape)
             dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
-            dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W, dxa)
+            dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W) + dxa
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dw3, dg3, db3,
                 dwd, dgd, dbd, None, None, None, None, None)
 
