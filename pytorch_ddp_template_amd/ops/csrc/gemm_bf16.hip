@@ -511,11 +511,16 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 // MODE_PLAIN reuses the same machinery for the plain TN GEMM
 // C[I,J] += sum_m A[m,I] B[m,J] (linear / ViT weight grads): the B gather is
 // then just rows of [M,J] and TAPS must be 1.  J == Cin for conv mode.
+// Batched via grid.z = nbatch * zsplit (blockIdx.z / zsplit selects the
+// batch, % zsplit the split-M slice) — ViT attention backward runs ~1.5K
+// small TN GEMMs per call and a host-side per-batch launch loop was 70%%
+// of its step time.
 template <typename T16, int TAPS, int MODE = MODE_CONV>
 __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
     float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
-    int I /*Kout*/, int J, long long ldc, ConvMeta cm) {
+    int I /*Kout*/, int J, long long ldc, ConvMeta cm, long long sA,
+    long long sB, long long sC, int zsplit) {
   constexpr int BI = 64, BJ = 64, BMC = 32;
   constexpr int IMG = 32 * 16;   // elements per [32 m][16 ch] image
   constexpr int TILE = 4 * IMG;  // 4 images = one 64-channel tile
@@ -527,9 +532,14 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   const int Cin = J;
   const int i0 = blockIdx.y * BI;
   const int j0 = blockIdx.x * BJ;
+  const int batch = (int)blockIdx.z / zsplit;
+  const int zidx = (int)blockIdx.z % zsplit;
+  dy += (long long)batch * sA;
+  x += (long long)batch * sB;
+  dw += (long long)batch * sC;
   const int n_chunks = (Mtot + BMC - 1) / BMC;
-  const int per_z = (n_chunks + gridDim.z - 1) / gridDim.z;
-  const int ch0 = blockIdx.z * per_z;
+  const int per_z = (n_chunks + zsplit - 1) / zsplit;
+  const int ch0 = zidx * per_z;
   const int ch1 = min(n_chunks, ch0 + per_z);
   if (ch0 >= ch1) return;  // uniform per block
 
@@ -946,20 +956,20 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   int z = std::max(1, std::min(n_chunks, (768 + tiles - 1) / std::max(1, tiles)));
   g16::ConvMeta cm{};
   auto& zp = zero_page(A.device(), A.scalar_type());
+  // one launch for the whole batch: grid.z = bsz * split (ViT attention
+  // backward has ~1.5K batches; a per-batch host loop dominated its step)
+  if (bsz > 1) z = std::max(1, std::min(n_chunks, (int)(2048 / bsz) + 1));
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
-    for (long long b = 0; b < bsz; ++b) {
-      dim3 grid((J + 63) / 64, (I + 63) / 64, z);
-      // glds + tr-read TN path (same kernel as conv wgrad, plain B gather)
-      hipLaunchKernelGGL(
-          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN>), grid,
-          dim3(g16::THREADS), 0, stream,
-          reinterpret_cast<const t16*>(A.data_ptr()) + b * (long long)M * I,
-          reinterpret_cast<const t16*>(B.data_ptr()) + b * (long long)M * J,
-          C.data_ptr<float>() + b * (long long)I * J,
-          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J,
-          /*ldc=*/J, cm);
-    }
+    dim3 grid((J + 63) / 64, (I + 63) / 64, (unsigned)(bsz * z));
+    hipLaunchKernelGGL(
+        (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN>), grid,
+        dim3(g16::THREADS), 0, stream,
+        reinterpret_cast<const t16*>(A.data_ptr()),
+        reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
+        reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J,
+        /*ldc=*/J, cm, (long long)M * I, (long long)M * J, (long long)I * J,
+        z);
   };
   if (A.scalar_type() == torch::kBFloat16) run(bf16{});
   else run(_Float16{});
@@ -997,7 +1007,7 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                          reinterpret_cast<const t16*>(x.data_ptr()),
                          dw.data_ptr<float>(),
                          reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
-                         Cin, ldc, cm);
+                         Cin, ldc, cm, 0, 0, 0, (int)grid.z);
     } else if (R == 1 && S == 1) {
       hipLaunchKernelGGL((g16::gemm_wgrad_tr_kernel<t16, 1>), grid,
                          dim3(g16::THREADS), 0, stream,
@@ -1005,7 +1015,7 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                          reinterpret_cast<const t16*>(x.data_ptr()),
                          dw.data_ptr<float>(),
                          reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
-                         Cin, ldc, cm);
+                         Cin, ldc, cm, 0, 0, 0, (int)grid.z);
     } else {  // generic R x S: per-tap TN launches (cold path)
       for (int r = 0; r < (int)R; ++r)
         for (int s2 = 0; s2 < (int)S; ++s2)

@@ -410,8 +410,7 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                               const T* __restrict__ gamma,
                               const float* __restrict__ mean,
                               const float* __restrict__ rstd,
-                              T* __restrict__ dx, float* __restrict__ dgamma,
-                              float* __restrict__ dbeta, int D) {
+                              T* __restrict__ dx, int D) {
   __shared__ float lds[4];
   const long long m = blockIdx.x;
   const T* dyr = dy + m * D;
@@ -433,9 +432,58 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
     float g = to_f(dyr[d]) * to_f(gamma[d]);
     float xh = (to_f(xr[d]) - mu) * rs;
     dxr[d] = to_t<T>(rs * (g - invD * (s1 + xh * s2)));
-    // dgamma/dbeta accumulated per row into f32 buffers (atomics)
-    atomicAdd(&dgamma[d], to_f(dyr[d]) * xh);
-    atomicAdd(&dbeta[d], to_f(dyr[d]));
+  }
+}
+
+// ---- LN bwd params: dgamma/dbeta column sums -> per-block ws partials ----
+// Same coalesced 8-col-slot layout as bn_bwd_stats (the old per-row atomic
+// accumulation was ~38M global atomics onto a D-sized array per ViT layer);
+// per-row mean/rstd are scalar loads per row iteration.
+template <typename T>
+__global__ void ln_bwd_param_kernel(const T* __restrict__ dy,
+                                    const T* __restrict__ x,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ rstd,
+                                    float* __restrict__ ws, long long M,
+                                    int D) {
+  __shared__ float sdg[kBnMaxC];
+  __shared__ float sdb[kBnMaxC];
+  for (int c = threadIdx.x; c < D; c += blockDim.x) {
+    sdg[c] = 0.f;
+    sdb[c] = 0.f;
+  }
+  __syncthreads();
+  using IO = VecIO<T>;
+  const int slots = D / 8;
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const int c0 = (int)(tid % slots) * 8;
+  const long long row0 = tid / slots;
+  const long long rstride = nthreads / slots;
+  float dg[8] = {}, db[8] = {};
+  for (long long m = row0; m < M; m += rstride) {
+    const float mu = mean[m], rs = rstd[m];
+    const T* pg = dy + m * D + c0;
+    const T* px = x + m * D + c0;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float g = to_f(pg[j]);
+      const float xh = (to_f(px[j]) - mu) * rs;
+      db[j] += g;
+      dg[j] += g * xh;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&sdg[c0 + j], dg[j]);
+    atomicAdd(&sdb[c0 + j], db[j]);
+  }
+  __syncthreads();
+  float* wdg = ws + (long long)blockIdx.x * D;
+  float* wdb = ws + ((long long)gridDim.x + blockIdx.x) * D;
+  for (int c = threadIdx.x; c < D; c += blockDim.x) {
+    wdg[c] = sdg[c];
+    wdb[c] = sdb[c];
   }
 }
 
@@ -697,21 +745,41 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   long long M = x.size(0);
   int D = (int)x.size(1);
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto dgamma = torch::zeros({D}, f32);
-  auto dbeta = torch::zeros({D}, f32);
+  auto dgamma = torch::empty({D}, f32);
+  auto dbeta = torch::empty({D}, f32);
   auto dx = torch::empty_like(dy);
   auto stream = c10::hip::getCurrentHIPStream();
+  TORCH_CHECK(D % 8 == 0 && D <= 2048,
+              "ln bwd needs D % 8 == 0 and D <= 2048 (got ", D, ")");
+  dim3 sgrid(grid_1d(M * D / 8, 256, 1024));
+  auto ws = torch::empty({2 * (long long)sgrid.x, D}, f32);
+  const bool wide = gamma.scalar_type() == torch::kFloat32;
+  auto dgamma_out =
+      wide ? dgamma : torch::empty({D}, x.options().dtype(gamma.scalar_type()));
+  auto dbeta_out =
+      wide ? dbeta : torch::empty({D}, x.options().dtype(gamma.scalar_type()));
   DDP_DISPATCH_FLOAT(x.scalar_type(), "ln_bwd", [&] {
+    const auto* dyp = reinterpret_cast<const scalar_t*>(dy.data_ptr());
+    const auto* xp = reinterpret_cast<const scalar_t*>(x.data_ptr());
     hipLaunchKernelGGL((nrm::ln_bwd_kernel<scalar_t>), dim3((unsigned)M),
-                       dim3(256), 0, stream,
-                       reinterpret_cast<const scalar_t*>(dy.data_ptr()),
-                       reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                       dim3(256), 0, stream, dyp, xp,
                        reinterpret_cast<const scalar_t*>(gamma.data_ptr()),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       reinterpret_cast<scalar_t*>(dx.data_ptr()),
-                       dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), D);
+                       reinterpret_cast<scalar_t*>(dx.data_ptr()), D);
+    hipLaunchKernelGGL((nrm::ln_bwd_param_kernel<scalar_t>), sgrid, dim3(256),
+                       0, stream, dyp, xp, mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), ws.data_ptr<float>(), M, D);
+    hipLaunchKernelGGL((nrm::bn_bwd_finalize_kernel<scalar_t>), dim3(D),
+                       dim3(256), 0, stream, ws.data_ptr<float>(),
+                       (int)sgrid.x, dgamma.data_ptr<float>(),
+                       dbeta.data_ptr<float>(),
+                       wide ? nullptr
+                            : reinterpret_cast<scalar_t*>(dgamma_out.data_ptr()),
+                       wide ? nullptr
+                            : reinterpret_cast<scalar_t*>(dbeta_out.data_ptr()),
+                       D);
   });
-  return {dx, dgamma, dbeta};
+  return {dx, dgamma_out, dbeta_out};
 }
 
 torch::Tensor softmax_fwd(torch::Tensor x, double scale) {
