@@ -488,46 +488,44 @@ __global__ void ln_bwd_param_kernel(const T* __restrict__ dy,
 }
 
 // ---- row softmax fwd (pre-scale) ----
+// One WAVE per row, 4 rows per block (attention rows are short — S=197 —
+// so a 256-thread block per row left 3/4 of the block idle and serialized
+// on block-wide reductions; wave shuffles reduce in-register).
 template <typename T>
 __global__ void softmax_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
-                                   int D, float scale) {
-  __shared__ float lds[4];
-  const long long m = blockIdx.x;
+                                   long long M, int D, float scale) {
+  const long long m = (long long)blockIdx.x * 4 + wave_id();
+  if (m >= M) return;
+  const int lane = lane_id();
   const T* row = x + m * D;
   T* yrow = y + m * D;
   float mx = -INFINITY;
-  for (int d = threadIdx.x; d < D; d += blockDim.x)
-    mx = fmaxf(mx, to_f(row[d]) * scale);
-  // block max via wave reduce + lds
-  float wmax = wave_reduce_max(mx);
-  if (lane_id() == 0) lds[wave_id()] = wmax;
-  __syncthreads();
-  float bmax = fmaxf(fmaxf(lds[0], lds[1]), fmaxf(lds[2], lds[3]));
-  __syncthreads();
+  for (int d = lane; d < D; d += 64) mx = fmaxf(mx, to_f(row[d]) * scale);
+  mx = wave_reduce_max(mx);
   float s = 0.f;
-  for (int d = threadIdx.x; d < D; d += blockDim.x)
-    s += __expf(to_f(row[d]) * scale - bmax);
-  s = block_reduce_sum<256>(s, lds);
+  for (int d = lane; d < D; d += 64) s += __expf(to_f(row[d]) * scale - mx);
+  s = wave_reduce_sum(s);
   const float inv = 1.f / s;
-  for (int d = threadIdx.x; d < D; d += blockDim.x)
-    yrow[d] = to_t<T>(__expf(to_f(row[d]) * scale - bmax) * inv);
+  for (int d = lane; d < D; d += 64)
+    yrow[d] = to_t<T>(__expf(to_f(row[d]) * scale - mx) * inv);
 }
 
-// ---- row softmax bwd: dx = scale * y * (dy - sum(y*dy)) ----
+// ---- row softmax bwd: dx = scale * y * (dy - sum(dy*y)) ----
 template <typename T>
 __global__ void softmax_bwd_kernel(const T* __restrict__ dy,
-                                   const T* __restrict__ y, T* __restrict__ dx,
-                                   int D, float scale) {
-  __shared__ float lds[4];
-  const long long m = blockIdx.x;
+                                   const T* __restrict__ y,
+                                   T* __restrict__ dx, long long M, int D,
+                                   float scale) {
+  const long long m = (long long)blockIdx.x * 4 + wave_id();
+  if (m >= M) return;
+  const int lane = lane_id();
   const T* dyr = dy + m * D;
   const T* yr = y + m * D;
   T* dxr = dx + m * D;
   float dot = 0.f;
-  for (int d = threadIdx.x; d < D; d += blockDim.x)
-    dot += to_f(dyr[d]) * to_f(yr[d]);
-  dot = block_reduce_sum<256>(dot, lds);
-  for (int d = threadIdx.x; d < D; d += blockDim.x)
+  for (int d = lane; d < D; d += 64) dot += to_f(dyr[d]) * to_f(yr[d]);
+  dot = wave_reduce_sum(dot);
+  for (int d = lane; d < D; d += 64)
     dxr[d] = to_t<T>(scale * to_f(yr[d]) * (to_f(dyr[d]) - dot));
 }
 
@@ -788,10 +786,10 @@ torch::Tensor softmax_fwd(torch::Tensor x, double scale) {
   auto y = torch::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream();
   DDP_DISPATCH_FLOAT(x.scalar_type(), "softmax_fwd", [&] {
-    hipLaunchKernelGGL((nrm::softmax_fwd_kernel<scalar_t>), dim3((unsigned)M),
+    hipLaunchKernelGGL((nrm::softmax_fwd_kernel<scalar_t>), dim3((unsigned)((M + 3) / 4)),
                        dim3(256), 0, stream,
                        reinterpret_cast<const scalar_t*>(x.data_ptr()),
-                       reinterpret_cast<scalar_t*>(y.data_ptr()), D,
+                       reinterpret_cast<scalar_t*>(y.data_ptr()), M, D,
                        (float)scale);
   });
   return y;
@@ -803,11 +801,11 @@ torch::Tensor softmax_bwd(torch::Tensor dy, torch::Tensor y, double scale) {
   auto dx = torch::empty_like(dy);
   auto stream = c10::hip::getCurrentHIPStream();
   DDP_DISPATCH_FLOAT(y.scalar_type(), "softmax_bwd", [&] {
-    hipLaunchKernelGGL((nrm::softmax_bwd_kernel<scalar_t>), dim3((unsigned)M),
+    hipLaunchKernelGGL((nrm::softmax_bwd_kernel<scalar_t>), dim3((unsigned)((M + 3) / 4)),
                        dim3(256), 0, stream,
                        reinterpret_cast<const scalar_t*>(dy.data_ptr()),
                        reinterpret_cast<const scalar_t*>(y.data_ptr()),
-                       reinterpret_cast<scalar_t*>(dx.data_ptr()), D,
+                       reinterpret_cast<scalar_t*>(dx.data_ptr()), M, D,
                        (float)scale);
   });
   return dx;
