@@ -515,19 +515,25 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 // batch, % zsplit the split-M slice) — ViT attention backward runs ~1.5K
 // small TN GEMMs per call and a host-side per-batch launch loop was 70%%
 // of its step time.
-template <typename T16, int TAPS, int MODE = MODE_CONV>
+// TI/TJ: MFMA fragments per wave along I/J — 2 gives a 64x64 block tile
+// (the conv wgrad default; TAPS=9 already fills the accumulators), 4 gives
+// 128x128 (TAPS=1 large plain GEMMs, e.g. ViT linear weight grads, where
+// 64x64 tiles were instruction-bound at 4 MFMAs per staged chunk).
+template <typename T16, int TAPS, int MODE = MODE_CONV, int TI = 2,
+          int TJ = 2>
 __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
     float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
     int I /*Kout*/, int J, long long ldc, ConvMeta cm, long long sA,
     long long sB, long long sC, int zsplit) {
-  constexpr int BI = 64, BJ = 64, BMC = 32;
+  constexpr int BI = 32 * TI, BJ = 32 * TJ, BMC = 32;
   constexpr int IMG = 32 * 16;   // elements per [32 m][16 ch] image
-  constexpr int TILE = 4 * IMG;  // 4 images = one 64-channel tile
-  constexpr int NOPS = 1 + TAPS;
+  constexpr int AIMGS = BI / 16, BIMGS = BJ / 16;
+  constexpr int TILE_A = AIMGS * IMG, TILE_B = BIMGS * IMG;
+  constexpr int NUNITS = AIMGS + TAPS * BIMGS;  // glds image stages / chunk
   typedef short v4s __attribute__((ext_vector_type(4)));
   using vec16 = typename M16<T16>::vec;
-  __shared__ __attribute__((aligned(16))) T16 lds[2 * NOPS * TILE];
+  __shared__ __attribute__((aligned(16))) T16 lds[2 * (TILE_A + TAPS * TILE_B)];
 
   const int Cin = J;
   const int i0 = blockIdx.y * BI;
@@ -545,13 +551,13 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;  // wave tile 32x32
+  const int wm = (wave >> 1) * (BI / 2), wn = (wave & 1) * (BJ / 2);
   const int fr = lane & 15;
   const int fs = lane >> 4;
   const int sm = lane >> 1;        // staged m row this lane covers
   const int sh8 = (lane & 1) * 8;  // 8-channel half within the 32-B row
 
-  f32x4 acc[2][2][TAPS] = {};
+  f32x4 acc[TI][TJ][TAPS] = {};
 
   // ---- stage chunk ch into LDS buffer buf: 4*NOPS glds, no VALU pack ----
   auto stage = [&](int buf, int ch) {
@@ -567,30 +573,34 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
       hb = ho * cm.stride - cm.pad;
       wb = wo * cm.stride - cm.pad;
     }
-    T16* base = lds + buf * NOPS * TILE;
-    // wave w issues units u = w, w+4, ... of the 4*NOPS image stages
-    for (int u = wave; u < 4 * NOPS; u += 4) {
-      const int tile = u >> 2;  // 0 = dy, 1+tap = x
-      const int ig = u & 3;     // 16-channel group
+    T16* base = lds + buf * (TILE_A + TAPS * TILE_B);
+    // wave w issues units u = w, w+4, ... of the NUNITS image stages
+    for (int u = wave; u < NUNITS; u += 4) {
       const T16* src = zpad;
-      if (tile == 0) {
+      T16* dst;
+      if (u < AIMGS) {
+        const int ig = u;
         const int ii = i0 + ig * 16 + sh8;
         if (mok && ii < I) src = dy + (long long)gm * I + ii;
+        dst = base + ig * IMG;
       } else if (MODE == MODE_PLAIN) {
+        const int ig = u - AIMGS;  // TAPS == 1
         const int jj = j0 + ig * 16 + sh8;
         if (mok && jj < J) src = x + (long long)gm * J + jj;
+        dst = base + TILE_A + ig * IMG;
       } else {
-        const int tap = tile - 1;
+        const int tap = (u - AIMGS) / BIMGS;
+        const int ig = (u - AIMGS) % BIMGS;
         // TAPS==9 is only launched for 3x3 kernels, so the decode is
-        // compile-time (tap is a constant in the unrolled loop)
+        // compile-time when the loop unrolls
         const int r = TAPS == 1 ? 0 : tap / 3;
         const int s = TAPS == 1 ? 0 : tap % 3;
         const int hi = hb + r, wi = wb + s;
         const int jj = j0 + ig * 16 + sh8;
         if (mok && hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W && jj < Cin)
           src = x + (((long long)n * cm.H + hi) * cm.W + wi) * Cin + jj;
+        dst = base + TILE_A + (tap * BIMGS + ig) * IMG;
       }
-      T16* dst = base + tile * TILE + ig * IMG;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
@@ -611,15 +621,16 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
     const int buf = (ch - ch0) & 1;
     const bool more = ch + 1 < ch1;
     if (more) stage(buf ^ 1, ch + 1);  // glds latency hides under the MFMAs
-    const T16* base = lds + buf * NOPS * TILE;
+    const T16* base = lds + buf * (TILE_A + TAPS * TILE_B);
 
     // A fragments (dy): images wm/16 + mi
-    vec16 af[2];
-    {
+    vec16 af[TI];
+#pragma unroll
+    for (int mi = 0; mi < TI; mi += 2) {
       const unsigned a0 =
-          LDS_BYTE(base + ((wm >> 4) + 0) * IMG) + tr_lane_off;
+          LDS_BYTE(base + ((wm >> 4) + mi + 0) * IMG) + tr_lane_off;
       const unsigned a1 =
-          LDS_BYTE(base + ((wm >> 4) + 1) * IMG) + tr_lane_off;
+          LDS_BYTE(base + ((wm >> 4) + mi + 1) * IMG) + tr_lane_off;
       v4s l0, h0, l1, h1;
       // "=&v" (early-clobber) is load-bearing: without it LLVM may alias an
       // output pair with an address input, and since ds_read results land
@@ -633,36 +644,39 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
           "s_waitcnt lgkmcnt(0)"
           : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
           : "v"(a0), "v"(a1));
-      reinterpret_cast<v4s*>(&af[0])[0] = l0;
-      reinterpret_cast<v4s*>(&af[0])[1] = h0;
-      reinterpret_cast<v4s*>(&af[1])[0] = l1;
-      reinterpret_cast<v4s*>(&af[1])[1] = h1;
+      reinterpret_cast<v4s*>(&af[mi + 0])[0] = l0;
+      reinterpret_cast<v4s*>(&af[mi + 0])[1] = h0;
+      reinterpret_cast<v4s*>(&af[mi + 1])[0] = l1;
+      reinterpret_cast<v4s*>(&af[mi + 1])[1] = h1;
     }
 #pragma unroll
     for (int tap = 0; tap < TAPS; ++tap) {
-      const T16* tb = base + (1 + tap) * TILE;
-      const unsigned b0 =
-          LDS_BYTE(tb + ((wn >> 4) + 0) * IMG) + tr_lane_off;
-      const unsigned b1 =
-          LDS_BYTE(tb + ((wn >> 4) + 1) * IMG) + tr_lane_off;
-      vec16 bfr[2];
-      v4s l0, h0, l1, h1;
-      asm volatile(
-          "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
-          "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
-          "ds_read_b64_tr_b16 %2, %5 offset:0\n\t"
-          "ds_read_b64_tr_b16 %3, %5 offset:128\n\t"
-          "s_waitcnt lgkmcnt(0)"
-          : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
-          : "v"(b0), "v"(b1));
-      reinterpret_cast<v4s*>(&bfr[0])[0] = l0;
-      reinterpret_cast<v4s*>(&bfr[0])[1] = h0;
-      reinterpret_cast<v4s*>(&bfr[1])[0] = l1;
-      reinterpret_cast<v4s*>(&bfr[1])[1] = h1;
+      const T16* tb = base + TILE_A + tap * TILE_B;
+      vec16 bfr[TJ];
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
+      for (int ni = 0; ni < TJ; ni += 2) {
+        const unsigned b0 =
+            LDS_BYTE(tb + ((wn >> 4) + ni + 0) * IMG) + tr_lane_off;
+        const unsigned b1 =
+            LDS_BYTE(tb + ((wn >> 4) + ni + 1) * IMG) + tr_lane_off;
+        v4s l0, h0, l1, h1;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
+            "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
+            "ds_read_b64_tr_b16 %2, %5 offset:0\n\t"
+            "ds_read_b64_tr_b16 %3, %5 offset:128\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
+            : "v"(b0), "v"(b1));
+        reinterpret_cast<v4s*>(&bfr[ni + 0])[0] = l0;
+        reinterpret_cast<v4s*>(&bfr[ni + 0])[1] = h0;
+        reinterpret_cast<v4s*>(&bfr[ni + 1])[0] = l1;
+        reinterpret_cast<v4s*>(&bfr[ni + 1])[1] = h1;
+      }
 #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
+      for (int mi = 0; mi < TI; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < TJ; ++ni)
           acc[mi][ni][tap] = M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni][tap]);
     }
     if (more) __syncthreads();  // drains the glds (vmcnt) + publishes buf^1
@@ -676,11 +690,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   for (int tap = 0; tap < TAPS; ++tap) {
     const long long coff = (long long)tap * Cin;
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) {
+    for (int ni = 0; ni < TJ; ++ni) {
       const int col = j0 + wn + ni * 16 + fr;
       if (col >= Cin) continue;
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
+      for (int mi = 0; mi < TI; ++mi)
 #pragma unroll
         for (int rr = 0; rr < 4; ++rr) {
           const int row = i0 + wm + mi * 16 + fs * 4 + rr;
@@ -959,8 +973,26 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   // one launch for the whole batch: grid.z = bsz * split (ViT attention
   // backward has ~1.5K batches; a per-batch host loop dominated its step)
   if (bsz > 1) z = std::max(1, std::min(n_chunks, (int)(2048 / bsz) + 1));
+  const bool wide = I >= 128 && J >= 128;  // 128x128 tiles for big GEMMs
+  if (wide) {
+    tiles = ((J + 127) / 128) * ((I + 127) / 128);
+    z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
+    if (bsz > 1) z = std::max(1, std::min(n_chunks, (int)(2048 / bsz) + 1));
+  }
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
+    if (wide) {
+      dim3 grid((J + 127) / 128, (I + 127) / 128, (unsigned)(bsz * z));
+      hipLaunchKernelGGL(
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4>), grid,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()),
+          reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J,
+          /*ldc=*/J, cm, (long long)M * I, (long long)M * J,
+          (long long)I * J, z);
+      return;
+    }
     dim3 grid((J + 63) / 64, (I + 63) / 64, (unsigned)(bsz * z));
     hipLaunchKernelGGL(
         (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN>), grid,
