@@ -19,9 +19,28 @@ BASELINE.json ResNet configs).
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from .native import native
+
+# Weight-gradient GEMMs and BN param-grad reductions are side outputs of the
+# backward chain; running them on a second HIP stream lets their (issue- and
+# LDS-bound) kernels co-reside with the (bandwidth-bound) dgrad/BN-dx chain.
+# PDT_FUSED_STREAMS=0 disables.
+_side = None
+
+
+def _side_stream():
+    global _side
+    if _side is None:
+        _side = torch.cuda.Stream()
+    return _side
+
+
+def _use_streams():
+    return os.environ.get("PDT_FUSED_STREAMS", "1") != "0"
 
 
 def _flat(t):
@@ -81,15 +100,28 @@ class _FusedBasicBlockFn(torch.autograd.Function):
         g2d = _flat(g)
         dx2, dg2, db2 = ext.bn_bwd(g2d, _flat(t2), g2, mean2, rstd2, None)
         dx2 = dx2.reshape(t2.shape)
-        # conv2 backward
+        use_side = _use_streams()
+        main = torch.cuda.current_stream()
+        side = _side_stream() if use_side else None
+        if use_side:
+            side.wait_stream(main)
+            with torch.cuda.stream(side):
+                dw2 = ext.conv2d_wgrad(dx2, z, 1, 1, 3, 3).to(w2.dtype)
+        # conv2 backward dgrad stays on the critical path
         dz = ext.conv2d_dgrad(dx2, w2, 1, 1, Hz, Wz)
-        dw2 = ext.conv2d_wgrad(dx2, z, 1, 1, 3, 3).to(w2.dtype)
+        if not use_side:
+            dw2 = ext.conv2d_wgrad(dx2, z, 1, 1, 3, 3).to(w2.dtype)
         # bn1 backward (its own fused ReLU: z is bn1's relu output)
         dt1, dg1, db1 = ext.bn_bwd(
             _flat(dz), _flat(t1), g1, mean1, rstd1, _flat(z)
         )
         dt1 = dt1.reshape(t1.shape)
-        dw1 = ext.conv2d_wgrad(dt1, x, stride, 1, 3, 3).to(w1.dtype)
+        if use_side:
+            side.wait_stream(main)
+            with torch.cuda.stream(side):
+                dw1 = ext.conv2d_wgrad(dt1, x, stride, 1, 3, 3).to(w1.dtype)
+        else:
+            dw1 = ext.conv2d_wgrad(dt1, x, stride, 1, 3, 3).to(w1.dtype)
 
         if wd is None:
             dx = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W) + g
@@ -98,8 +130,19 @@ class _FusedBasicBlockFn(torch.autograd.Function):
             dxa = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W)
             dtd, dgd, dbd = ext.bn_bwd(g2d, _flat(td), gd, meand, rstdd, None)
             dtd = dtd.reshape(td.shape)
-            dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
+            if use_side:
+                side.wait_stream(main)
+                with torch.cuda.stream(side):
+                    dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
+            else:
+                dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
             dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W) + dxa
+        if use_side:
+            # weight grads are consumed (bucket accumulation) on the main
+            # stream after this node returns
+            main.wait_stream(side)
+            for t in (dw2, dw1) + ((dwd,) if wd is not None else ()):
+                t.record_stream(main)
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dwd, dgd, dbd,
                 None, None, None, None)
 
