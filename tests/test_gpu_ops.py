@@ -78,7 +78,8 @@ def test_gemm_nt_f32(m, n, k):
     close_f32(out, a @ b.t(), tol=1e-4)
 
 
-@pytest.mark.parametrize("m,i,j", [(256, 64, 64), (1000, 70, 33), (64, 10, 512)])
+@pytest.mark.parametrize("m,i,j", [(256, 64, 64), (1000, 70, 33), (64, 10, 512),
+                                   (512, 256, 384), (2048, 768, 768)])
 def test_gemm_tn_bf16(m, i, j):
     a = t32(m, i, seed=8).to(torch.bfloat16)
     b = t32(m, j, seed=9).to(torch.bfloat16)
