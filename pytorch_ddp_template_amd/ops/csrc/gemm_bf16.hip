@@ -35,6 +35,7 @@ namespace g16 {
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
 
 // 16-bit dtype traits: fragment vector type + the matching MFMA intrinsic.
 template <typename T16>
@@ -45,12 +46,22 @@ struct M16<bf16> {
   DEV_INLINE static f32x4 mma(vec a, vec b, f32x4 c) {
     return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
   }
+  // 32x32x16 is gfx950's full-rate bf16 shape (~2.5 PF vs ~2.0 for
+  // 16x16x32) and does 2x the FLOPs per instruction — issue-bound kernels
+  // want it.  A/B: lane holds 8 k at k = (lane>>5)*8, row/col = lane&31.
+  // C/D: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
+  DEV_INLINE static f32x16 mma32(vec a, vec b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+  }
 };
 template <>
 struct M16<_Float16> {
   using vec = f16x8;
   DEV_INLINE static f32x4 mma(vec a, vec b, f32x4 c) {
     return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+  DEV_INLINE static f32x16 mma32(vec a, vec b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_f16(a, b, c, 0, 0, 0);
   }
 };
 
@@ -211,11 +222,12 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     }
   };
 
-  // ----- main loop -----
-  f32x4 acc[4][NI] = {};
+  // ----- main loop (32x32x16 MFMA: full-rate shape, half the instrs) -----
+  constexpr int NI32 = BNT / 64;  // 32-wide B tiles per wave
+  f32x16 acc[2][NI32] = {};
   const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BNT / 2);
-  const int fr = lane & 15;   // fragment row (A) / col (B, D)
-  const int fs = lane >> 4;   // k-slot
+  const int r32 = lane & 31;  // row (A) / col (B, D) within a 32-tile
+  const int ks = lane >> 5;   // k half-slot (8 halfwords)
 
   // per-wave glds count of one stage() call (the vmcnt budget at barriers)
   const bool light = (BNT == 64) && (wave >= 2);  // no B chunks to stage
@@ -243,24 +255,28 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     const char* baseA = &smem[buf * BUF_BYTES];
     const char* baseB = baseA + TILE_BYTES;
     using vec16 = typename M16<T16>::vec;
-    vec16 af[4], bf[NI];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int ra = wm + i * 16 + fr;
-      af[i] = *reinterpret_cast<const vec16*>(
-          baseA + ra * 64 + kswz(ra, fs) * 16);
+    for (int kh = 0; kh < 2; ++kh) {  // two k-16 halves of the BK=32 step
+      const int chunk = kh * 2 + ks;
+      vec16 af[2], bf[NI32];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int ra = wm + i * 32 + r32;
+        af[i] = *reinterpret_cast<const vec16*>(
+            baseA + ra * 64 + kswz(ra, chunk) * 16);
+      }
+#pragma unroll
+      for (int i = 0; i < NI32; ++i) {
+        const int rb = wn + i * 32 + r32;
+        bf[i] = *reinterpret_cast<const vec16*>(
+            baseB + rb * 64 + kswz(rb, chunk) * 16);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NI32; ++ni)
+          acc[mi][ni] = M16<T16>::mma32(af[mi], bf[ni], acc[mi][ni]);
     }
-#pragma unroll
-    for (int i = 0; i < NI; ++i) {
-      const int rb = wn + i * 16 + fr;
-      bf[i] = *reinterpret_cast<const vec16*>(
-          baseB + rb * 64 + kswz(rb, fs) * 16);
-    }
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-      for (int ni = 0; ni < NI; ++ni)
-        acc[mi][ni] = M16<T16>::mma(af[mi], bf[ni], acc[mi][ni]);
     if (kt + 1 < KT) {
       if (kt + 2 < KT) NT_WAIT_STAGE(); else NT_WAIT_ALL();
       asm volatile("s_barrier" ::: "memory");
@@ -270,38 +286,39 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
 #undef NT_WAIT_ALL
 
   // ----- epilogue: bias + relu (+addend) + bf16 store (+BN stat partials) --
-  float col_sum[NI] = {}, col_sq[NI] = {};
+  // C/D layout of 32x32x16: col = lane&31, row = (reg&3)+8*(reg>>2)+4*ks
+  float col_sum[NI32] = {}, col_sq[NI32] = {};
 #pragma unroll
-  for (int ni = 0; ni < NI; ++ni) {
-    const int col = n0 + wn + ni * 16 + fr;
+  for (int ni = 0; ni < NI32; ++ni) {
+    const int col = n0 + wn + ni * 32 + r32;
     if (col >= N) continue;
     float bv = HAS_BIAS ? to_f(bias[col]) : 0.f;
-    // gather the residual addend (+mask) up front: 16 independent loads
-    // issue back-to-back and overlap one memory latency, instead of
-    // load->use->store serializing per element
-    float av[4][4];
+    // gather the residual addend (+mask) up front: independent loads issue
+    // back-to-back and overlap one memory latency
+    float av[2][16];
     if (EXTRAS && addend) {
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = m0 + wm + mi * 16 + fs * 4 + r;
+        for (int reg = 0; reg < 16; ++reg) {
+          const int row =
+              m0 + wm + mi * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * ks;
           const long long idx = (long long)row * N + col;
-          av[mi][r] = row < M ? to_f(addend[idx]) : 0.f;
+          av[mi][reg] = row < M ? to_f(addend[idx]) : 0.f;
           if (addend_mask && row < M && !(to_f(addend_mask[idx]) > 0.f))
-            av[mi][r] = 0.f;
+            av[mi][reg] = 0.f;
         }
     }
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
-      const int row_base = m0 + wm + mi * 16 + fs * 4;
+    for (int mi = 0; mi < 2; ++mi) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = row_base + r;
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row =
+            m0 + wm + mi * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * ks;
         if (row >= M) continue;
-        float v = acc[mi][ni][r] + bv;
+        float v = acc[mi][ni][reg] + bv;
         if (RELU) v = fmaxf(v, 0.f);
-        if (EXTRAS && addend) v += av[mi][r];
+        if (EXTRAS && addend) v += av[mi][reg];
         C[(long long)row * N + col] = to_t<T16>(v);
         if (EXTRAS && stats_ws) {
           col_sum[ni] += v;
@@ -311,22 +328,18 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     }
   }
   if (EXTRAS && stats_ws) {
-    // barrier-free: reduce over the fs lanes (same col, different rows) via
-    // two shuffles, then each wave PAIR (wave>>1 disambiguates waves that
-    // share a column range but cover different row halves) stores its
-    // 32-col slice straight into its own workspace row.  ws therefore has
-    // 2 rows per block (ws_nblocks = 2 * grid blocks); the finalize pass
-    // reduces rows regardless.
+    // barrier-free: lanes l and l^32 share a column (ks differs) — one
+    // shuffle reduces them; each wave PAIR (wave>>1 disambiguates waves
+    // sharing a column range at different row halves) stores its slice into
+    // its own workspace row (ws_nblocks = 2 * grid blocks).
     const long long wsrow = (long long)flat_id * 2 + (wave >> 1);
 #pragma unroll
-    for (int ni = 0; ni < NI; ++ni) {
+    for (int ni = 0; ni < NI32; ++ni) {
       float sv = col_sum[ni], qv = col_sq[ni];
-      sv += __shfl_xor(sv, 16);
       sv += __shfl_xor(sv, 32);
-      qv += __shfl_xor(qv, 16);
       qv += __shfl_xor(qv, 32);
-      const int col = n0 + wn + ni * 16 + fr;
-      if (fs == 0 && col < N) {
+      const int col = n0 + wn + ni * 32 + r32;
+      if (ks == 0 && col < N) {
         stats_ws[wsrow * N + col] = sv;
         stats_ws[((long long)ws_nblocks + wsrow) * N + col] = qv;
       }
