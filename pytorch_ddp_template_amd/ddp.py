@@ -375,10 +375,21 @@ def train(args, model):
     tr_loss_total = 0.0
     log_count = 0
 
-    for epoch in range(int(args.num_train_epochs)):
+    # epoch/step progress bars on the main rank only (reference ddp.py:212,
+    # 215; disabled elsewhere).  The loss postfix updates at logging
+    # boundaries, not per step — the reference's per-step set_postfix forced
+    # a device sync every iteration (ddp.py:232).
+    from tqdm import tqdm, trange
+
+    show_bars = is_main_process() and not args.no_progress_bar
+    epoch_iter = trange(
+        int(args.num_train_epochs), desc="Epoch", disable=not show_bars
+    )
+    for epoch in epoch_iter:
         if isinstance(train_sampler, ShardedSampler):
             train_sampler.set_epoch(epoch)  # reference ddp.py:213-214
-        for step, (x, y) in enumerate(loader):
+        step_iter = tqdm(loader, desc="Iteration", disable=not show_bars)
+        for step, (x, y) in enumerate(step_iter):
             x = x.to(args.device, non_blocking=True)
             y = y.to(args.device, non_blocking=True)
             if x.dtype.is_floating_point:
@@ -446,6 +457,8 @@ def train(args, model):
                         dict(step=global_step, loss=window,
                              lr=scheduler.get_last_lr()[0]),
                     )
+                    if show_bars:
+                        step_iter.set_postfix(loss=window)
 
                 if (
                     is_main_process()
@@ -510,6 +523,7 @@ def build_parser():
     parser.add_argument("--find_unused_parameters", action="store_true")
     parser.add_argument("--num_workers", type=int, default=0)
     parser.add_argument("--no_tensorboard", action="store_true")
+    parser.add_argument("--no_progress_bar", action="store_true")
     return parser
 
 

@@ -570,7 +570,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   const int sm = lane >> 1;        // staged m row this lane covers
   const int sh8 = (lane & 1) * 8;  // 8-channel half within the 32-B row
 
-  f32x4 acc[TI][TJ][TAPS] = {};
+  f32x16 acc[TI / 2][TJ / 2][TAPS] = {};  // 32x32 tiles
 
   // ---- stage chunk ch into LDS buffer buf: 4*NOPS glds, no VALU pack ----
   auto stage = [&](int buf, int ch) {
@@ -620,9 +620,14 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
     }
   };
 
-  // per-lane byte offset of this lane's tr-read address within an image
+  // tr-read addressing for 32x32x16 fragments: lane L covers column
+  // (L&31) -> image (L>>4)&1 at in-image col L&15; k half-slot ks = L>>5
+  // selects the 8-m window.  Two reads (offset 0/+128B) give m +0..7; the
+  // second MFMA of a 32-m chunk reads +512B (16 rows further).
+  const int ks = lane >> 5;
   const unsigned tr_lane_off =
-      (unsigned)((fs * 8 + ((lane & 15) >> 2)) * 32 + (lane & 3) * 8);
+      (unsigned)((ks * 8 + ((lane & 15) >> 2)) * 32 + (lane & 3) * 8);
+  const int img_sel = (lane >> 4) & 1;  // which 16-ch image of a 32-col pair
 #define LDS_BYTE(p)                                           \
   ((unsigned)(unsigned long long)(__attribute__((            \
       address_space(3))) const T16*)(p))
@@ -630,20 +635,19 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   stage(0, ch0);
   __syncthreads();
 
+  constexpr int TI32 = TI / 2, TJ32 = TJ / 2;
   for (int ch = ch0; ch < ch1; ++ch) {
     const int buf = (ch - ch0) & 1;
     const bool more = ch + 1 < ch1;
     if (more) stage(buf ^ 1, ch + 1);  // glds latency hides under the MFMAs
     const T16* base = lds + buf * (TILE_A + TAPS * TILE_B);
 
-    // A fragments (dy): images wm/16 + mi
-    vec16 af[TI];
+    // A fragments (dy): per 32-col tile mi, per 16-m half kh
+    vec16 af[TI32][2];
 #pragma unroll
-    for (int mi = 0; mi < TI; mi += 2) {
-      const unsigned a0 =
-          LDS_BYTE(base + ((wm >> 4) + mi + 0) * IMG) + tr_lane_off;
-      const unsigned a1 =
-          LDS_BYTE(base + ((wm >> 4) + mi + 1) * IMG) + tr_lane_off;
+    for (int mi = 0; mi < TI32; ++mi) {
+      const unsigned a0 = LDS_BYTE(base + ((wm >> 4) + mi * 2 + img_sel) * IMG) +
+                          tr_lane_off;
       v4s l0, h0, l1, h1;
       // "=&v" (early-clobber) is load-bearing: without it LLVM may alias an
       // output pair with an address input, and since ds_read results land
@@ -652,68 +656,69 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
       asm volatile(
           "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
           "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
-          "ds_read_b64_tr_b16 %2, %5 offset:0\n\t"
-          "ds_read_b64_tr_b16 %3, %5 offset:128\n\t"
+          "ds_read_b64_tr_b16 %2, %4 offset:512\n\t"
+          "ds_read_b64_tr_b16 %3, %4 offset:640\n\t"
           "s_waitcnt lgkmcnt(0)"
           : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
-          : "v"(a0), "v"(a1));
-      reinterpret_cast<v4s*>(&af[mi + 0])[0] = l0;
-      reinterpret_cast<v4s*>(&af[mi + 0])[1] = h0;
-      reinterpret_cast<v4s*>(&af[mi + 1])[0] = l1;
-      reinterpret_cast<v4s*>(&af[mi + 1])[1] = h1;
+          : "v"(a0));
+      reinterpret_cast<v4s*>(&af[mi][0])[0] = l0;
+      reinterpret_cast<v4s*>(&af[mi][0])[1] = h0;
+      reinterpret_cast<v4s*>(&af[mi][1])[0] = l1;
+      reinterpret_cast<v4s*>(&af[mi][1])[1] = h1;
     }
 #pragma unroll
     for (int tap = 0; tap < TAPS; ++tap) {
       const T16* tb = base + TILE_A + tap * TILE_B;
-      vec16 bfr[TJ];
 #pragma unroll
-      for (int ni = 0; ni < TJ; ni += 2) {
+      for (int ni = 0; ni < TJ32; ++ni) {
         const unsigned b0 =
-            LDS_BYTE(tb + ((wn >> 4) + ni + 0) * IMG) + tr_lane_off;
-        const unsigned b1 =
-            LDS_BYTE(tb + ((wn >> 4) + ni + 1) * IMG) + tr_lane_off;
+            LDS_BYTE(tb + ((wn >> 4) + ni * 2 + img_sel) * IMG) + tr_lane_off;
         v4s l0, h0, l1, h1;
         asm volatile(
             "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
             "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
-            "ds_read_b64_tr_b16 %2, %5 offset:0\n\t"
-            "ds_read_b64_tr_b16 %3, %5 offset:128\n\t"
+            "ds_read_b64_tr_b16 %2, %4 offset:512\n\t"
+            "ds_read_b64_tr_b16 %3, %4 offset:640\n\t"
             "s_waitcnt lgkmcnt(0)"
             : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
-            : "v"(b0), "v"(b1));
-        reinterpret_cast<v4s*>(&bfr[ni + 0])[0] = l0;
-        reinterpret_cast<v4s*>(&bfr[ni + 0])[1] = h0;
-        reinterpret_cast<v4s*>(&bfr[ni + 1])[0] = l1;
-        reinterpret_cast<v4s*>(&bfr[ni + 1])[1] = h1;
+            : "v"(b0));
+        vec16 bf0, bf1;
+        reinterpret_cast<v4s*>(&bf0)[0] = l0;
+        reinterpret_cast<v4s*>(&bf0)[1] = h0;
+        reinterpret_cast<v4s*>(&bf1)[0] = l1;
+        reinterpret_cast<v4s*>(&bf1)[1] = h1;
+#pragma unroll
+        for (int mi = 0; mi < TI32; ++mi) {
+          acc[mi][ni][tap] = M16<T16>::mma32(af[mi][0], bf0, acc[mi][ni][tap]);
+          acc[mi][ni][tap] = M16<T16>::mma32(af[mi][1], bf1, acc[mi][ni][tap]);
+        }
       }
-#pragma unroll
-      for (int mi = 0; mi < TI; ++mi)
-#pragma unroll
-        for (int ni = 0; ni < TJ; ++ni)
-          acc[mi][ni][tap] = M16<T16>::mma(af[mi], bfr[ni], acc[mi][ni][tap]);
     }
     if (more) __syncthreads();  // drains the glds (vmcnt) + publishes buf^1
   }
+
 #undef LDS_BYTE
 
   // ---- writeback: f32 atomics into dw[I][taps][Cin] slices ----
   // (static tap order: a runtime-rotated index into acc would demote the
   // accumulators to scratch)
+  // C/D layout of 32x32x16: col = lane&31, row = (reg&3)+8*(reg>>2)+4*ks
 #pragma unroll
   for (int tap = 0; tap < TAPS; ++tap) {
     const long long coff = (long long)tap * Cin;
 #pragma unroll
-    for (int ni = 0; ni < TJ; ++ni) {
-      const int col = j0 + wn + ni * 16 + fr;
+    for (int ni = 0; ni < TJ / 2; ++ni) {
+      const int col = j0 + wn + ni * 32 + (lane & 31);
       if (col >= Cin) continue;
 #pragma unroll
-      for (int mi = 0; mi < TI; ++mi)
+      for (int mi = 0; mi < TI / 2; ++mi)
 #pragma unroll
-        for (int rr = 0; rr < 4; ++rr) {
-          const int row = i0 + wm + mi * 16 + fs * 4 + rr;
+        for (int reg = 0; reg < 16; ++reg) {
+          const int row =
+              i0 + wm + mi * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * ks;
           if (row >= I) continue;
           atomicAdd(&dw[(long long)row * ldc + coff + col],
-                    acc[mi][ni][tap][rr]);
+                    acc[mi][ni][tap][reg]);
         }
     }
   }
