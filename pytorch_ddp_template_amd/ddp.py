@@ -371,6 +371,11 @@ def train(args, model):
     device_loss = torch.zeros((), device=args.device)
     logged_loss = 0.0
     steps_since_log = 0
+    # per-step wall timers (SURVEY.md §5.1 — the reference had none): window
+    # throughput measured between logging boundaries, no extra device syncs
+    import time as _time
+
+    window_t0 = _time.perf_counter()
     done = False
     tr_loss_total = 0.0
     log_count = 0
@@ -441,6 +446,14 @@ def train(args, model):
                     and global_step % args.logging_steps == 0
                 ):
                     # ONE readback per logging window (not per step)
+                    now = _time.perf_counter()
+                    win_steps = steps_since_log
+                    ms_per_step = (now - window_t0) / max(1, win_steps) * 1e3
+                    sps = (
+                        args.train_batch_size * max(1, args.world_size)
+                        * win_steps / max(1e-9, now - window_t0)
+                    )
+                    window_t0 = now
                     window = float(device_loss) / max(1, steps_since_log)
                     tr_loss_total += float(device_loss)
                     log_count += steps_since_log
@@ -452,10 +465,15 @@ def train(args, model):
                             "lr", scheduler.get_last_lr()[0], global_step
                         )
                         tb_writer.add_scalar("loss", window, global_step)
+                        tb_writer.add_scalar(
+                            "samples_per_sec", sps, global_step
+                        )
                     logger.info(
                         "train step",
                         dict(step=global_step, loss=window,
-                             lr=scheduler.get_last_lr()[0]),
+                             lr=scheduler.get_last_lr()[0],
+                             ms_per_step=round(ms_per_step, 2),
+                             samples_per_sec=round(sps, 1)),
                     )
                     if show_bars:
                         step_iter.set_postfix(loss=window)
