@@ -538,7 +538,8 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
     float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
     int I /*Kout*/, int J, long long ldc, ConvMeta cm, long long sA,
-    long long sB, long long sC, int zsplit) {
+    long long sB, long long sC, int zsplit,
+    float* __restrict__ colsum = nullptr /* f32 [I]: bias grad of dy */) {
   constexpr int BI = 32 * TI, BJ = 32 * TJ, BMC = 32;
   constexpr int IMG = 32 * 16;   // elements per [32 m][16 ch] image
   constexpr int AIMGS = BI / 16, BIMGS = BJ / 16;
@@ -636,6 +637,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   __syncthreads();
 
   constexpr int TI32 = TI / 2, TJ32 = TJ / 2;
+  float bsum[TI32] = {};
   for (int ch = ch0; ch < ch1; ++ch) {
     const int buf = (ch - ch0) & 1;
     const bool more = ch + 1 < ch1;
@@ -665,6 +667,19 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
       reinterpret_cast<v4s*>(&af[mi][0])[1] = h0;
       reinterpret_cast<v4s*>(&af[mi][1])[0] = l1;
       reinterpret_cast<v4s*>(&af[mi][1])[1] = h1;
+      if (colsum && (wave & 1) == 0) {
+        // bias grad rides along: the staged dy tile is already in
+        // registers; out-of-range m/i lanes were zero-page staged.
+        // Only wn==0 waves contribute (wn pairs re-read the same A tile).
+        float a = 0.f;
+#pragma unroll
+        for (int kh = 0; kh < 2; ++kh) {
+          const vec16 v = af[mi][kh];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) a += (float)v[j];
+        }
+        bsum[mi] += a;
+      }
     }
 #pragma unroll
     for (int tap = 0; tap < TAPS; ++tap) {
@@ -702,6 +717,15 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   // ---- writeback: f32 atomics into dw[I][taps][Cin] slices ----
   // (static tap order: a runtime-rotated index into acc would demote the
   // accumulators to scratch)
+  if (colsum && (wave & 1) == 0) {
+#pragma unroll
+    for (int mi = 0; mi < TI32; ++mi) {
+      float v = bsum[mi] + __shfl_xor(bsum[mi], 32);  // join the ks pair
+      const int col = i0 + wm + mi * 32 + (lane & 31);
+      if (ks == 0 && col < I) atomicAdd(&colsum[col], v);
+    }
+  }
+
   // C/D layout of 32x32x16: col = lane&31, row = (reg&3)+8*(reg>>2)+4*ks
 #pragma unroll
   for (int tap = 0; tap < TAPS; ++tap) {
@@ -1024,6 +1048,53 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   if (A.scalar_type() == torch::kBFloat16) run(bf16{});
   else run(_Float16{});
   return C;
+}
+
+// TN GEMM + bias grad in one pass: returns {C=f32[I,J], db=f32[I]} where
+// db = column sums of A (the dy operand) — the kernel already stages every
+// dy tile, so the separate col_sum pass over dy disappears (linear layers'
+// backward).  2-D only.
+std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.dim() == 2 && B.dim() == 2);
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  int M = (int)A.size(0), I = (int)A.size(1);
+  int J = (int)B.size(1);
+  TORCH_CHECK((int)B.size(0) == M);
+  auto C = torch::zeros({I, J}, A.options().dtype(torch::kFloat32));
+  auto db = torch::zeros({I}, A.options().dtype(torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream();
+  int n_chunks = (M + 31) / 32;
+  g16::ConvMeta cm{};
+  auto& zp = zero_page(A.device(), A.scalar_type());
+  const bool wide = I >= 128 && J >= 128;
+  int tiles = wide ? ((J + 127) / 128) * ((I + 127) / 128)
+                   : ((J + 63) / 64) * ((I + 63) / 64);
+  int z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
+  auto run = [&](auto tag) {
+    using t16 = decltype(tag);
+    if (wide) {
+      dim3 grid((J + 127) / 128, (I + 127) / 128, (unsigned)z);
+      hipLaunchKernelGGL(
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4>), grid,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()),
+          reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J, J, cm, 0, 0,
+          0, z, db.data_ptr<float>());
+    } else {
+      dim3 grid((J + 63) / 64, (I + 63) / 64, (unsigned)z);
+      hipLaunchKernelGGL(
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN>), grid,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()),
+          reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J, J, cm, 0, 0,
+          0, z, db.data_ptr<float>());
+    }
+  };
+  if (A.scalar_type() == torch::kBFloat16) run(bf16{});
+  else run(_Float16{});
+  return {C, db};
 }
 
 // conv wgrad: dw[Kout,R,S,C] (f32) from dy[N,HO,WO,Kout], x[N,H,W,C]

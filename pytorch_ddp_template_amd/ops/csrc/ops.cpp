@@ -34,6 +34,7 @@ torch::Tensor im2col(torch::Tensor x, int64_t R, int64_t S, int64_t stride,
 torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
                           c10::optional<torch::Tensor> bias, bool relu);
 torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B);
+std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B);
 torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
                               c10::optional<torch::Tensor> bias, int64_t stride,
                               int64_t pad, bool relu);
@@ -258,6 +259,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias") = c10::nullopt, py::arg("relu") = false,
         py::arg("out_f32") = false);
   m.def("gemm_tn", &gemm_tn);
+  m.def("gemm_tn_bias", [](torch::Tensor A, torch::Tensor B) {
+    TORCH_CHECK(A.scalar_type() == torch::kBFloat16 ||
+                A.scalar_type() == torch::kHalf,
+                "gemm_tn_bias: 16-bit dtypes only");
+    return bmm_tn_bias_bf16(A.contiguous(), B.contiguous());
+  });
   m.def("bmm_nt", &bmm_nt);
   m.def("bmm_nn", &bmm_nn);
   m.def("bmm_tn", &bmm_tn);

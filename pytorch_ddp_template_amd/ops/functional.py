@@ -70,8 +70,14 @@ class _LinearFn(torch.autograd.Function):
                 dys = ext.relu_bwd(dys, y_relu.reshape_as(dys))
             wt = ext.transpose2d(w)  # (K, N)
             dx = ext.gemm_nt(dys, wt, None, False, False)
-            dw = ext.gemm_tn(dys, xs).to(w.dtype)
-            db = ext.col_sum(dys).to(w.dtype) if ctx.has_bias else None
+            if ctx.has_bias and dys.dtype in (torch.bfloat16, torch.float16):
+                # bias grad rides inside the TN GEMM (dy is staged anyway)
+                dwf, dbf = ext.gemm_tn_bias(dys, xs)
+                dw = dwf.to(w.dtype)
+                db = dbf.to(w.dtype)
+            else:
+                dw = ext.gemm_tn(dys, xs).to(w.dtype)
+                db = ext.col_sum(dys).to(w.dtype) if ctx.has_bias else None
         else:
             if ctx.act == "relu":
                 dys = dys * (y_relu.reshape_as(dys) > 0).to(dys.dtype)

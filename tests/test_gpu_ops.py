@@ -382,3 +382,16 @@ def test_l2norm_and_scale():
     EXT.scale_by_tensor_(ts, torch.tensor(3.0, device=DEV))  # clamped to 1
     for t, b in zip(ts, before):
         torch.testing.assert_close(t.cpu(), b * 0.5, rtol=1e-6, atol=1e-7)
+
+
+def test_gemm_tn_bias():
+    """TN GEMM with the bias grad riding along: db must equal dy's column
+    sums and C must match the plain TN result."""
+    for m, i, j in [(256, 64, 64), (2048, 768, 768), (197, 64, 40)]:
+        a = t32(m, i, seed=30).to(torch.bfloat16).to(DEV)
+        b = t32(m, j, seed=31).to(torch.bfloat16).to(DEV)
+        c, db = EXT.gemm_tn_bias(a, b)
+        ref_c = a.float().t().cpu() @ b.float().cpu()
+        ref_db = a.float().sum(0).cpu()
+        close_bf16(c, ref_c, scale=ref_c.abs().max().clamp(min=0.5))
+        close_bf16(db, ref_db, scale=ref_db.abs().max().clamp(min=0.5))
