@@ -667,10 +667,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
       reinterpret_cast<v4s*>(&af[mi][0])[1] = h0;
       reinterpret_cast<v4s*>(&af[mi][1])[0] = l1;
       reinterpret_cast<v4s*>(&af[mi][1])[1] = h1;
-      if (colsum && (wave & 1) == 0) {
+      if (colsum && blockIdx.x == 0 && (wave & 1) == 0) {
         // bias grad rides along: the staged dy tile is already in
         // registers; out-of-range m/i lanes were zero-page staged.
-        // Only wn==0 waves contribute (wn pairs re-read the same A tile).
+        // Only one J-tile column of blocks and only wn==0 waves contribute
+        // (everyone else re-reads the same A tile).
         float a = 0.f;
 #pragma unroll
         for (int kh = 0; kh < 2; ++kh) {
@@ -717,7 +718,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   // ---- writeback: f32 atomics into dw[I][taps][Cin] slices ----
   // (static tap order: a runtime-rotated index into acc would demote the
   // accumulators to scratch)
-  if (colsum && (wave & 1) == 0) {
+  if (colsum && blockIdx.x == 0 && (wave & 1) == 0) {
 #pragma unroll
     for (int mi = 0; mi < TI32; ++mi) {
       float v = bsum[mi] + __shfl_xor(bsum[mi], 32);  // join the ks pair
