@@ -43,6 +43,7 @@ struct Bucket {
   std::vector<int64_t> param_indices;  // global param indices in this bucket
   std::vector<int64_t> offsets;        // element offset of each param in flat
   int64_t pending = 0;             // grads not yet accumulated this iteration
+  bool ready = false;              // all grads accumulated, awaiting launch
   c10::intrusive_ptr<c10d::Work> work;  // in-flight all-reduce
 };
 
@@ -129,6 +130,7 @@ class Reducer {
   void reset() {
     for (auto& b : buckets_) {
       b.pending = static_cast<int64_t>(b.param_indices.size());
+      b.ready = false;
       b.work = nullptr;
     }
     next_expected_bucket_ = 0;
@@ -154,7 +156,18 @@ class Reducer {
     TORCH_CHECK(b.pending > 0, "gradient marked ready twice for bucket ", bi,
                 " (param ", param_index, ")");
     if (--b.pending == 0 && sync_) {
-      launch(bi);
+      // RCCL/NCCL requires every rank to issue collectives in the SAME
+      // order.  Buckets are built in reverse registration order (the usual
+      // ready order), and launches are strictly in bucket-index order: a
+      // bucket that becomes ready early waits for its turn, so dynamic
+      // graphs or racy hook orderings can never produce a cross-rank
+      // collective-order mismatch.
+      b.ready = true;
+      while (next_expected_bucket_ < static_cast<int64_t>(buckets_.size()) &&
+             buckets_[next_expected_bucket_].ready) {
+        launch(next_expected_bucket_);
+        ++next_expected_bucket_;
+      }
     }
   }
 
@@ -174,6 +187,7 @@ class Reducer {
     for (size_t bi = 0; bi < buckets_.size(); ++bi) {
       if (!buckets_[bi].work) launch(static_cast<int64_t>(bi));
     }
+    next_expected_bucket_ = static_cast<int64_t>(buckets_.size());
     for (auto& b : buckets_) {
       if (b.work) {
         b.work->wait();
