@@ -377,28 +377,31 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                               const T* __restrict__ gamma,
                               const T* __restrict__ beta,
                               float* __restrict__ mean_out,
-                              float* __restrict__ rstd_out, int D, float eps) {
-  __shared__ float lds[4];
-  const long long m = blockIdx.x;
+                              float* __restrict__ rstd_out, long long M,
+                              int D, float eps) {
+  // one WAVE per row, 4 rows per block: rows are short (ViT D=768) and a
+  // 256-thread block per row serialized on block-wide reductions
+  const long long m = (long long)blockIdx.x * 4 + wave_id();
+  if (m >= M) return;
+  const int lane = lane_id();
   const T* row = x + m * D;
   T* yrow = y + m * D;
   float s = 0.f, ss = 0.f;
-  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+  for (int d = lane; d < D; d += 64) {
     float v = to_f(row[d]);
     s += v;
     ss += v * v;
   }
-  s = block_reduce_sum<256>(s, lds);
-  __syncthreads();
-  ss = block_reduce_sum<256>(ss, lds);
+  s = wave_reduce_sum(s);
+  ss = wave_reduce_sum(ss);
   const float mu = s / D;
   const float var = fmaxf(ss / D - mu * mu, 0.f);
   const float rs = rsqrtf(var + eps);
-  if (threadIdx.x == 0) {
+  if (lane == 0) {
     mean_out[m] = mu;
     rstd_out[m] = rs;
   }
-  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+  for (int d = lane; d < D; d += 64) {
     float v = (to_f(row[d]) - mu) * rs * to_f(gamma[d]) + to_f(beta[d]);
     yrow[d] = to_t<T>(v);
   }
@@ -410,25 +413,25 @@ __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                               const T* __restrict__ gamma,
                               const float* __restrict__ mean,
                               const float* __restrict__ rstd,
-                              T* __restrict__ dx, int D) {
-  __shared__ float lds[4];
-  const long long m = blockIdx.x;
+                              T* __restrict__ dx, long long M, int D) {
+  const long long m = (long long)blockIdx.x * 4 + wave_id();
+  if (m >= M) return;
+  const int lane = lane_id();
   const T* dyr = dy + m * D;
   const T* xr = x + m * D;
   T* dxr = dx + m * D;
   const float mu = mean[m], rs = rstd[m];
   float s1 = 0.f, s2 = 0.f;
-  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+  for (int d = lane; d < D; d += 64) {
     float g = to_f(dyr[d]) * to_f(gamma[d]);
     float xh = (to_f(xr[d]) - mu) * rs;
     s1 += g;
     s2 += g * xh;
   }
-  s1 = block_reduce_sum<256>(s1, lds);
-  __syncthreads();
-  s2 = block_reduce_sum<256>(s2, lds);
+  s1 = wave_reduce_sum(s1);
+  s2 = wave_reduce_sum(s2);
   const float invD = 1.f / D;
-  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+  for (int d = lane; d < D; d += 64) {
     float g = to_f(dyr[d]) * to_f(gamma[d]);
     float xh = (to_f(xr[d]) - mu) * rs;
     dxr[d] = to_t<T>(rs * (g - invD * (s1 + xh * s2)));
@@ -724,13 +727,13 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
   auto y = torch::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream();
   DDP_DISPATCH_FLOAT(x.scalar_type(), "ln_fwd", [&] {
-    hipLaunchKernelGGL((nrm::ln_fwd_kernel<scalar_t>), dim3((unsigned)M),
-                       dim3(256), 0, stream,
+    hipLaunchKernelGGL((nrm::ln_fwd_kernel<scalar_t>),
+                       dim3((unsigned)((M + 3) / 4)), dim3(256), 0, stream,
                        reinterpret_cast<const scalar_t*>(x.data_ptr()),
                        reinterpret_cast<scalar_t*>(y.data_ptr()),
                        reinterpret_cast<const scalar_t*>(gamma.data_ptr()),
                        reinterpret_cast<const scalar_t*>(beta.data_ptr()),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(), D,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(), M, D,
                        (float)eps);
   });
   return {y, mean, rstd};
@@ -759,11 +762,12 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   DDP_DISPATCH_FLOAT(x.scalar_type(), "ln_bwd", [&] {
     const auto* dyp = reinterpret_cast<const scalar_t*>(dy.data_ptr());
     const auto* xp = reinterpret_cast<const scalar_t*>(x.data_ptr());
-    hipLaunchKernelGGL((nrm::ln_bwd_kernel<scalar_t>), dim3((unsigned)M),
-                       dim3(256), 0, stream, dyp, xp,
+    hipLaunchKernelGGL((nrm::ln_bwd_kernel<scalar_t>),
+                       dim3((unsigned)((M + 3) / 4)), dim3(256), 0, stream,
+                       dyp, xp,
                        reinterpret_cast<const scalar_t*>(gamma.data_ptr()),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       reinterpret_cast<scalar_t*>(dx.data_ptr()), D);
+                       reinterpret_cast<scalar_t*>(dx.data_ptr()), M, D);
     hipLaunchKernelGGL((nrm::ln_bwd_param_kernel<scalar_t>), sgrid, dim3(256),
                        0, stream, dyp, xp, mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), ws.data_ptr<float>(), M, D);
