@@ -18,7 +18,8 @@ import torch
 from torch import nn
 
 from ..ops import GELU, LayerNorm, Linear
-from ..ops.functional import attention
+from ..ops.functional import attention, attention_qkv
+from ..ops.native import native_available
 
 
 class MultiHeadAttention(nn.Module):
@@ -34,6 +35,14 @@ class MultiHeadAttention(nn.Module):
         N, S, D = x.shape
         h, dh = self.heads, D // self.heads
         qkv = self.qkv(x)  # (N, S, 3D)
+        if (
+            x.is_cuda and dh == 64 and S <= 224
+            and x.dtype in (torch.bfloat16, torch.float16)
+            and native_available()
+        ):
+            # fused kernel: stages q/k/v straight from the packed qkv
+            out = attention_qkv(qkv.contiguous(), h, 1.0 / math.sqrt(dh))
+            return self.proj(out)
         qkv = qkv.reshape(N, S, 3, h, dh).permute(2, 0, 3, 1, 4)
         q, k, v = (t.reshape(N * h, S, dh).contiguous() for t in qkv)
         out = attention(q, k, v, 1.0 / math.sqrt(dh))  # (N*h, S, dh)

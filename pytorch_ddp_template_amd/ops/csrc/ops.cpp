@@ -91,6 +91,9 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor target,
                      torch::Tensor lse, torch::Tensor dloss);
 torch::Tensor mse_fwd(torch::Tensor p, torch::Tensor t);
 torch::Tensor mse_bwd(torch::Tensor p, torch::Tensor t, torch::Tensor dloss);
+// attention.hip
+std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
+                                    double scale);
 // multi_tensor.hip
 void sgd_step(std::vector<torch::Tensor> params,
               std::vector<torch::Tensor> grads,
@@ -298,6 +301,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_fwd", &softmax_fwd);
+  m.def("attn_fwd", &attn_fwd);
   m.def("softmax_bwd", &softmax_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);

@@ -638,6 +638,55 @@ def softmax(x, scale=1.0):
     return _SoftmaxFn.apply(x, scale)
 
 
+class _AttentionQKVFn(torch.autograd.Function):
+    """Fused attention forward from the PACKED qkv tensor (one kernel:
+    staging from qkv, QK^T, softmax, PV — no q/k/v permute+copy kernels).
+    Backward composes the framework's batched MFMA GEMMs on the saved
+    probability matrix P, exactly like the unfused path."""
+
+    @staticmethod
+    def forward(ctx, qkv, heads, scale):
+        out, P = native().attn_fwd(qkv, heads, scale)
+        ctx.save_for_backward(qkv, P)
+        ctx.heads, ctx.scale = heads, scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, P = ctx.saved_tensors
+        ext = native()
+        N, S, D3 = qkv.shape
+        h = ctx.heads
+        dh = D3 // (3 * h)
+        B = N * h
+        parts = qkv.reshape(N, S, 3, h, dh).permute(2, 0, 3, 1, 4)
+        q, k, v = (parts[i].reshape(B, S, dh).contiguous() for i in range(3))
+        do = (
+            dout.reshape(N, S, h, dh).permute(0, 2, 1, 3)
+            .reshape(B, S, dh).contiguous()
+        )
+        dv = ext.bmm_tn(P, do)                      # P^T @ dO
+        dp = ext.bmm_nt(do, v)                      # dO @ V^T
+        ds = ext.softmax_bwd(
+            dp.reshape(-1, S).contiguous(), P.reshape(-1, S), ctx.scale
+        ).reshape(B, S, S)
+        dq = ext.bmm_nn(ds, k)                      # dS @ K
+        dk = ext.bmm_tn(ds, q)                      # dS^T @ Q
+        dqkv = (
+            torch.stack([dq, dk, dv], 0)
+            .reshape(3, N, h, S, dh)
+            .permute(1, 3, 0, 2, 4)
+            .reshape(N, S, D3)
+            .contiguous()
+        )
+        return dqkv, None, None
+
+
+def attention_qkv(qkv, heads, scale):
+    """qkv: [N, S, 3*heads*dh] packed (the qkv Linear output)."""
+    return _AttentionQKVFn.apply(qkv, heads, scale)
+
+
 def attention(q, k, v, scale):
     """Batched attention out = softmax(Q K^T * scale) V.
 

@@ -395,3 +395,43 @@ def test_gemm_tn_bias():
         ref_db = a.float().sum(0).cpu()
         close_bf16(c, ref_c, scale=ref_c.abs().max().clamp(min=0.5))
         close_bf16(db, ref_db, scale=ref_db.abs().max().clamp(min=0.5))
+
+
+def test_attn_fwd_fused_matches_composed():
+    """Fused attention forward (attn_fwd) vs the composed bmm+softmax path,
+    and its saved P vs the composed softmax output."""
+    from pytorch_ddp_template_amd.ops.functional import attention
+
+    torch.manual_seed(3)
+    N, S, h, dh = 3, 197, 4, 64
+    qkv = (torch.randn(N, S, 3 * h * dh) * 0.5).to(torch.bfloat16).to(DEV)
+    out, P = EXT.attn_fwd(qkv, h, 1.0 / 8.0)
+    parts = qkv.reshape(N, S, 3, h, dh).permute(2, 0, 3, 1, 4)
+    q, k, v = (t.reshape(N * h, S, dh).contiguous() for t in parts)
+    ref = attention(q, k, v, 1.0 / 8.0)
+    ref_out = ref.reshape(N, h, S, dh).permute(0, 2, 1, 3).reshape(N, S, h * dh)
+    close_bf16(out, ref_out.float(), scale=ref_out.float().abs().max().clamp(min=0.5))
+    # P vs composed softmax
+    s_ref = torch.softmax(
+        (q.float() @ k.float().transpose(1, 2)) / 8.0, dim=-1
+    )
+    close_bf16(P, s_ref, scale=torch.tensor(1.0))
+
+
+def test_attention_qkv_grads_match_composed():
+    from pytorch_ddp_template_amd.ops.functional import attention, attention_qkv
+
+    torch.manual_seed(4)
+    N, S, h, dh = 2, 197, 4, 64
+    qkv = (torch.randn(N, S, 3 * h * dh) * 0.5).to(torch.bfloat16).to(DEV)
+    a = qkv.clone().requires_grad_(True)
+    b = qkv.clone().requires_grad_(True)
+    out_f = attention_qkv(a, h, 1.0 / 8.0)
+    out_f.float().square().mean().backward()
+    parts = b.reshape(N, S, 3, h, dh).permute(2, 0, 3, 1, 4)
+    q, k, v = (t.reshape(N * h, S, dh) for t in parts)
+    ref = attention(q.contiguous(), k.contiguous(), v.contiguous(), 1.0 / 8.0)
+    ref_out = ref.reshape(N, h, S, dh).permute(0, 2, 1, 3).reshape(N, S, h * dh)
+    ref_out.float().square().mean().backward()
+    assert torch.allclose(out_f.float(), ref_out.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(a.grad.float(), b.grad.float(), atol=3e-2, rtol=3e-2)
