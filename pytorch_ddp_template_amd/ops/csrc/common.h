@@ -54,16 +54,18 @@ DEV_INLINE short f2bfbits(float f) {
 DEV_INLINE int lane_id() { return threadIdx.x & (kWave - 1); }
 DEV_INLINE int wave_id() { return threadIdx.x >> 6; }
 
-// wave-wide f32 sum (64 lanes)
+// wave-wide f32 sum (64 lanes) — butterfly: the result is valid in EVERY
+// lane (shfl_down cascades are lane-0-only and silently wrong for callers
+// that broadcast the reduction, e.g. row softmax / LayerNorm)
 DEV_INLINE float wave_reduce_sum(float v) {
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off);
-  return v;  // valid in lane 0
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off);
+  return v;
 }
 
 DEV_INLINE float wave_reduce_max(float v) {
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off));
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off));
   return v;
 }
 

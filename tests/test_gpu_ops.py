@@ -409,11 +409,14 @@ def test_attn_fwd_fused_matches_composed():
     parts = qkv.reshape(N, S, 3, h, dh).permute(2, 0, 3, 1, 4)
     q, k, v = (t.reshape(N * h, S, dh).contiguous() for t in parts)
     ref = attention(q, k, v, 1.0 / 8.0)
-    ref_out = ref.reshape(N, h, S, dh).permute(0, 2, 1, 3).reshape(N, S, h * dh)
-    close_bf16(out, ref_out.float(), scale=ref_out.float().abs().max().clamp(min=0.5))
+    ref_out = (
+        ref.reshape(N, h, S, dh).permute(0, 2, 1, 3).reshape(N, S, h * dh)
+        .float().cpu()
+    )
+    close_bf16(out, ref_out, scale=ref_out.abs().max().clamp(min=0.5))
     # P vs composed softmax
     s_ref = torch.softmax(
-        (q.float() @ k.float().transpose(1, 2)) / 8.0, dim=-1
+        (q.float() @ k.float().transpose(1, 2)).cpu() / 8.0, dim=-1
     )
     close_bf16(P, s_ref, scale=torch.tensor(1.0))
 
