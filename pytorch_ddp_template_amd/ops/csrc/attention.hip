@@ -68,37 +68,36 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   T16* ldsQ = ldsV + kMaxSP * kDh;
   T16* ldsP = ldsQ + kQT * kDh;
   float* red = reinterpret_cast<float*>(ldsP + kQT * kMaxSP);
-  float* oacc = reinterpret_cast<float*>(ldsK);  // overlays K after QK^T
+  // out accumulator overlays the P tile once PV has consumed it
+  float* oacc = reinterpret_cast<float*>(ldsP);
 
   using vec16 = typename MM<T16>::vec;
   const int SP = (S + 31) & ~31;  // padded S actually used
   const int n = blockIdx.z;
   const int hh = blockIdx.y;
-  const int q0 = blockIdx.x * kQT;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int D = H * kDh;
   const long long bh = (long long)n * H + hh;
+  const int r32 = lane & 31;
+  const int ks = lane >> 5;
+  const int nst = SP / 32;  // s-tiles (<= 7)
+  const int r8 = lane >> 3;
+  const int p16 = (lane & 7) * 8;  // halfword offset within a row
 
-  // ---- stage K, V, Q via glds (1 KiB / 8 rows per instruction) ----
-  // lane -> (row = u*8 + (lane>>3), 16-B piece lane&7) of the image
+  // ---- stage K and V ONCE for this (n, head): 1 KiB / 8 rows per glds ----
   {
-    const int r8 = lane >> 3;
-    const int p16 = (lane & 7) * 8;  // halfword offset within a row
-    const int nK = SP / 8;           // 8-row groups per image
-    for (int u = wave; u < 2 * nK + kQT / 8; u += 4) {
+    const int nK = SP / 8;
+    for (int u = wave; u < 2 * nK; u += 4) {
       const T16* src = zpad;
       T16* dst;
       int row, which;
-      if (u < nK) {            // K
+      if (u < nK) {
         row = u * 8 + r8; which = 1;
         dst = ldsK + (u * 8) * kDh;
-      } else if (u < 2 * nK) { // V
+      } else {
         row = (u - nK) * 8 + r8; which = 2;
         dst = ldsV + ((u - nK) * 8) * kDh;
-      } else {                 // Q
-        row = q0 + (u - 2 * nK) * 8 + r8; which = 0;
-        dst = ldsQ + ((u - 2 * nK) * 8) * kDh;
       }
       if (row < S)
         src = qkv + (((long long)n * S + row) * 3 + which) * D + hh * kDh +
@@ -108,179 +107,213 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
           (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
     }
   }
-  __syncthreads();  // drains the glds (vmcnt) and publishes K/V/Q
 
-  // ---- phase 1: scores = scale * Q @ K^T  (wave w: s-tiles w and w+4) ----
-  const int r32 = lane & 31;
-  const int ks = lane >> 5;
-  const int nst = SP / 32;  // s-tiles (<= 7)
-  f32x16 acc[2] = {};
-#pragma unroll
-  for (int t = 0; t < 2; ++t) {
-    const int st = wave + t * 4;
-    if (st >= nst) break;
-#pragma unroll
-    for (int kc = 0; kc < 4; ++kc) {  // dh in 16-halfword chunks
-      vec16 a = *reinterpret_cast<const vec16*>(
-          ldsQ + r32 * kDh + kc * 16 + ks * 8);
-      vec16 b = *reinterpret_cast<const vec16*>(
-          ldsK + (st * 32 + r32) * kDh + kc * 16 + ks * 8);
-      acc[t] = MM<T16>::mma32(a, b, acc[t]);
-    }
-  }
-
-  // ---- softmax over s (rows = q) ----
-  // lane holds 16 regs; reg -> q row (reg&3) + 8*(reg>>2) + 4*ks.
-  // Pass 1: row max.  Reduce over the 32 cols of each tile via shuffles
-  // (masks < 32 stay within the ks half), combine this wave's tiles, then
-  // cross-wave through LDS.
-  float rmax[16], rsum[16];
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    float m = -3.0e38f;
-#pragma unroll
-    for (int t = 0; t < 2; ++t) {
-      const int st = wave + t * 4;
-      if (st >= nst) break;
-      const int col = st * 32 + r32;
-      float v = col < S ? acc[t][r] * scale : -3.0e38f;
-      m = fmaxf(m, v);
-    }
-#pragma unroll
-    for (int w = 16; w >= 1; w >>= 1) m = fmaxf(m, __shfl_xor(m, w));
-    rmax[r] = m;
-  }
-  // publish per-wave row maxes: one lane per ks-half writes its 16 rows
-  if ((lane & 31) == 0) {
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
-      red[(0 * kQT + row) * 4 + wave] = rmax[r];
-    }
-  }
-  __syncthreads();
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
-    const float* rw = &red[(0 * kQT + row) * 4];
-    rmax[r] = fmaxf(fmaxf(rw[0], rw[1]), fmaxf(rw[2], rw[3]));
-  }
-  // Pass 2: exp + row sum (same reduction path), P kept in acc
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    float s = 0.f;
-#pragma unroll
-    for (int t = 0; t < 2; ++t) {
-      const int st = wave + t * 4;
-      if (st >= nst) break;
-      const int col = st * 32 + r32;
-      float p = col < S ? __expf(acc[t][r] * scale - rmax[r]) : 0.f;
-      acc[t][r] = p;
-      s += p;
-    }
-#pragma unroll
-    for (int w = 16; w >= 1; w >>= 1) s += __shfl_xor(s, w);
-    rsum[r] = s;
-  }
-  if ((lane & 31) == 0) {
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
-      red[(1 * kQT + row) * 4 + wave] = rsum[r];
-    }
-  }
-  __syncthreads();
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
-    const float* rw = &red[(1 * kQT + row) * 4];
-    rsum[r] = 1.f / (rw[0] + rw[1] + rw[2] + rw[3]);
-  }
-
-  // ---- write P: LDS tile (PV operand) + global (saved for backward) ----
-#pragma unroll
-  for (int t = 0; t < 2; ++t) {
-    const int st = wave + t * 4;
-    if (st >= nst) continue;
-    const int col = st * 32 + r32;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
-      const T16 pv = to_t<T16>(acc[t][r] * rsum[r]);
-      ldsP[row * kMaxSP + col] = pv;
-      if (col < S && q0 + row < S)
-        P[(bh * S + q0 + row) * S + col] = pv;
-    }
-  }
-  // zero the LDS P pad columns (SP..kMaxSP) so PV reads stay clean
-  for (int c = SP + threadIdx.x; c < kMaxSP; c += 256) {
-#pragma unroll
-    for (int row = 0; row < kQT; ++row) ldsP[row * kMaxSP + c] = to_t<T16>(0.f);
-  }
-  // also zero the out accumulator (overlays dead K image)
-  for (int i = threadIdx.x; i < kQT * kDh; i += 256)
-    oacc[i] = 0.f;
-  __syncthreads();
-
-  // ---- phase 2: out = P @ V (wave w sums its s-tiles; LDS-accumulated) --
-  // A = P rows (q, k=s contiguous); B = V[s][d] via the hardware transpose
-  // read (ds_read_b64_tr_b16): V image rows are m(=s)-major exactly like
-  // the wgrad kernel's tiles; the per-lane address pattern is identical.
   const unsigned tr_off =
       (unsigned)((ks * 8 + ((lane & 15) >> 2)) * (kDh * 2) + (lane & 3) * 8);
-  const int img_sel = (lane >> 4) & 1;  // d 0..15 vs 16..31 of a 32-col pair
+  const int img_sel = (lane >> 4) & 1;
 #define LDSB(p)                                               \
   ((unsigned)(unsigned long long)(__attribute__((            \
       address_space(3))) const T16*)(p))
-  f32x16 oaccr[2] = {};  // d-tiles 0 (d 0..31) and 1 (d 32..63)
-  for (int st = wave; st < nst; st += 4) {
+
+  // ---- loop over the q tiles of this (n, head) ----
+  for (int q0 = 0; q0 < S; q0 += kQT) {
+    // stage the Q tile (waits the K/V glds too on the first pass)
+    for (int u = wave; u < kQT / 8; u += 4) {
+      const int row = q0 + u * 8 + r8;
+      const T16* src = zpad;
+      if (row < S)
+        src = qkv + (((long long)n * S + row) * 3 + 0) * D + hh * kDh + p16;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)(
+              ldsQ + (u * 8) * kDh),
+          16, 0, 0);
+    }
+    __syncthreads();  // drains glds (vmcnt): K/V/Q visible
+
+    // ---- QK^T (wave w: s-tiles w and w+4) ----
+    f32x16 acc[2] = {};
 #pragma unroll
-    for (int kh = 0; kh < 2; ++kh) {  // two 16-s halves of the 32-s tile
-      vec16 a = *reinterpret_cast<const vec16*>(
-          ldsP + r32 * kMaxSP + st * 32 + kh * 16 + ks * 8);
+    for (int t = 0; t < 2; ++t) {
+      const int st = wave + t * 4;
+      if (st >= nst) break;
 #pragma unroll
-      for (int dt = 0; dt < 2; ++dt) {
-        // V fragment: rows s = st*32 + kh*16 + .., cols d = dt*32 + lane&31
-        const T16* vimg = ldsV + (st * 32 + kh * 16) * kDh + (dt * 32 + img_sel * 16);
-        const unsigned b0 = LDSB(vimg) + tr_off;
-        v4s l0, h0;
-        asm volatile(
-            "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
-            "ds_read_b64_tr_b16 %1, %2 offset:512\n\t"  /* +4 s-rows */
-            "s_waitcnt lgkmcnt(0)"
-            : "=&v"(l0), "=&v"(h0)
-            : "v"(b0));
-        vec16 b;
-        reinterpret_cast<v4s*>(&b)[0] = l0;
-        reinterpret_cast<v4s*>(&b)[1] = h0;
-        oaccr[dt] = MM<T16>::mma32(a, b, oaccr[dt]);
+      for (int kc = 0; kc < 4; ++kc) {
+        vec16 a = *reinterpret_cast<const vec16*>(
+            ldsQ + r32 * kDh + kc * 16 + ks * 8);
+        vec16 b = *reinterpret_cast<const vec16*>(
+            ldsK + (st * 32 + r32) * kDh + kc * 16 + ks * 8);
+        acc[t] = MM<T16>::mma32(a, b, acc[t]);
       }
     }
-  }
-  // cross-wave accumulate in LDS
+
+    // ---- row softmax (two passes; regs -> rows (r&3)+8*(r>>2)+4*ks) ----
+    float rmax[16], rsum[16];
 #pragma unroll
-  for (int dt = 0; dt < 2; ++dt) {
-    const int col = dt * 32 + r32;
+    for (int r = 0; r < 16; ++r) {
+      float m = -3.0e38f;
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const int st = wave + t * 4;
+        if (st >= nst) break;
+        const int col = st * 32 + r32;
+        float v = col < S ? acc[t][r] * scale : -3.0e38f;
+        m = fmaxf(m, v);
+      }
+#pragma unroll
+      for (int w = 16; w >= 1; w >>= 1) m = fmaxf(m, __shfl_xor(m, w));
+      rmax[r] = m;
+    }
+    if ((lane & 31) == 0) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
+        red[(0 * kQT + row) * 4 + wave] = rmax[r];
+      }
+    }
+    __syncthreads();
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
-      atomicAdd(&oacc[row * kDh + col], oaccr[dt][r]);
+      const float* rw = &red[(0 * kQT + row) * 4];
+      rmax[r] = fmaxf(fmaxf(rw[0], rw[1]), fmaxf(rw[2], rw[3]));
     }
-  }
-  __syncthreads();
-  // ---- store out[N, S, h*dh]: rows q0+0..31, this head's 128-B slice ----
-  {
-    const int r8 = lane >> 3;
-    const int p16 = (lane & 7) * 8;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      float sm = 0.f;
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const int st = wave + t * 4;
+        if (st >= nst) break;
+        const int col = st * 32 + r32;
+        float pv = col < S ? __expf(acc[t][r] * scale - rmax[r]) : 0.f;
+        acc[t][r] = pv;
+        sm += pv;
+      }
+#pragma unroll
+      for (int w = 16; w >= 1; w >>= 1) sm += __shfl_xor(sm, w);
+      rsum[r] = sm;
+    }
+    if ((lane & 31) == 0) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
+        red[(1 * kQT + row) * 4 + wave] = rsum[r];
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
+      const float* rw = &red[(1 * kQT + row) * 4];
+      rsum[r] = 1.f / (rw[0] + rw[1] + rw[2] + rw[3]);
+    }
+
+    // ---- P into LDS (the PV operand) ----
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      const int st = wave + t * 4;
+      if (st >= nst) continue;
+      const int col = st * 32 + r32;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
+        ldsP[row * kMaxSP + col] = to_t<T16>(acc[t][r] * rsum[r]);
+      }
+    }
+    for (int c = SP + (int)threadIdx.x; c < kMaxSP; c += 256) {
+#pragma unroll
+      for (int row = 0; row < kQT; ++row)
+        ldsP[row * kMaxSP + c] = to_t<T16>(0.f);
+    }
+    __syncthreads();  // ldsP complete
+
+    // ---- bulk-copy P to global from LDS: 16-B stores aligned to the DST
+    // (P rows start at (..)*S elements and S is odd for ViT, so row starts
+    // are only 2-B aligned: each row gets a scalar head up to the next
+    // 16-B boundary, then gathered 16-B chunks; the per-register scatter
+    // this replaces was 2-B stores and store-issue-bound) ----
+    {
+      const int spans = (S + 7) / 8 + 1;  // per-row 8-col units (+ head slack)
+      for (int idx = (int)threadIdx.x; idx < kQT * spans; idx += 256) {
+        const int row = idx / spans;
+        if (q0 + row >= S) continue;
+        const long long rbase = (bh * S + q0 + row) * (long long)S;
+        // head: elements before the first 16-B-aligned dst position
+        const int head = (int)((8 - (rbase & 7)) & 7);
+        const int unit = idx % spans;
+        T16* dst = P + rbase;
+        const T16* srcp = ldsP + row * kMaxSP;
+        if (unit == 0) {
+          for (int j = 0; j < head && j < S; ++j) dst[j] = srcp[j];
+        } else {
+          const int c8 = head + (unit - 1) * 8;
+          if (c8 >= S) continue;
+          if (c8 + 8 <= S) {
+            T16 tmp[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) tmp[j] = srcp[c8 + j];
+            *reinterpret_cast<uint4*>(dst + c8) =
+                *reinterpret_cast<const uint4*>(tmp);
+          } else {
+            for (int j = 0; c8 + j < S; ++j) dst[c8 + j] = srcp[c8 + j];
+          }
+        }
+      }
+    }
+
+    // ---- PV: wave-partial over its s-tiles, V^T via hardware tr reads ----
+    f32x16 oaccr[2] = {};
+    for (int st = wave; st < nst; st += 4) {
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+        vec16 a = *reinterpret_cast<const vec16*>(
+            ldsP + r32 * kMaxSP + st * 32 + kh * 16 + ks * 8);
+#pragma unroll
+        for (int dt = 0; dt < 2; ++dt) {
+          const T16* vimg =
+              ldsV + (st * 32 + kh * 16) * kDh + (dt * 32 + img_sel * 16);
+          const unsigned b0 = LDSB(vimg) + tr_off;
+          v4s l0, h0;
+          asm volatile(
+              "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+              "ds_read_b64_tr_b16 %1, %2 offset:512\n\t" /* +4 s-rows */
+              "s_waitcnt lgkmcnt(0)"
+              : "=&v"(l0), "=&v"(h0)
+              : "v"(b0));
+          vec16 b;
+          reinterpret_cast<v4s*>(&b)[0] = l0;
+          reinterpret_cast<v4s*>(&b)[1] = h0;
+          oaccr[dt] = MM<T16>::mma32(a, b, oaccr[dt]);
+        }
+      }
+    }
+    __syncthreads();  // everyone done reading ldsP; reuse it as oacc
+    for (int i = (int)threadIdx.x; i < kQT * kDh; i += 256) oacc[i] = 0.f;
+    __syncthreads();
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt) {
+      const int col = dt * 32 + r32;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
+        atomicAdd(&oacc[row * kDh + col], oaccr[dt][r]);
+      }
+    }
+    __syncthreads();
+    // ---- out[N, S, h*dh]: this head's 128-B slice, 16-B stores ----
     for (int u = wave; u < kQT / 8; u += 4) {
       const int row = u * 8 + r8;
       const int qrow = q0 + row;
       if (qrow >= S) continue;
       T16* dst = out + ((long long)n * S + qrow) * D + hh * kDh + p16;
-      const float* src = &oacc[row * kDh + p16];
+      const float* srcp = &oacc[row * kDh + p16];
+      T16 tmp[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) dst[j] = to_t<T16>(src[j]);
+      for (int j = 0; j < 8; ++j) tmp[j] = to_t<T16>(srcp[j]);
+      *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(tmp);
     }
+    __syncthreads();  // oacc (= ldsP) must be fully read before next q-tile
   }
 #undef LDSB
 }
@@ -308,7 +341,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
   static torch::Tensor zp;
   if (!zp.defined() || zp.device() != qkv.device())
     zp = torch::zeros({64}, qkv.options());
-  dim3 grid((S + attn::kQT - 1) / attn::kQT, H, N);
+  dim3 grid(1, H, N);  // q-tiles looped in-kernel (K/V staged once)
   auto stream = c10::hip::getCurrentHIPStream();
   DDP_DISPATCH_FLOAT(qkv.scalar_type(), "attn_fwd", [&] {
     if constexpr (!std::is_same_v<scalar_t, float>) {
