@@ -62,11 +62,11 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const T16* __restrict__ qkv, T16* __restrict__ out, T16* __restrict__ P,
     const T16* __restrict__ zpad, int N, int S, int H, float scale) {
   __shared__ __attribute__((aligned(16))) char smem[
-      kMaxSP * kRow * 2 + kQT * kRow + kQT * kMaxSP * 2 + 2 * kQT * 4 * 4];
+      kMaxSP * kRow * 2 + 2 * kQT * kRow + kQT * kMaxSP * 2 + 3 * kQT * 4 * 4];
   T16* ldsK = reinterpret_cast<T16*>(smem);
   T16* ldsV = ldsK + kMaxSP * kDh;
-  T16* ldsQ = ldsV + kMaxSP * kDh;
-  T16* ldsP = ldsQ + kQT * kDh;
+  T16* ldsQb = ldsV + kMaxSP * kDh;  // [2][kQT][kDh] double-buffered Q
+  T16* ldsP = ldsQb + 2 * kQT * kDh;
   float* red = reinterpret_cast<float*>(ldsP + kQT * kMaxSP);
   // out accumulator overlays the P tile once PV has consumed it
   float* oacc = reinterpret_cast<float*>(ldsP);
@@ -116,8 +116,9 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       address_space(3))) const T16*)(p))
 
   // ---- loop over the q tiles of this (n, head) ----
-  for (int q0 = 0; q0 < S; q0 += kQT) {
-    // stage the Q tile (waits the K/V glds too on the first pass)
+  // Q is double-buffered: tile t+1's loads are issued before tile t's
+  // compute so their latency hides under the MFMAs/softmax.
+  auto stage_q = [&](int q0, int buf) {
     for (int u = wave; u < kQT / 8; u += 4) {
       const int row = q0 + u * 8 + r8;
       const T16* src = zpad;
@@ -126,10 +127,21 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)(
-              ldsQ + (u * 8) * kDh),
+              ldsQb + (buf * kQT + u * 8) * kDh),
           16, 0, 0);
     }
-    __syncthreads();  // drains glds (vmcnt): K/V/Q visible
+  };
+  stage_q(0, 0);
+  int qbuf = 0;
+  bool first = true;
+  for (int q0 = 0; q0 < S; q0 += kQT) {
+    const T16* ldsQ = ldsQb + qbuf * kQT * kDh;
+    if (q0 + kQT < S) stage_q(q0 + kQT, qbuf ^ 1);
+    if (first) {
+      __syncthreads();  // drains glds (vmcnt): K/V/Q(0) visible
+      first = false;
+    }
+    // (later barriers in this iteration drain the prefetched Q tile)
 
     // ---- QK^T (wave w: s-tiles w and w+4) ----
     f32x16 acc[2] = {};
@@ -313,7 +325,8 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       for (int j = 0; j < 8; ++j) tmp[j] = to_t<T16>(srcp[j]);
       *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(tmp);
     }
-    __syncthreads();  // oacc (= ldsP) must be fully read before next q-tile
+    __syncthreads();  // oacc (= ldsP) read + next Q tile (vmcnt) published
+    qbuf ^= 1;
   }
 #undef LDSB
 }
