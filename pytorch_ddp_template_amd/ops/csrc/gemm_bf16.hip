@@ -332,7 +332,12 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     // shuffle reduces them; each wave PAIR (wave>>1 disambiguates waves
     // sharing a column range at different row halves) stores its slice into
     // its own workspace row (ws_nblocks = 2 * grid blocks).
-    const long long wsrow = (long long)flat_id * 2 + (wave >> 1);
+    // ws rows are capped (host: ws_nblocks <= 8192); blocks beyond the cap
+    // wrap around and accumulate atomically (big-M launches would otherwise
+    // scale the workspace and its zero-fill/finalize with M)
+    const long long vrow = (long long)flat_id * 2 + (wave >> 1);
+    const long long wsrow = vrow % ws_nblocks;
+    const bool wrap = vrow >= ws_nblocks;
 #pragma unroll
     for (int ni = 0; ni < NI32; ++ni) {
       float sv = col_sum[ni], qv = col_sq[ni];
@@ -340,8 +345,13 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
       qv += __shfl_xor(qv, 32);
       const int col = n0 + wn + ni * 32 + r32;
       if (ks == 0 && col < N) {
-        stats_ws[wsrow * N + col] = sv;
-        stats_ws[((long long)ws_nblocks + wsrow) * N + col] = qv;
+        if (wrap) {
+          atomicAdd(&stats_ws[wsrow * N + col], sv);
+          atomicAdd(&stats_ws[((long long)ws_nblocks + wsrow) * N + col], qv);
+        } else {
+          stats_ws[wsrow * N + col] = sv;
+          stats_ws[((long long)ws_nblocks + wsrow) * N + col] = qv;
+        }
       }
     }
   }
@@ -905,9 +915,11 @@ std::vector<torch::Tensor> conv2d_fwd_bf16_impl(
   auto y = torch::empty({N, HO, WO, Kout}, x.options());
   auto& zp = zero_page(x.device(), x.scalar_type());
   dim3 grid((Kout + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
-  // 2 workspace rows per block (per wave-pair; see the epilogue)
-  const int nblocks =
-      2 * (int)(((Kout <= 64 ? (Kout + 63) / 64 : grid.x)) * grid.y);
+  // 2 workspace rows per block (per wave-pair; see the epilogue), capped —
+  // overflow blocks wrap around with atomic accumulation
+  const int nblocks = std::min(
+      8192,
+      2 * (int)(((Kout <= 64 ? (Kout + 63) / 64 : grid.x)) * grid.y));
   g16::ConvMeta cm{H, W, cl, S, R, (int)stride, (int)pad, HO, WO};
   NtExtras ex{};
   torch::Tensor ws;
