@@ -1110,6 +1110,188 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
   return {C, db};
 }
 
+namespace g16 {
+// -------------------------------------- TN conv wgrad, segment-staged ----
+// 3x3 stride-1 wgrad with pow2 HO/WO: instead of gathering 9 shifted x
+// tiles per chunk, stage dy + THREE spatial x row-tiles with a 2-column
+// halo (the s-shift becomes a +s*32B tr-read base offset into the same
+// tile).  Cuts the per-chunk glds count 40 -> 28 and the x LDS/gather
+// traffic 3x; all addressing is shift arithmetic (pow2 gate).
+//
+// x tile layout: G = max(1, 32/WO) spatial segments of L = min(32, WO)
+// output columns each, stored as q = seg*(L+2) + col rows of [16 ch]
+// (E = G*(L+2) <= 48 rows, image padded to 64 so each is exactly 2 glds).
+// A fragment m-window of 4 always sits inside one segment (L >= 4, windows
+// 4-aligned); the second tr read's row jump is L>=8 ? +4 : +6 (uniform).
+template <typename T16>
+__global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
+    const T16* __restrict__ dy, const T16* __restrict__ x,
+    float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
+    int I /*Kout*/, int J /*Cin*/, long long ldc, ConvMeta cm, int lgWO,
+    int lgHO, int zsplit) {
+  constexpr int BI = 64, BJ = 64, BMC = 32;
+  constexpr int IMG_A = 32 * 16;   // dy images: [32 m][16 ch]
+  constexpr int IMG_X = 64 * 16;   // x images: [<=48 q rows][16 ch], padded
+  constexpr int TILE_A = 4 * IMG_A;       // 4 KiB
+  constexpr int TILE_X = 4 * IMG_X;       // 8 KiB per kernel row r
+  typedef short v4s __attribute__((ext_vector_type(4)));
+  using vec16 = typename M16<T16>::vec;
+  __shared__ __attribute__((aligned(16))) T16 lds[2 * (TILE_A + 3 * TILE_X)];
+
+  const int Cin = J;
+  const int i0 = blockIdx.y * BI;
+  const int j0 = blockIdx.x * BJ;
+  const int n_chunks = (Mtot + BMC - 1) / BMC;
+  const int per_z = (n_chunks + zsplit - 1) / zsplit;
+  const int ch0 = (int)blockIdx.z * per_z;
+  const int ch1 = min(n_chunks, ch0 + per_z);
+  if (ch0 >= ch1) return;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
+  const int sm = lane >> 1;
+  const int sh8 = (lane & 1) * 8;
+  const int L = min(32, 1 << lgWO);  // seg length (pow2)
+  const int lgL = lgWO < 5 ? lgWO : 5;
+  const int segs = max(1, 32 >> lgL);
+  const int E = segs * (L + 2);
+  const int rcpL2 = 65536 / (L + 2) + 1;  // magic recip (q <= 63)
+
+  f32x16 acc[9] = {};
+
+  // ---- stage chunk: dy (4 glds) + 3 x row-tiles (8 glds each) ----
+  auto stage = [&](int buf, int ch) {
+    const int m0 = ch * BMC;
+    T16* base = lds + buf * (TILE_A + 3 * TILE_X);
+    for (int u = wave; u < 4 + 24; u += 4) {
+      const T16* src = zpad;
+      T16* dst;
+      if (u < 4) {  // dy image u
+        const int gm = m0 + sm;
+        const int ii = i0 + u * 16 + sh8;
+        if (gm < Mtot && ii < I) src = dy + (long long)gm * I + ii;
+        dst = base + u * IMG_A;
+      } else {
+        const int v = u - 4;          // 0..23
+        const int rt = v / 8;         // kernel row 0..2
+        const int g2 = (v / 4) & 1;   // which 32-row half of the image
+        const int ig = v & 3;         // 16-ch group
+        const int q = g2 * 32 + sm;   // lds row within the image
+        dst = base + TILE_A + rt * TILE_X + ig * IMG_X + g2 * (32 * 16);
+        if (q < E) {
+          const int seg = (q * rcpL2) >> 16;
+          const int c = q - seg * (L + 2);
+          const long long m_seg = (long long)m0 + (long long)seg * L;
+          if (m_seg < Mtot) {
+            const int wo0 = (int)(m_seg & ((1 << lgWO) - 1));
+            const long long t = m_seg >> lgWO;
+            const int ho = (int)(t & ((1 << lgHO) - 1));
+            const int n = (int)(t >> lgHO);
+            const int hi = ho - cm.pad + rt;
+            const int wi = wo0 - cm.pad + c;
+            const int jj = j0 + ig * 16 + sh8;
+            if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W && jj < Cin)
+              src = x + (((long long)n * cm.H + hi) * cm.W + wi) * Cin + jj;
+          }
+        }
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
+    }
+  };
+
+  const int ks = lane >> 5;
+  const unsigned tr_lane_off =
+      (unsigned)((ks * 8 + ((lane & 15) >> 2)) * 32 + (lane & 3) * 8);
+  const int img_sel = (lane >> 4) & 1;
+  // x-tile q rows for this lane's two m-windows (kh = 0/1), plus the
+  // uniform row jump of the second tr read of each window pair
+  const int jump = (L >= 8 ? 4 : 6) * 32;  // bytes = rows*32B
+  int qw[2];
+#pragma unroll
+  for (int kh = 0; kh < 2; ++kh) {
+    const int mw = kh * 16 + ks * 8;
+    qw[kh] = (mw >> lgL) * (L + 2) + (mw & (L - 1));
+  }
+#define LDS_BYTE(p)                                           \
+  ((unsigned)(unsigned long long)(__attribute__((            \
+      address_space(3))) const T16*)(p))
+
+  stage(0, ch0);
+  __syncthreads();
+
+  for (int ch = ch0; ch < ch1; ++ch) {
+    const int buf = (ch - ch0) & 1;
+    const bool more = ch + 1 < ch1;
+    if (more) stage(buf ^ 1, ch + 1);
+    const T16* base = lds + buf * (TILE_A + 3 * TILE_X);
+
+    // A fragments (dy), as in the generic tr kernel
+    vec16 af[2];
+    {
+      const unsigned a0 =
+          LDS_BYTE(base + ((wm >> 4) + img_sel) * IMG_A) + tr_lane_off;
+      v4s l0, h0, l1, h1;
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
+          "ds_read_b64_tr_b16 %2, %4 offset:512\n\t"
+          "ds_read_b64_tr_b16 %3, %4 offset:640\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
+          : "v"(a0), "v"(a0));
+      reinterpret_cast<v4s*>(&af[0])[0] = l0;
+      reinterpret_cast<v4s*>(&af[0])[1] = h0;
+      reinterpret_cast<v4s*>(&af[1])[0] = l1;
+      reinterpret_cast<v4s*>(&af[1])[1] = h1;
+    }
+#pragma unroll
+    for (int rt = 0; rt < 3; ++rt) {
+      const T16* timg =
+          base + TILE_A + rt * TILE_X + (img_sel + (wn >> 4)) * IMG_X;
+#pragma unroll
+      for (int s2 = 0; s2 < 3; ++s2) {
+#pragma unroll
+        for (int kh = 0; kh < 2; ++kh) {
+          const unsigned b0 =
+              LDS_BYTE(timg) + (unsigned)((qw[kh] + s2) * 32) +
+              (unsigned)(((lane & 15) >> 2) * 32 + (lane & 3) * 8);
+          const unsigned b1 = b0 + (unsigned)jump;
+          v4s l0, h0;
+          asm volatile(
+              "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+              "ds_read_b64_tr_b16 %1, %3 offset:0\n\t"
+              "s_waitcnt lgkmcnt(0)"
+              : "=&v"(l0), "=&v"(h0)
+              : "v"(b0), "v"(b1));
+          vec16 bf;
+          reinterpret_cast<v4s*>(&bf)[0] = l0;
+          reinterpret_cast<v4s*>(&bf)[1] = h0;
+          acc[rt * 3 + s2] = M16<T16>::mma32(af[kh], bf, acc[rt * 3 + s2]);
+        }
+      }
+    }
+    if (more) __syncthreads();
+  }
+
+  // ---- writeback (32x32x16 C/D layout) ----
+#pragma unroll
+  for (int tap = 0; tap < 9; ++tap) {
+    const long long coff = (long long)tap * Cin;
+    const int col = j0 + wn + (lane & 31);
+    if (col >= Cin) continue;
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int row = i0 + wm + (reg & 3) + 8 * (reg >> 2) + 4 * ks;
+      if (row >= I) continue;
+      atomicAdd(&dw[(long long)row * ldc + coff + col], acc[tap][reg]);
+    }
+  }
+}
+}  // namespace g16
+
 // conv wgrad: dw[Kout,R,S,C] (f32) from dy[N,HO,WO,Kout], x[N,H,W,C]
 torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                                 int64_t stride, int64_t pad, int64_t R,
@@ -1132,9 +1314,25 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   int z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
   const long long ldc = (long long)R * S * Cin;
   dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
+  auto pow2l = [](int v) {
+    int l = 0;
+    while ((1 << l) < v) ++l;
+    return ((1 << l) == v && v >= 4) ? l : -1;
+  };
+  const int lgWO = pow2l(WO), lgHO = pow2l(HO);
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
-    if (R == 3 && S == 3) {
+    if (R == 3 && S == 3 && stride == 1 && lgWO >= 2 && lgHO >= 2) {
+      // segment-staged fast path (stride-1, pow2 spatial): 28 glds/chunk
+      // instead of 40, x gathered once per kernel ROW not per tap
+      hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16>), grid,
+                         dim3(g16::THREADS), 0, stream,
+                         reinterpret_cast<const t16*>(dy.data_ptr()),
+                         reinterpret_cast<const t16*>(x.data_ptr()),
+                         dw.data_ptr<float>(),
+                         reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
+                         Cin, ldc, cm, lgWO, lgHO, (int)grid.z);
+    } else if (R == 3 && S == 3) {
       hipLaunchKernelGGL((g16::gemm_wgrad_tr_kernel<t16, 9>), grid,
                          dim3(g16::THREADS), 0, stream,
                          reinterpret_cast<const t16*>(dy.data_ptr()),

@@ -172,7 +172,8 @@ def test_conv2d_dgrad_bf16(n, h, c, k, r, stride, pad):
 
 @pytest.mark.parametrize(
     "n,h,c,k,r,stride,pad",
-    [(2, 16, 64, 64, 3, 1, 1), (2, 16, 64, 128, 3, 2, 1), (2, 16, 3, 64, 3, 1, 1)],
+    [(2, 16, 64, 64, 3, 1, 1), (2, 16, 64, 128, 3, 2, 1), (2, 16, 3, 64, 3, 1, 1),
+     (2, 4, 64, 64, 3, 1, 1), (2, 8, 128, 64, 3, 1, 1), (3, 12, 64, 64, 3, 1, 1)],
 )
 def test_conv2d_wgrad_bf16(n, h, c, k, r, stride, pad):
     ho = (h + 2 * pad - r) // stride + 1
