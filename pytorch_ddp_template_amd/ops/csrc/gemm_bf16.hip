@@ -896,6 +896,11 @@ torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
+// conv_halo.hip: direct-from-LDS-halo 3x3 stride-1 fast path
+bool conv2d_fwd_halo(torch::Tensor x, torch::Tensor w, torch::Tensor y,
+                     torch::Tensor zp, int64_t stride, int64_t pad,
+                     float* stats_ws, int ws_nblocks);
+
 // conv2d forward, NHWC x[N,H,W,C] * w[Kout,R,S,C] -> y[N,HO,WO,Kout]
 // stats=true additionally returns the BN partial workspace ws
 // ([2*nblocks][Kout], block-level sums/sumsqs of y) so the following
@@ -928,6 +933,12 @@ std::vector<torch::Tensor> conv2d_fwd_bf16_impl(
                       x.options().dtype(torch::kFloat32));
     ex.stats_ws = ws.data_ptr<float>();
     ex.ws_nblocks = nblocks;
+  }
+  if (!relu && !bias.has_value() &&
+      conv2d_fwd_halo(x, w, y, zp, stride, pad, ex.stats_ws,
+                      ex.ws_nblocks)) {
+    if (stats) return {y, ws};
+    return {y};
   }
   if (x.scalar_type() == torch::kBFloat16)
     launch_nt16<bf16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M, Kout,
