@@ -992,6 +992,11 @@ torch::Tensor conv2d_dgrad_bf16(torch::Tensor dy, torch::Tensor wr,
   dim3 grid((Cin + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM, 1);
   g16::ConvMeta cm{HS,   WS,   cl,   S,  R, 1, dpad, HOut, WOut,
                    (int)stride, HOs, WOs};
+  // dense (stride-1) dgrad is exactly a 3x3 stride-1 pad-1 conv of dy with
+  // the rotated weight: take the LDS-halo fast path when it applies
+  if ((int)stride == 1 && !addend.has_value() &&
+      conv2d_fwd_halo(dy, wr, dx, zp, 1, dpad, nullptr, 0))
+    return dx;
   NtExtras ex{};
   if (addend.has_value()) {
     TORCH_CHECK(addend->is_contiguous() && addend->numel() == dx.numel() &&
