@@ -160,7 +160,16 @@ def _conv_ref_nhwc(x, w, b, stride, pad):
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride, pad, act):
+        cpad = 0
         if use_native(x, w):
+            C = x.shape[-1]
+            if C < 8:
+                # narrow input (the 3-channel stem): zero-pad channels to 8
+                # so the fast implicit-GEMM path applies instead of
+                # materializing a huge im2col (740 MB for ResNet-50's stem)
+                cpad = 8 - C
+                x = torch.nn.functional.pad(x, (0, cpad))
+                w = torch.nn.functional.pad(w, (0, cpad))
             y = native().conv2d_fwd(
                 x.contiguous(), w.contiguous(), b, stride, pad, act == "relu"
             )
@@ -169,22 +178,30 @@ class _Conv2dFn(torch.autograd.Function):
             if act == "relu":
                 y = torch.relu(y)
         ctx.save_for_backward(x, w, y if act == "relu" else None)
-        ctx.meta = (stride, pad, act, b is not None)
+        ctx.meta = (stride, pad, act, b is not None, cpad)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x, w, y_relu = ctx.saved_tensors
-        stride, pad, act, has_bias = ctx.meta
+        stride, pad, act, has_bias, cpad = ctx.meta
         dy = dy.contiguous()
         if use_native(dy, w):
             ext = native()
             if act == "relu":
                 dy = ext.relu_bwd(dy, y_relu)
-            dx = ext.conv2d_dgrad(dy, w, stride, pad, x.shape[1], x.shape[2])
+            dx = (
+                ext.conv2d_dgrad(dy, w, stride, pad, x.shape[1], x.shape[2])
+                if ctx.needs_input_grad[0]
+                else None
+            )
             dw = ext.conv2d_wgrad(dy, x, stride, pad, w.shape[1], w.shape[2]).to(
                 w.dtype
             )
+            if cpad:
+                dw = dw[..., : w.shape[-1] - cpad].contiguous()
+                if dx is not None:
+                    dx = dx[..., : w.shape[-1] - cpad].contiguous()
             db = (
                 ext.col_sum(dy.reshape(-1, dy.shape[-1])).to(w.dtype)
                 if has_bias
