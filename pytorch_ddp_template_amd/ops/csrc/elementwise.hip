@@ -87,35 +87,6 @@ struct GeluBwdOp {  // (dy, x) -> dy * (Phi(x) + x phi(x))
 };
 
 // ---------------- column sum (bias gradient): dy[M,N] -> f32 [N] ---------
-// vectorized: each lane owns an 8-column slot (16-B row segments instead of
-// 2-B); scalar variant kept for N % 8 != 0
-template <typename T>
-__global__ void col_sum_kernel_v8(const T* __restrict__ dy,
-                                  float* __restrict__ out, long long M,
-                                  long long N) {
-  using IO = VecIO<T>;
-  const long long c0 = ((long long)blockIdx.x * kWave + lane_id()) * 8;
-  if (c0 >= N) return;
-  const int nw = blockDim.x / kWave;
-  const long long rows_per_blk = (M + gridDim.y - 1) / gridDim.y;
-  const long long m0 = blockIdx.y * rows_per_blk;
-  const long long m1 = min(M, m0 + rows_per_blk);
-  float acc[8] = {};
-  for (long long m = m0 + wave_id(); m < m1; m += nw) {
-    if (IO::kPerLane == 8) {
-      auto v = *reinterpret_cast<const typename VecIO<T>::Vec*>(
-          dy + m * N + c0);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc[j] += IO::get(v, j);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc[j] += to_f(dy[m * N + c0 + j]);
-    }
-  }
-#pragma unroll
-  for (int j = 0; j < 8; ++j) atomicAdd(&out[c0 + j], acc[j]);
-}
-
 template <typename T>
 __global__ void col_sum_kernel(const T* __restrict__ dy,
                                float* __restrict__ out, long long M,
