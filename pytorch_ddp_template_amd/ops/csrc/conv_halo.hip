@@ -48,8 +48,10 @@ constexpr int BM = 128;              // output rows per block
 constexpr int THREADS = 256;
 constexpr int HALO_MAX = 24576;      // halfwords (48 KiB)
 
-// weight tile: [BNT rows][32 k] like the NT kernel's B tile, 3 buffers
-template <typename T16, int BNT, bool EXTRAS>
+// weight tile: [BNT rows][32 k] like the NT kernel's B tile, 3 buffers.
+// BMT: output rows per block (256 for narrow-N layers: more MFMA work per
+// addressed fragment in this issue-bound regime).
+template <typename T16, int BNT, bool EXTRAS, int BMT = BM>
 __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
     const T16* __restrict__ x, const T16* __restrict__ w,
     T16* __restrict__ y, const T16* __restrict__ zpad, int M, int N, int K,
@@ -62,11 +64,11 @@ __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
   T16* bt = lds + HALO_MAX;
 
   const int flat_id = (int)(blockIdx.y * gridDim.x + blockIdx.x);
-  const int m0 = (int)blockIdx.y * BM;
+  const int m0 = (int)blockIdx.y * BMT;
   const int n0 = (int)blockIdx.x * BNT;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int G = BM >> lgW;            // output rows in this block
+  const int G = BMT >> lgW;           // output rows in this block
   const int Wp2 = W + 2;
   const int HROWS = G + 2;
   const int halo_hw = HROWS * Wp2 * C;  // halfwords used
@@ -123,17 +125,17 @@ __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
   };
 
   const int KT = (K + BK - 1) / BK;
-  const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BNT / 2);
+  constexpr int MI32 = BMT / 128;  // 32-row tiles per wave (wave dim = 2)
+  const int wm = (wave >> 1) * (BMT / 2), wn = (wave & 1) * (BNT / 2);
   const int r32 = lane & 31;
   const int ks = lane >> 5;
   constexpr int NI32 = BNT / 64;
-  f32x16 acc[2][NI32] = {};
+  f32x16 acc[2 * MI32][NI32] = {};
 
-  // per-lane A-row geometry (fixed over the k loop): for mi in {0,1} the
-  // m row is wm + mi*32 + r32 -> (segment, output column)
-  int seg_[2], wo_[2];
+  // per-lane A-row geometry (fixed over the k loop)
+  int seg_[2 * MI32], wo_[2 * MI32];
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi) {
+  for (int mi = 0; mi < 2 * MI32; ++mi) {
     const int m = wm + mi * 32 + r32;
     seg_[mi] = m >> lgW;
     wo_[mi] = m & (W - 1);
@@ -162,9 +164,9 @@ __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
       const int r = (rs * 21846) >> 16;  // /3 for rs < 9
       const int s = rs - r * 3;
       const int c0 = kw & (C - 1);
-      vec16 af[2];
+      vec16 af[2 * MI32];
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi) {
+      for (int mi = 0; mi < 2 * MI32; ++mi) {
         const int pos = (seg_[mi] + r) * Wp2 + wo_[mi] + s;
         const int cslot = (c0 >> 3) ^ (pos & (cch - 1));
         af[mi] = *reinterpret_cast<const vec16*>(
@@ -178,7 +180,7 @@ __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
         vec16 bf = *reinterpret_cast<const vec16*>(
             bbase + rb * 32 + swz * 8);
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
+        for (int mi = 0; mi < 2 * MI32; ++mi)
           acc[mi][ni] = MM<T16>::mma32(af[mi], bf, acc[mi][ni]);
       }
     }
@@ -199,7 +201,7 @@ __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
     const int col = n0 + wn + ni * 32 + r32;
     if (col >= N) continue;
 #pragma unroll
-    for (int mi = 0; mi < 2; ++mi) {
+    for (int mi = 0; mi < 2 * MI32; ++mi) {
 #pragma unroll
       for (int reg = 0; reg < 16; ++reg) {
         const int row =
@@ -256,21 +258,26 @@ bool conv2d_fwd_halo(torch::Tensor x, torch::Tensor w, torch::Tensor y,
   const int lgW = lg(W), lgH = lg(H), lgC = lg(C);
   if (lgW < 2 || lgH < 0 || lgC < 3) return false;
   if ((long long)H * W % ch::BM != 0) return false;  // block spans one image
-  const int G = ch::BM >> lgW;
-  const long long halo_hw = (long long)(G + 2) * (W + 2) * C;
+  const int M = N_ * H * W, K = 9 * C;
+  const int BNT = Kout <= 64 ? 64 : 128;
+  // narrow-N layers take 256 output rows per block: 2x the MFMA work per
+  // addressed fragment (this regime is instruction-issue bound)
+  // (the template BMT must match: narrow-N is compiled at 256 only)
+  const int BMT = BNT == 64 ? 256 : ch::BM;
+  const long long halo_hw = (long long)((BMT >> lgW) + 2) * (W + 2) * C;
   // +512: the last glds unit's lane-linear writes may run past the
   // used region (their sources are zpad) — keep them inside the carve
   if (halo_hw + 512 > ch::HALO_MAX) return false;
-  const int M = N_ * H * W, K = 9 * C;
-  const int BNT = Kout <= 64 ? 64 : 128;
-  dim3 grid((Kout + BNT - 1) / BNT, (M + ch::BM - 1) / ch::BM, 1);
+  if ((long long)H * W % BMT != 0) return false;
+  dim3 grid((Kout + BNT - 1) / BNT, (M + BMT - 1) / BMT, 1);
   auto stream = c10::hip::getCurrentHIPStream();
   const bool ex = stats_ws != nullptr;
   auto do_launch = [&](auto tag, auto bntc, auto exc) {
     using scalar_t = decltype(tag);
+    constexpr int BMTv = decltype(bntc)::value == 64 ? 256 : ch::BM;
     hipLaunchKernelGGL(
         (ch::conv_halo_kernel<scalar_t, decltype(bntc)::value,
-                              decltype(exc)::value>),
+                              decltype(exc)::value, BMTv>),
         grid, dim3(ch::THREADS), 0, stream,
         reinterpret_cast<const scalar_t*>(x.data_ptr()),
         reinterpret_cast<const scalar_t*>(w.data_ptr()),
