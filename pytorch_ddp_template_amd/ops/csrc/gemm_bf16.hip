@@ -82,7 +82,7 @@ struct ConvMeta {
   int ss, SH, SW;
 };
 
-enum { MODE_PLAIN = 0, MODE_CONV = 1 };
+enum { MODE_PLAIN = 0, MODE_CONV = 1, MODE_CONVJ = 2 };
 
 // ---------------------------------------------------------------- NT -----
 // BNT: the N tile (128, or 64 when N <= 64 — ResNet layer1 Kout, C=64
@@ -534,6 +534,12 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 // MODE_PLAIN reuses the same machinery for the plain TN GEMM
 // C[I,J] += sum_m A[m,I] B[m,J] (linear / ViT weight grads): the B gather is
 // then just rows of [M,J] and TAPS must be 1.  J == Cin for conv mode.
+// MODE_CONVJ (TAPS == 1) puts ALL taps of a generic RxS conv wgrad on the
+// J axis instead: dw[Kout][R*S*Cin] is ONE TN GEMM where logical column
+// j = tap*Cin + c gathers x at that tap's spatial shift (tap = j >> C_log2,
+// r = tap/S, s = tap%S).  One launch stages each dy chunk grid.x times
+// instead of R*S times — the ResNet-50 stem (7x7 s2, Cin-padded 8) was 49
+// per-tap TN launches re-reading all of dy, 28% of its whole train step.
 // Batched via grid.z = nbatch * zsplit (blockIdx.z / zsplit selects the
 // batch, % zsplit the split-M slice) — ViT attention backward runs ~1.5K
 // small TN GEMMs per call and a host-side per-batch launch loop was 70%%
@@ -583,12 +589,15 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
 
   f32x16 acc[TI / 2][TJ / 2][TAPS] = {};  // 32x32 tiles
 
+  // magic reciprocal for the CONVJ tap -> (r,s) decode (S <= 256)
+  const int rcpS = MODE == MODE_PLAIN ? 0 : 65536 / cm.S + 1;
+
   // ---- stage chunk ch into LDS buffer buf: 4*NOPS glds, no VALU pack ----
   auto stage = [&](int buf, int ch) {
     const int gm = ch * BMC + sm;
     const bool mok = gm < Mtot;
     int n = 0, hb = 0, wb = 0;
-    if (MODE == MODE_CONV && mok) {
+    if (MODE != MODE_PLAIN && mok) {
       int t = gm;
       const int wo = t % cm.WO;
       t /= cm.WO;
@@ -611,6 +620,20 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
         const int ig = u - AIMGS;  // TAPS == 1
         const int jj = j0 + ig * 16 + sh8;
         if (mok && jj < J) src = x + (long long)gm * J + jj;
+        dst = base + TILE_A + ig * IMG;
+      } else if (MODE == MODE_CONVJ) {
+        const int ig = u - AIMGS;  // TAPS == 1
+        // logical col jc = tap*Cin_phys + c; each lane's 8-elem group stays
+        // inside one tap because Cin_phys is pow2 >= 8 and jc % 8 == 0
+        const int jc = j0 + ig * 16 + sh8;
+        const int tap = jc >> cm.C_log2;
+        const int r = (tap * rcpS) >> 16;
+        const int s = tap - r * cm.S;
+        const int hi = hb + r, wi = wb + s;
+        if (mok && jc < J && hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W)
+          src = x +
+                (((long long)n * cm.H + hi) * cm.W + wi) * (1LL << cm.C_log2) +
+                (jc & ((1 << cm.C_log2) - 1));
         dst = base + TILE_A + ig * IMG;
       } else {
         const int tap = (u - AIMGS) / BIMGS;
@@ -1364,16 +1387,22 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                          dw.data_ptr<float>(),
                          reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
                          Cin, ldc, cm, 0, 0, 0, (int)grid.z);
-    } else {  // generic R x S: per-tap TN launches (cold path)
-      for (int r = 0; r < (int)R; ++r)
-        for (int s2 = 0; s2 < (int)S; ++s2)
-          hipLaunchKernelGGL(
-              (g16::gemm_tn_bf16_kernel<t16, g16::MODE_CONV>), grid,
-              dim3(g16::THREADS), 0, stream,
-              reinterpret_cast<const t16*>(dy.data_ptr()),
-              reinterpret_cast<const t16*>(x.data_ptr()),
-              dw.data_ptr<float>(), M, Kout, Cin, r, s2, ldc,
-              ((long long)r * S + s2) * Cin, cm);
+    } else {  // generic R x S: taps-on-J, ONE launch (dw[I][R*S*Cin] as a
+      // single TN GEMM; col j = tap*Cin + c gathers x at tap's shift).
+      // The per-tap TN alternative re-read all of dy R*S times — the 7x7
+      // ResNet-50 stem spent 13.4 ms/step (28% of the step) there.
+      const int Jl = (int)ldc;  // = R*S*Cin
+      int tiles2 = ((Jl + 127) / 128) * ((Kout + 63) / 64);
+      int z2 = std::max(
+          1, std::min(n_chunks, (512 + tiles2 - 1) / std::max(1, tiles2)));
+      dim3 grid2((Jl + 127) / 128, (Kout + 63) / 64, (unsigned)z2);
+      hipLaunchKernelGGL(
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_CONVJ, 2, 4>), grid2,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(dy.data_ptr()),
+          reinterpret_cast<const t16*>(x.data_ptr()), dw.data_ptr<float>(),
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout, Jl, ldc, cm,
+          0, 0, 0, z2);
     }
   };
   if (x.scalar_type() == torch::kBFloat16) run(bf16{});
