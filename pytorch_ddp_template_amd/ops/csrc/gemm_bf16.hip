@@ -1379,6 +1379,20 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
                          dw.data_ptr<float>(),
                          reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
                          Cin, ldc, cm, 0, 0, 0, (int)grid.z);
+    } else if (R == 1 && S == 1 && Kout >= 128 && Cin >= 128) {
+      // wide 128x128 tiles: the bottleneck 1x1s (up to 512x2048 dw) re-read
+      // dy grid.x times; halving grid.x halves that traffic
+      int tiles2 = ((Cin + 127) / 128) * ((Kout + 127) / 128);
+      int z2 = std::max(
+          1, std::min(n_chunks, (512 + tiles2 - 1) / std::max(1, tiles2)));
+      dim3 grid2((Cin + 127) / 128, (Kout + 127) / 128, (unsigned)z2);
+      hipLaunchKernelGGL(
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_CONV, 4, 4>), grid2,
+          dim3(g16::THREADS), 0, stream,
+          reinterpret_cast<const t16*>(dy.data_ptr()),
+          reinterpret_cast<const t16*>(x.data_ptr()), dw.data_ptr<float>(),
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout, Cin, ldc, cm,
+          0, 0, 0, z2);
     } else if (R == 1 && S == 1) {
       hipLaunchKernelGGL((g16::gemm_wgrad_tr_kernel<t16, 1>), grid,
                          dim3(g16::THREADS), 0, stream,
