@@ -40,12 +40,29 @@ from .native import native, use_native
 # ---------------------------------------------------------------------------
 
 
+def _plain_gemm_to_blas(m: int, k: int, n: int) -> bool:
+    """Big plain GEMMs go to rocBLAS/hipBLASLt (the MI355X library path);
+    the hand-written NT/TN kernels keep every fused shape (bias/relu
+    epilogues, conv modes, bias-grad-in-GEMM).  Measured on the ViT-B/16
+    linear shapes (tools/gemm_ab.py, 1xMI355X): rocBLAS 846-1097 TF vs
+    510-687 TF for the in-house NT kernel; the TN wgrad with its fused
+    bias grad stays with the in-house kernel (competitive or faster).
+    """
+    return m >= 4096 and k >= 512 and n >= 512
+
+
 class _LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, act):
         # x: (..., K) ; w: (N, K) ; b: (N,) or None
         xs = x.reshape(-1, x.shape[-1])
-        if use_native(x, w):
+        if use_native(x, w) and _plain_gemm_to_blas(
+            xs.shape[0], xs.shape[1], w.shape[0]
+        ):
+            y = torch.nn.functional.linear(xs.contiguous(), w, b)
+            if act == "relu":
+                y = torch.relu(y)
+        elif use_native(x, w):
             y = native().gemm_nt(xs.contiguous(), w, b, act == "relu", False)
         else:
             y = xs @ w.t()
@@ -68,8 +85,11 @@ class _LinearFn(torch.autograd.Function):
             ext = native()
             if ctx.act == "relu":
                 dys = ext.relu_bwd(dys, y_relu.reshape_as(dys))
-            wt = ext.transpose2d(w)  # (K, N)
-            dx = ext.gemm_nt(dys, wt, None, False, False)
+            if _plain_gemm_to_blas(dys.shape[0], dys.shape[1], w.shape[1]):
+                dx = dys @ w  # rocBLAS NN, no w-transpose materialization
+            else:
+                wt = ext.transpose2d(w)  # (K, N)
+                dx = ext.gemm_nt(dys, wt, None, False, False)
             if ctx.has_bias and dys.dtype in (torch.bfloat16, torch.float16):
                 # bias grad rides inside the TN GEMM (dy is staged anyway)
                 dwf, dbf = ext.gemm_tn_bias(dys, xs)
