@@ -49,25 +49,47 @@ constexpr int kDh = 64;        // head dim (fixed)
 constexpr int kQT = 32;        // q rows per workgroup
 constexpr int kMaxSP = 224;    // padded S capacity (ViT-B/16: S=197)
 constexpr int kRow = kDh * 2;  // 128 B per K/V/Q row
+// P-tile row stride (elements).  224 would put every row start on the same
+// 2 banks (448 B = 112 words, 112 % 32 = 16); 232 gives 8 distinct starts,
+// and combined with the chunk swizzle below the PV operand reads are
+// conflict-free.
+constexpr int kPS = 232;
+
+// K/Q images are 16-B-chunk XOR-swizzled: chunk c of row r is stored at
+// chunk position c ^ (r & 7).  Plain row-major 128-B rows put every lane of
+// a b128 operand read on the same 4 banks (the row stride is a multiple of
+// the 32-bank span -> 32-way conflict, which PMC showed as the kernel being
+// wait-bound at 5x SQ_BUSY).  global_load_lds writes lane-linearly, so the
+// swizzle is applied to the SOURCE address each lane fetches (the 16-B
+// chunks of one 128-B row land permuted but stay in the same cache lines).
+// V is NOT swizzled: it is read through ds_read_b64_tr_b16 whose addressing
+// is fixed by the transpose unit.
+DEV_INLINE int kqswz(int lane8, int row8) { return ((lane8 ^ row8) & 7) * 8; }
+// P-tile chunk swizzle (chunks < 28 carry data; chunk 28 is stride pad)
+DEV_INLINE int pswz(int row, int col) {
+  int c = col >> 3;
+  c = c < 28 ? (c ^ (row & 3)) : c;
+  return row * kPS + c * 8 + (col & 7);
+}
 
 // LDS layout (bytes):
-//   K image   [kMaxSP][128]          28672
-//   V image   [kMaxSP][128]          28672
-//   Q tile    [kQT][128]              4096
-//   P tile    [kQT][kMaxSP*2 + pad]  14336 (bf16, row stride 448 B)
+//   K image   [kMaxSP][128]          28672   (chunk-swizzled)
+//   V image   [kMaxSP][128]          28672   (natural; tr reads)
+//   Q tile    [2][kQT][128]           8192   (chunk-swizzled)
+//   P tile    [kQT][kPS*2]           14848   (bf16, swizzled, stride 464 B)
 //   red       [2][kQT][4] f32         1024
-//   out acc   overlays the K image (K is dead after QK^T)
+//   out acc   overlays the P tile (P is dead after PV)
 template <typename T16>
 __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const T16* __restrict__ qkv, T16* __restrict__ out, T16* __restrict__ P,
     const T16* __restrict__ zpad, int N, int S, int H, float scale) {
   __shared__ __attribute__((aligned(16))) char smem[
-      kMaxSP * kRow * 2 + 2 * kQT * kRow + kQT * kMaxSP * 2 + 3 * kQT * 4 * 4];
+      kMaxSP * kRow * 2 + 2 * kQT * kRow + kQT * kPS * 2 + 2 * kQT * 4 * 4];
   T16* ldsK = reinterpret_cast<T16*>(smem);
   T16* ldsV = ldsK + kMaxSP * kDh;
   T16* ldsQb = ldsV + kMaxSP * kDh;  // [2][kQT][kDh] double-buffered Q
   T16* ldsP = ldsQb + 2 * kQT * kDh;
-  float* red = reinterpret_cast<float*>(ldsP + kQT * kMaxSP);
+  float* red = reinterpret_cast<float*>(ldsP + kQT * kPS);
   // out accumulator overlays the P tile once PV has consumed it
   float* oacc = reinterpret_cast<float*>(ldsP);
 
@@ -92,16 +114,19 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       const T16* src = zpad;
       T16* dst;
       int row, which;
+      int chunk_off;
       if (u < nK) {
         row = u * 8 + r8; which = 1;
         dst = ldsK + (u * 8) * kDh;
+        chunk_off = kqswz(lane, r8);  // K is chunk-swizzled
       } else {
         row = (u - nK) * 8 + r8; which = 2;
         dst = ldsV + ((u - nK) * 8) * kDh;
+        chunk_off = p16;  // V natural (tr reads)
       }
       if (row < S)
         src = qkv + (((long long)n * S + row) * 3 + which) * D + hh * kDh +
-              p16;
+              chunk_off;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
@@ -123,7 +148,8 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       const int row = q0 + u * 8 + r8;
       const T16* src = zpad;
       if (row < S)
-        src = qkv + (((long long)n * S + row) * 3 + 0) * D + hh * kDh + p16;
+        src = qkv + (((long long)n * S + row) * 3 + 0) * D + hh * kDh +
+              kqswz(lane, r8);  // Q chunk-swizzled like K
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)(
@@ -144,18 +170,23 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     // (later barriers in this iteration drain the prefetched Q tile)
 
     // ---- QK^T (wave w: s-tiles w and w+4) ----
+    // chunk index (kc*2+ks) XOR'd with the LDS row's swizzle key; Q
+    // fragments hoisted out of the s-tile loop (same row both iterations)
     f32x16 acc[2] = {};
+    vec16 aq[4];
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc)
+      aq[kc] = *reinterpret_cast<const vec16*>(
+          ldsQ + r32 * kDh + kqswz(kc * 2 + ks, r32));
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
       const int st = wave + t * 4;
       if (st >= nst) break;
 #pragma unroll
       for (int kc = 0; kc < 4; ++kc) {
-        vec16 a = *reinterpret_cast<const vec16*>(
-            ldsQ + r32 * kDh + kc * 16 + ks * 8);
         vec16 b = *reinterpret_cast<const vec16*>(
-            ldsK + (st * 32 + r32) * kDh + kc * 16 + ks * 8);
-        acc[t] = MM<T16>::mma32(a, b, acc[t]);
+            ldsK + (st * 32 + r32) * kDh + kqswz(kc * 2 + ks, r32));
+        acc[t] = MM<T16>::mma32(aq[kc], b, acc[t]);
       }
     }
 
@@ -230,13 +261,13 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
-        ldsP[row * kMaxSP + col] = to_t<T16>(acc[t][r] * rsum[r]);
+        ldsP[pswz(row, col)] = to_t<T16>(acc[t][r] * rsum[r]);
       }
     }
     for (int c = SP + (int)threadIdx.x; c < kMaxSP; c += 256) {
 #pragma unroll
       for (int row = 0; row < kQT; ++row)
-        ldsP[row * kMaxSP + c] = to_t<T16>(0.f);
+        ldsP[pswz(row, c)] = to_t<T16>(0.f);
     }
     __syncthreads();  // ldsP complete
 
@@ -246,29 +277,31 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     // 16-B boundary, then gathered 16-B chunks; the per-register scatter
     // this replaces was 2-B stores and store-issue-bound) ----
     {
+      // one row per 8 threads (kQT*8 = 256: no runtime div), units strided 8
       const int spans = (S + 7) / 8 + 1;  // per-row 8-col units (+ head slack)
-      for (int idx = (int)threadIdx.x; idx < kQT * spans; idx += 256) {
-        const int row = idx / spans;
-        if (q0 + row >= S) continue;
+      const int row = (int)threadIdx.x >> 3;
+      if (q0 + row < S) {
         const long long rbase = (bh * S + q0 + row) * (long long)S;
         // head: elements before the first 16-B-aligned dst position
         const int head = (int)((8 - (rbase & 7)) & 7);
-        const int unit = idx % spans;
         T16* dst = P + rbase;
-        const T16* srcp = ldsP + row * kMaxSP;
-        if (unit == 0) {
-          for (int j = 0; j < head && j < S; ++j) dst[j] = srcp[j];
-        } else {
-          const int c8 = head + (unit - 1) * 8;
-          if (c8 >= S) continue;
-          if (c8 + 8 <= S) {
-            T16 tmp[8];
-#pragma unroll
-            for (int j = 0; j < 8; ++j) tmp[j] = srcp[c8 + j];
-            *reinterpret_cast<uint4*>(dst + c8) =
-                *reinterpret_cast<const uint4*>(tmp);
+        for (int unit = (int)threadIdx.x & 7; unit < spans; unit += 8) {
+          if (unit == 0) {
+            for (int j = 0; j < head && j < S; ++j)
+              dst[j] = ldsP[pswz(row, j)];
           } else {
-            for (int j = 0; c8 + j < S; ++j) dst[c8 + j] = srcp[c8 + j];
+            const int c8 = head + (unit - 1) * 8;
+            if (c8 >= S) continue;
+            if (c8 + 8 <= S) {
+              T16 tmp[8];
+#pragma unroll
+              for (int j = 0; j < 8; ++j) tmp[j] = ldsP[pswz(row, c8 + j)];
+              *reinterpret_cast<uint4*>(dst + c8) =
+                  *reinterpret_cast<const uint4*>(tmp);
+            } else {
+              for (int j = 0; c8 + j < S; ++j)
+                dst[c8 + j] = ldsP[pswz(row, c8 + j)];
+            }
           }
         }
       }
@@ -280,7 +313,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
       for (int kh = 0; kh < 2; ++kh) {
         vec16 a = *reinterpret_cast<const vec16*>(
-            ldsP + r32 * kMaxSP + st * 32 + kh * 16 + ks * 8);
+            ldsP + pswz(r32, st * 32 + kh * 16 + ks * 8));
 #pragma unroll
         for (int dt = 0; dt < 2; ++dt) {
           const T16* vimg =
