@@ -368,6 +368,78 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 // are kDh*2 = 128 B so the wgrad-style constant folds out — asserted here.
 static_assert(kDh * 2 == 128, "tr-read pattern assumes 128-B V rows");
 
+// ---- packed-qkv <-> per-head layout movers for the composed backward ----
+// torch's generic 5-D permute+contiguous copies ran at ~140 GB/s and were
+// ~6% of the ViT step (89 eager elementwise launches); these are plain
+// 16-B-unit copies with both sides coalesced within each dh row.
+
+// All three movers keep per-thread index math to shifts/masks: the big
+// non-pow2 dims (n*h, n*S) live on blockIdx.z and are decoded ONCE per
+// block with a scalar divide; dh8 must be a pow2 (dh in {8,16,...,128}).
+// A first cut decoded a flat index with per-thread 64-bit `/ S` and `% h`
+// — it measured SLOWER than torch's generic permute (which uses magic
+// dividers); these versions have no per-thread division at all.
+
+// q/k/v[(n*h+hd), s, :] = qkv[n, s, which, hd, :]
+// grid: (ceil(S*dh8/256), 3, N*h)
+template <typename T16>
+__global__ void qkv_unpack_kernel(const T16* __restrict__ qkv,
+                                  T16* __restrict__ q, T16* __restrict__ k,
+                                  T16* __restrict__ v, int S, int h,
+                                  int lg_dh8) {
+  const int u2 = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (u2 >= (S << lg_dh8)) return;
+  const int s = u2 >> lg_dh8;
+  const int d8 = u2 & ((1 << lg_dh8) - 1);
+  const int nh = (int)blockIdx.z;           // n*h + hd
+  const int n = nh / h, hd = nh - n * h;    // scalar, once per block
+  const int which = blockIdx.y;
+  T16* outp = which == 0 ? q : which == 1 ? k : v;
+  reinterpret_cast<uint4*>(outp)[((long long)nh * S << lg_dh8) + u2] =
+      reinterpret_cast<const uint4*>(
+          qkv)[((((long long)n * S + s) * 3 + which) * h + hd) << lg_dh8 | d8];
+}
+
+// dqkv[n, s, which, hd, :] = {dq,dk,dv}[which][(n*h+hd), s, :]
+// grid: (ceil(3*h*dh8/256), S, N)
+template <typename T16>
+__global__ void qkv_pack_kernel(const T16* __restrict__ dq,
+                                const T16* __restrict__ dk,
+                                const T16* __restrict__ dv,
+                                T16* __restrict__ dqkv, int S, int h,
+                                int lg_dh8, int rcp_h) {
+  const int u2 = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (u2 >= ((3 * h) << lg_dh8)) return;
+  const int d8 = u2 & ((1 << lg_dh8) - 1);
+  const int t = u2 >> lg_dh8;               // which*h + hd, < 3*h
+  const int which = (t * rcp_h) >> 16;      // magic /h (t tiny: exact)
+  const int hd = t - which * h;
+  const int s = (int)blockIdx.y;
+  const long long n = blockIdx.z;
+  const T16* srcp = which == 0 ? dq : which == 1 ? dk : dv;
+  reinterpret_cast<uint4*>(
+      dqkv)[(((n * S + s) * 3 * h) << lg_dh8) + u2] =
+      reinterpret_cast<const uint4*>(
+          srcp)[((n * h + hd) * S + s) << lg_dh8 | d8];
+}
+
+// y[(n*h+hd), s, :] = x[n, s, hd, :]   (dout head split)
+// grid: (ceil(S*dh8/256), 1, N*h)
+template <typename T16>
+__global__ void head_split_kernel(const T16* __restrict__ x,
+                                  T16* __restrict__ y, int S, int h,
+                                  int lg_dh8) {
+  const int u2 = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (u2 >= (S << lg_dh8)) return;
+  const int s = u2 >> lg_dh8;
+  const int d8 = u2 & ((1 << lg_dh8) - 1);
+  const int nh = (int)blockIdx.z;
+  const int n = nh / h, hd = nh - n * h;    // scalar, once per block
+  reinterpret_cast<uint4*>(y)[((long long)nh * S << lg_dh8) + u2] =
+      reinterpret_cast<const uint4*>(
+          x)[(((long long)n * S + s) * h + hd) << lg_dh8 | d8];
+}
+
 }  // namespace attn
 
 // out[N,S,H*dh], P[N*H, S, S]
@@ -403,4 +475,96 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
     }
   });
   return {out, P};
+}
+
+namespace {
+int attn_log2_exact(int v) {
+  int l = 0;
+  while ((1 << l) < v) ++l;
+  return ((1 << l) == v) ? l : -1;
+}
+}  // namespace
+
+// qkv [N, S, 3*h*dh] -> (q, k, v) each [N*h, S, dh]
+std::vector<torch::Tensor> qkv_unpack(torch::Tensor qkv, int64_t heads) {
+  TORCH_CHECK(qkv.dim() == 3 && qkv.is_contiguous());
+  const int N = (int)qkv.size(0), S = (int)qkv.size(1);
+  const int h = (int)heads;
+  const int dh = (int)(qkv.size(2) / (3 * h));
+  const int lg = attn_log2_exact(dh / 8);
+  TORCH_CHECK(dh % 8 == 0 && lg >= 0, "qkv_unpack needs pow2 dh/8");
+  auto opt = qkv.options();
+  auto q = torch::empty({(long long)N * h, S, dh}, opt);
+  auto k = torch::empty({(long long)N * h, S, dh}, opt);
+  auto v = torch::empty({(long long)N * h, S, dh}, opt);
+  dim3 grid((unsigned)(((S << lg) + 255) / 256), 3, (unsigned)(N * h));
+  auto stream = c10::hip::getCurrentHIPStream();
+  DDP_DISPATCH_FLOAT(qkv.scalar_type(), "qkv_unpack", [&] {
+    if constexpr (!std::is_same_v<scalar_t, float>) {
+      hipLaunchKernelGGL((attn::qkv_unpack_kernel<scalar_t>), grid, dim3(256),
+                         0, stream,
+                         reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
+                         reinterpret_cast<scalar_t*>(q.data_ptr()),
+                         reinterpret_cast<scalar_t*>(k.data_ptr()),
+                         reinterpret_cast<scalar_t*>(v.data_ptr()), S, h, lg);
+    } else {
+      TORCH_CHECK(false, "qkv_unpack: bf16/f16 only");
+    }
+  });
+  return {q, k, v};
+}
+
+// (dq, dk, dv) each [N*h, S, dh] -> dqkv [N, S, 3*h*dh]
+torch::Tensor qkv_pack(torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                       int64_t N, int64_t heads) {
+  TORCH_CHECK(dq.is_contiguous() && dk.is_contiguous() && dv.is_contiguous());
+  TORCH_CHECK(dq.scalar_type() == dk.scalar_type() &&
+              dq.scalar_type() == dv.scalar_type());
+  const int h = (int)heads;
+  const int S = (int)dq.size(1), dh = (int)dq.size(2);
+  const int lg = attn_log2_exact(dh / 8);
+  TORCH_CHECK(dh % 8 == 0 && lg >= 0, "qkv_pack needs pow2 dh/8");
+  auto dqkv = torch::empty({N, S, (long long)3 * h * dh}, dq.options());
+  TORCH_CHECK(N <= 65535 && S <= 65535, "qkv_pack grid limits");
+  dim3 grid((unsigned)((((3 * h) << lg) + 255) / 256), (unsigned)S,
+            (unsigned)N);
+  auto stream = c10::hip::getCurrentHIPStream();
+  DDP_DISPATCH_FLOAT(dq.scalar_type(), "qkv_pack", [&] {
+    if constexpr (!std::is_same_v<scalar_t, float>) {
+      hipLaunchKernelGGL((attn::qkv_pack_kernel<scalar_t>), grid, dim3(256),
+                         0, stream,
+                         reinterpret_cast<const scalar_t*>(dq.data_ptr()),
+                         reinterpret_cast<const scalar_t*>(dk.data_ptr()),
+                         reinterpret_cast<const scalar_t*>(dv.data_ptr()),
+                         reinterpret_cast<scalar_t*>(dqkv.data_ptr()), S, h,
+                         lg, 65536 / h + 1);
+    } else {
+      TORCH_CHECK(false, "qkv_pack: bf16/f16 only");
+    }
+  });
+  return dqkv;
+}
+
+// x [N, S, h*dh] -> y [N*h, S, dh]
+torch::Tensor head_split(torch::Tensor x, int64_t heads) {
+  TORCH_CHECK(x.dim() == 3 && x.is_contiguous());
+  const int N = (int)x.size(0), S = (int)x.size(1);
+  const int h = (int)heads;
+  const int dh = (int)(x.size(2) / h);
+  const int lg = attn_log2_exact(dh / 8);
+  TORCH_CHECK(dh % 8 == 0 && lg >= 0, "head_split needs pow2 dh/8");
+  auto y = torch::empty({(long long)N * h, S, dh}, x.options());
+  dim3 grid((unsigned)(((S << lg) + 255) / 256), 1, (unsigned)(N * h));
+  auto stream = c10::hip::getCurrentHIPStream();
+  DDP_DISPATCH_FLOAT(x.scalar_type(), "head_split", [&] {
+    if constexpr (!std::is_same_v<scalar_t, float>) {
+      hipLaunchKernelGGL((attn::head_split_kernel<scalar_t>), grid, dim3(256),
+                         0, stream,
+                         reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                         reinterpret_cast<scalar_t*>(y.data_ptr()), S, h, lg);
+    } else {
+      TORCH_CHECK(false, "head_split: bf16/f16 only");
+    }
+  });
+  return y;
 }

@@ -92,6 +92,10 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor target,
 torch::Tensor mse_fwd(torch::Tensor p, torch::Tensor t);
 torch::Tensor mse_bwd(torch::Tensor p, torch::Tensor t, torch::Tensor dloss);
 // attention.hip
+std::vector<torch::Tensor> qkv_unpack(torch::Tensor qkv, int64_t heads);
+torch::Tensor qkv_pack(torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                       int64_t N, int64_t heads);
+torch::Tensor head_split(torch::Tensor x, int64_t heads);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
                                     double scale);
 // multi_tensor.hip
@@ -302,6 +306,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_fwd", &softmax_fwd);
   m.def("attn_fwd", &attn_fwd);
+  m.def("qkv_unpack", &qkv_unpack);
+  m.def("qkv_pack", &qkv_pack);
+  m.def("head_split", &head_split);
   m.def("softmax_bwd", &softmax_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
