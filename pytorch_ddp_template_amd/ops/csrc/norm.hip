@@ -125,15 +125,19 @@ __global__ void bn_finalize_kernel(const float* __restrict__ ws, int nblocks,
   }
 }
 
-// ---- BN pass 3: normalize (+optional fused relu) ----
+// ---- BN pass 3: normalize (+optional fused relu, +optional residual) ----
 // Vectorized: each thread owns one 8-channel slot and strides rows, so the
-// per-channel scale/shift loads hoist out of the row loop.
-template <typename T, bool RELU>
+// per-channel scale/shift loads hoist out of the row loop.  ADD fuses the
+// residual skip into this pass (y = relu(bn(x) + addend)): the separate
+// add_relu pass (read bn-out + skip, write out) disappears — one fewer
+// full-tensor read+write per residual block.
+template <typename T, bool RELU, bool ADD = false>
 __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
                                const float* __restrict__ mean,
                                const float* __restrict__ rstd,
                                const T* __restrict__ gamma,
-                               const T* __restrict__ beta, long long M, int C) {
+                               const T* __restrict__ beta, long long M, int C,
+                               const T* __restrict__ addend = nullptr) {
   using IO = VecIO<T>;
   const int slots = C / 8;
   const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -150,10 +154,15 @@ __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
   for (long long m = row0; m < M; m += rstride) {
     if (IO::kPerLane == 8) {
       auto v = *reinterpret_cast<const typename VecIO<T>::Vec*>(x + m * C + c0);
+      typename VecIO<T>::Vec va{};
+      if (ADD)
+        va = *reinterpret_cast<const typename VecIO<T>::Vec*>(
+            addend + m * C + c0);
       typename VecIO<T>::Vec o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float r = IO::get(v, j) * sc[j] + sh[j];
+        if (ADD) r += IO::get(va, j);
         IO::set(o, j, RELU ? fmaxf(r, 0.f) : r);
       }
       *reinterpret_cast<typename VecIO<T>::Vec*>(y + m * C + c0) = o;
@@ -161,6 +170,7 @@ __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float r = to_f(x[m * C + c0 + j]) * sc[j] + sh[j];
+        if (ADD) r += to_f(addend[m * C + c0 + j]);
         y[m * C + c0 + j] = to_t<T>(RELU ? fmaxf(r, 0.f) : r);
       }
     }
@@ -588,7 +598,8 @@ std::vector<torch::Tensor> bn_fwd_ws(torch::Tensor x, torch::Tensor gamma,
                                      torch::Tensor beta, torch::Tensor ws,
                                      c10::optional<torch::Tensor> running_mean,
                                      c10::optional<torch::Tensor> running_var,
-                                     double momentum, double eps, bool relu) {
+                                     double momentum, double eps, bool relu,
+                                     c10::optional<torch::Tensor> addend) {
   TORCH_CHECK(x.dim() == 2 && x.is_contiguous());
   long long M = x.size(0);
   int C = (int)x.size(1);
@@ -612,7 +623,16 @@ std::vector<torch::Tensor> bn_fwd_ws(torch::Tensor x, torch::Tensor gamma,
     const auto* gp = reinterpret_cast<const scalar_t*>(gamma.data_ptr());
     const auto* bp = reinterpret_cast<const scalar_t*>(beta.data_ptr());
     auto* yp = reinterpret_cast<scalar_t*>(y.data_ptr());
-    if (relu)
+    if (addend.has_value()) {
+      TORCH_CHECK(relu, "bn_fwd_ws addend is only fused with relu=true");
+      TORCH_CHECK(addend->is_contiguous() && addend->numel() == x.numel() &&
+                  addend->scalar_type() == x.scalar_type());
+      hipLaunchKernelGGL((nrm::bn_norm_kernel<scalar_t, true, true>),
+                         dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
+                         xp, yp, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         gp, bp, M, C,
+                         reinterpret_cast<const scalar_t*>(addend->data_ptr()));
+    } else if (relu)
       hipLaunchKernelGGL((nrm::bn_norm_kernel<scalar_t, true>),
                          dim3(grid_1d(M * C / 8, 256)), dim3(256), 0, stream,
                          xp, yp, mean.data_ptr<float>(), rstd.data_ptr<float>(),

@@ -69,7 +69,8 @@ std::vector<torch::Tensor> bn_fwd_ws(torch::Tensor x, torch::Tensor gamma,
                                      torch::Tensor beta, torch::Tensor ws,
                                      c10::optional<torch::Tensor> running_mean,
                                      c10::optional<torch::Tensor> running_var,
-                                     double momentum, double eps, bool relu);
+                                     double momentum, double eps, bool relu,
+                                     c10::optional<torch::Tensor> addend);
 torch::Tensor bn_infer(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
                        torch::Tensor rmean, torch::Tensor rvar, double eps,
                        bool relu);
@@ -298,7 +299,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2d_fwd", &maxpool2d_fwd);
   m.def("maxpool2d_bwd", &maxpool2d_bwd);
   m.def("bn_fwd", &bn_fwd);
-  m.def("bn_fwd_ws", &bn_fwd_ws);
+  m.def("bn_fwd_ws", &bn_fwd_ws, py::arg("x"), py::arg("gamma"),
+        py::arg("beta"), py::arg("ws"), py::arg("running_mean"),
+        py::arg("running_var"), py::arg("momentum"), py::arg("eps"),
+        py::arg("relu"), py::arg("addend") = c10::nullopt);
   m.def("bn_infer", &bn_infer);
   m.def("bn_bwd", &bn_bwd, py::arg("dy"), py::arg("x"), py::arg("gamma"),
         py::arg("mean"), py::arg("rstd"), py::arg("y_relu") = c10::nullopt);

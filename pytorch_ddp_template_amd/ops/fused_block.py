@@ -62,10 +62,6 @@ class _FusedBasicBlockFn(torch.autograd.Function):
         )
         z = z2.reshape(t1.shape)
         t2, ws2 = ext.conv2d_fwd_stats(z, w2, None, 1, 1, False)
-        y2, mean2, rstd2 = ext.bn_fwd_ws(
-            _flat(t2), g2, b2, ws2, bn2.running_mean, bn2.running_var,
-            mom, eps, False,
-        )
         if wd is not None:
             td, wsd = ext.conv2d_fwd_stats(x, wd, None, stride, 0, False)
             idn2, meand, rstdd = ext.bn_fwd_ws(
@@ -76,7 +72,13 @@ class _FusedBasicBlockFn(torch.autograd.Function):
         else:
             td = meand = rstdd = None
             idn = x
-        out = ext.add_relu_fwd(y2.reshape(t2.shape), idn)
+        # residual add + relu fused into bn2's normalize pass (one fewer
+        # full-tensor read+write than a separate add_relu)
+        out2, mean2, rstd2 = ext.bn_fwd_ws(
+            _flat(t2), g2, b2, ws2, bn2.running_mean, bn2.running_var,
+            mom, eps, True, _flat(idn.contiguous()),
+        )
+        out = out2.reshape(t2.shape)
         ctx.save_for_backward(
             x, w1, g1, t1, z, mean1, rstd1, w2, g2, t2, mean2, rstd2,
             wd, gd, td, meand, rstdd, out,
@@ -180,9 +182,6 @@ class _FusedBottleneckFn(torch.autograd.Function):
             mom, eps, True)
         z2 = z2f.reshape(t2.shape)
         t3, ws3 = ext.conv2d_fwd_stats(z2, w3, None, 1, 0, False)
-        y3, mean3, rstd3 = ext.bn_fwd_ws(
-            _flat(t3), g3, b3, ws3, bn3.running_mean, bn3.running_var,
-            mom, eps, False)
         if wd is not None:
             td, wsd = ext.conv2d_fwd_stats(x, wd, None, stride, 0, False)
             idnf, meand, rstdd = ext.bn_fwd_ws(
@@ -192,7 +191,11 @@ class _FusedBottleneckFn(torch.autograd.Function):
         else:
             td = meand = rstdd = None
             idn = x
-        out = ext.add_relu_fwd(y3.reshape(t3.shape), idn)
+        # residual add + relu fused into bn3's normalize pass
+        out3, mean3, rstd3 = ext.bn_fwd_ws(
+            _flat(t3), g3, b3, ws3, bn3.running_mean, bn3.running_var,
+            mom, eps, True, _flat(idn.contiguous()))
+        out = out3.reshape(t3.shape)
         ctx.save_for_backward(
             x, w1, g1, t1, z1, mean1, rstd1, w2, g2, t2, z2, mean2, rstd2,
             w3, g3, t3, mean3, rstd3, wd, gd, td, meand, rstdd, out)
