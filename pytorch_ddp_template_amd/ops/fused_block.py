@@ -7,9 +7,12 @@ Why (measured on MI355X, ResNet-18/CIFAR bf16 step profile):
   stats pass that re-read every conv output disappears;
 * BN backward applies the downstream ReLU mask inline (``bn_bwd(y_relu=·)``)
   — no relu_bwd pass;
-* the residual skip gradient (ReLU-masked dy) is accumulated inside the
-  conv1/downsample dgrad epilogue (``conv2d_dgrad(addend=·)``) — the
-  autograd-engine at::add of two full-size grad tensors disappears.
+* the residual add + final ReLU ride in the last BN's normalize epilogue
+  (``bn_fwd_ws(addend=·)``) — the separate add_relu pass disappears.
+  (The backward-side equivalent — accumulating the skip grad inside the
+  conv dgrad epilogue — was measured SLOWER (+200 us/launch of per-lane
+  2-B gathers against the MFMA C-layout scatter) and reverted; the skip
+  grad uses a plain bandwidth-bound add.)
 
 The unfused module path (ops/modules.py) remains the CPU/eval/odd-shape
 reference; tests compare the two.  Reference scope note: the reference

@@ -442,3 +442,41 @@ def test_attention_qkv_grads_match_composed():
     ref_out.float().square().mean().backward()
     assert torch.allclose(out_f.float(), ref_out.float(), atol=3e-2, rtol=3e-2)
     assert torch.allclose(a.grad.float(), b.grad.float(), atol=3e-2, rtol=3e-2)
+
+
+def test_qkv_movers_match_permute():
+    """qkv_unpack / qkv_pack / head_split vs the torch permute they replace
+    (the attention backward's layout movers, attention.hip)."""
+    torch.manual_seed(7)
+    N, S, h, dh = 3, 197, 4, 64
+    qkv = torch.randn(N, S, 3 * h * dh).to(torch.bfloat16).to(DEV)
+    q, k, v = EXT.qkv_unpack(qkv, h)
+    parts = qkv.reshape(N, S, 3, h, dh).permute(2, 0, 3, 1, 4)
+    for got, want in zip((q, k, v), parts):
+        assert torch.equal(got.cpu(), want.reshape(N * h, S, dh).contiguous().cpu())
+    x = torch.randn(N, S, h * dh).to(torch.bfloat16).to(DEV)
+    ys = EXT.head_split(x, h)
+    want = x.reshape(N, S, h, dh).permute(0, 2, 1, 3).reshape(N * h, S, dh)
+    assert torch.equal(ys.cpu(), want.contiguous().cpu())
+    dqkv = EXT.qkv_pack(q, k, v, N, h)
+    assert torch.equal(dqkv.cpu(), qkv.cpu())
+
+
+def test_bn_fwd_ws_addend_relu():
+    """bn_fwd_ws(addend=·): y = relu(bn(x) + skip) fused into the normalize
+    pass (norm.hip bn_norm_kernel<.., ADD=true>)."""
+    torch.manual_seed(8)
+    M, C = 4096, 64
+    x = torch.randn(M, C).to(torch.bfloat16).to(DEV)
+    skip = torch.randn(M, C).to(torch.bfloat16).to(DEV)
+    g = torch.randn(C).to(torch.bfloat16).to(DEV)
+    b = torch.randn(C).to(torch.bfloat16).to(DEV)
+    xf0 = x.float()
+    ws0 = torch.stack([xf0.sum(0), (xf0 * xf0).sum(0)])  # [2*nb=2, C] partials
+    y, mean, rstd = EXT.bn_fwd_ws(x, g, b, ws0, None, None, 0.1, 1e-5, True, skip)
+    xf = x.float()
+    mu = xf.mean(0)
+    var = xf.var(0, unbiased=False)
+    ref = (xf - mu) / (var + 1e-5).sqrt() * g.float() + b.float()
+    ref = torch.relu(ref + skip.float()).cpu()
+    close_bf16(y, ref, scale=ref.abs().max().clamp(min=0.5))
