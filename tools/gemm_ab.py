@@ -42,7 +42,6 @@ def main():
         t_ours_f = bench(lambda: EXT.gemm_nt(x, w, bias, False, False))
         t_blas_f = bench(lambda: torch.nn.functional.linear(x, w, bias))
         # dgrad: dx = dy @ w  (ours: explicit transpose + NT)
-        wt = EXT.transpose2d(w)
         t_ours_dx = bench(lambda: EXT.gemm_nt(dy, EXT.transpose2d(w), None, False, False))
         t_blas_dx = bench(lambda: dy @ w)
         # wgrad: dw = dy^T @ x
