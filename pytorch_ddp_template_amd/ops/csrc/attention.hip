@@ -497,6 +497,7 @@ std::vector<torch::Tensor> qkv_unpack(torch::Tensor qkv, int64_t heads) {
   auto q = torch::empty({(long long)N * h, S, dh}, opt);
   auto k = torch::empty({(long long)N * h, S, dh}, opt);
   auto v = torch::empty({(long long)N * h, S, dh}, opt);
+  TORCH_CHECK((long long)N * h <= 65535, "qkv_unpack grid.z limit");
   dim3 grid((unsigned)(((S << lg) + 255) / 256), 3, (unsigned)(N * h));
   auto stream = c10::hip::getCurrentHIPStream();
   DDP_DISPATCH_FLOAT(qkv.scalar_type(), "qkv_unpack", [&] {
@@ -554,6 +555,7 @@ torch::Tensor head_split(torch::Tensor x, int64_t heads) {
   const int lg = attn_log2_exact(dh / 8);
   TORCH_CHECK(dh % 8 == 0 && lg >= 0, "head_split needs pow2 dh/8");
   auto y = torch::empty({(long long)N * h, S, dh}, x.options());
+  TORCH_CHECK((long long)N * h <= 65535, "head_split grid.z limit");
   dim3 grid((unsigned)(((S << lg) + 255) / 256), 1, (unsigned)(N * h));
   auto stream = c10::hip::getCurrentHIPStream();
   DDP_DISPATCH_FLOAT(x.scalar_type(), "head_split", [&] {
