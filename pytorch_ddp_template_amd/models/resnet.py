@@ -3,8 +3,11 @@
 Built for the BASELINE.json benchmark configs (ResNet-18 on CIFAR-shape
 3x32x32, ResNet-50 on ImageNet-shape 3x224x224).  Architecture follows the
 standard torchvision ResNet v1 definitions; layout is NHWC end-to-end and
-conv->bn->relu chains use the fused BN+ReLU epilogue, residual tails use the
-fused add+relu kernel.
+conv->bn->relu chains use the fused BN+ReLU epilogue; in the fused-block
+training path the residual add + final ReLU ride in the last BN's normalize
+epilogue (ops/fused_block.py).  The reference template has no conv models at
+all (FooModel is a 2-layer MLP, reference model.py:8-16) —
+these exist for the BASELINE.json benchmark configs.
 
 Two stems:
 * ``stem="cifar"`` — 3x3 s1 conv, no maxpool (the standard CIFAR ResNet stem;
