@@ -527,6 +527,11 @@ torch::Tensor qkv_pack(torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
   TORCH_CHECK(dh % 8 == 0 && lg >= 0, "qkv_pack needs pow2 dh/8");
   auto dqkv = torch::empty({N, S, (long long)3 * h * dh}, dq.options());
   TORCH_CHECK(N <= 65535 && S <= 65535, "qkv_pack grid limits");
+  // The kernel decodes which/hd with the magic reciprocal 65536/h + 1, exact
+  // only while (3h-1)*(65536/h + 1) < 65536*(t/h + 1) — holds for h <= 128;
+  // larger head counts would silently mis-route gradients.
+  TORCH_CHECK(h >= 1 && h <= 128, "qkv_pack: heads must be in [1,128], got ",
+              h);
   dim3 grid((unsigned)((((3 * h) << lg) + 255) / 256), (unsigned)S,
             (unsigned)N);
   auto stream = c10::hip::getCurrentHIPStream();

@@ -219,7 +219,11 @@ __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
   if (EXTRAS && stats_ws) {
     const long long vrow = (long long)flat_id * 2 + (wave >> 1);
     const long long wsrow = vrow % ws_nblocks;
-    const bool wrap = vrow >= ws_nblocks;
+    // If ANY block of this launch wraps, all blocks accumulate atomically
+    // into the pre-zeroed workspace (plain stores by row owners race with
+    // wrapped blocks' atomics — block execution order is unordered).
+    const bool wrap =
+        2ll * gridDim.x * gridDim.y > (long long)ws_nblocks || gridDim.z > 1;
 #pragma unroll
     for (int ni = 0; ni < NI32; ++ni) {
       float sv = col_sum[ni], qv = col_sq[ni];

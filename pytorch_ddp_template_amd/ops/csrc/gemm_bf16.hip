@@ -337,7 +337,12 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     // scale the workspace and its zero-fill/finalize with M)
     const long long vrow = (long long)flat_id * 2 + (wave >> 1);
     const long long wsrow = vrow % ws_nblocks;
-    const bool wrap = vrow >= ws_nblocks;
+    // If ANY vrow in this launch wraps (2*nwg > ws_nblocks), every block
+    // accumulates atomically into the pre-zeroed workspace: block execution
+    // order is not guaranteed, so a wrapped block's atomicAdd landing before
+    // the row owner's plain store would be silently overwritten.  (z-batched
+    // blocks share flat ids, so gridDim.z > 1 forces atomics too.)
+    const bool wrap = 2ll * nwg > (long long)ws_nblocks || gridDim.z > 1;
 #pragma unroll
     for (int ni = 0; ni < NI32; ++ni) {
       float sv = col_sum[ni], qv = col_sq[ni];

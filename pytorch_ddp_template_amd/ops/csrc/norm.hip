@@ -45,6 +45,10 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ ws,
   const long long rstride = nthreads / slots;
   float a0 = 0, a1 = 0, a2 = 0, a3 = 0, a4 = 0, a5 = 0, a6 = 0, a7 = 0;
   float q0 = 0, q1 = 0, q2 = 0, q3 = 0, q4 = 0, q5 = 0, q6 = 0, q7 = 0;
+  // When nthreads % slots != 0 the trailing threads have row0 == rstride and
+  // would re-accumulate rows already owned by the row0 == 0 threads (double
+  // counting).  They skip the sweep but still reach the barrier + epilogue.
+  if (row0 < rstride)
   for (long long m = row0; m < M; m += rstride) {
     const T* p = x + m * C + c0;
     float v[8];
@@ -145,6 +149,7 @@ __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
   const int c0 = (int)(tid % slots) * 8;
   const long long row0 = tid / slots;
   const long long rstride = nthreads / slots;
+  if (row0 >= rstride) return;  // trailing threads would duplicate rows
   float sc[8], sh[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
@@ -229,6 +234,9 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ dy,
     rs[j] = rstd[c0 + j];
   }
   float dg[8] = {}, db[8] = {};
+  // row0 >= rstride threads would double-count rows (nthreads % slots != 0);
+  // they skip the sweep but still reach the barrier + workspace store.
+  if (row0 < rstride)
   for (long long m = row0; m < M; m += rstride) {
     const T* pg = dy + m * C + c0;
     const T* px = x + m * C + c0;
@@ -474,6 +482,11 @@ __global__ void ln_bwd_param_kernel(const T* __restrict__ dy,
   const long long row0 = tid / slots;
   const long long rstride = nthreads / slots;
   float dg[8] = {}, db[8] = {};
+  // For ViT-B D=768 (slots=96) a 1024-block launch leaves 262144 % 96 = 64
+  // trailing threads with row0 == rstride that would re-accumulate rows
+  // ≡ 0 (mod rstride) — deterministic dgamma/dbeta overcount at production
+  // M.  Skip the sweep for them; they still join the barrier + store.
+  if (row0 < rstride)
   for (long long m = row0; m < M; m += rstride) {
     const float mu = mean[m], rs = rstd[m];
     const T* pg = dy + m * D + c0;

@@ -218,10 +218,6 @@ class _Conv2dFn(torch.autograd.Function):
             dw = ext.conv2d_wgrad(dy, x, stride, pad, w.shape[1], w.shape[2]).to(
                 w.dtype
             )
-            if cpad:
-                dw = dw[..., : w.shape[-1] - cpad].contiguous()
-                if dx is not None:
-                    dx = dx[..., : w.shape[-1] - cpad].contiguous()
             db = (
                 ext.col_sum(dy.reshape(-1, dy.shape[-1])).to(w.dtype)
                 if has_bias
@@ -238,6 +234,13 @@ class _Conv2dFn(torch.autograd.Function):
             dx = dxc.permute(0, 2, 3, 1).contiguous().to(x.dtype)
             dw = dwc.permute(0, 2, 3, 1).contiguous().to(w.dtype)
             db = dy.sum(dim=(0, 1, 2)).to(w.dtype) if has_bias else None
+        if cpad:
+            # saved x/w were channel-padded for the stem fast path: un-slice
+            # dw/dx back to the caller's channel count in BOTH branches (the
+            # fallback branch computed grads against the padded tensors too).
+            dw = dw[..., : w.shape[-1] - cpad].contiguous()
+            if dx is not None:
+                dx = dx[..., : w.shape[-1] - cpad].contiguous()
         return dx, dw, db, None, None, None
 
 
