@@ -37,9 +37,9 @@ import re
 
 import numpy as np
 import torch
-from torch.utils.data import DataLoader, RandomSampler
+from torch.utils.data import DataLoader
 
-from .data import ShardedSampler, build_dataset
+from .data import CudaPrefetcher, ShardedSampler, build_dataset
 from .models import build_model
 from .optim import SGD, clip_grad_norm_, get_linear_schedule_with_warmup
 from .ops import scale_grads_
@@ -143,13 +143,30 @@ def save_model(model, output_dir):
     torch.save(model_to_save.state_dict(), os.path.join(output_dir, "model.bin"))
 
 
-def save_checkpoint(args, model, optimizer, scheduler, global_step):
-    """Full checkpoint in the reference layout (ddp.py:255-277)."""
+def save_checkpoint(
+    args, model, optimizer, scheduler, global_step, epoch=0, batches_in_epoch=0
+):
+    """Full checkpoint in the reference layout (ddp.py:255-277), plus
+    ``training_state.pt`` (epoch / in-epoch position / RNG states) so a
+    resume continues mid-epoch without replaying data — the reference never
+    loaded at all."""
     output_dir = os.path.join(args.output_dir, f"checkpoint-{global_step}")
     save_model(model, output_dir)
     torch.save(args, os.path.join(output_dir, "training_args.bin"))
     torch.save(optimizer.state_dict(), os.path.join(output_dir, "optimizer.pt"))
     torch.save(scheduler.state_dict(), os.path.join(output_dir, "scheduler.pt"))
+    state = {
+        "epoch": epoch,
+        "batches_in_epoch": batches_in_epoch,
+        "rng": {
+            "torch": torch.get_rng_state(),
+            "numpy": np.random.get_state(),
+            "random": random.getstate(),
+        },
+    }
+    if torch.cuda.is_available() and args.n_gpu > 0:
+        state["rng"]["cuda"] = torch.cuda.get_rng_state_all()
+    torch.save(state, os.path.join(output_dir, "training_state.pt"))
     logger.info("Saved checkpoint.", dict(dir=output_dir, step=global_step))
 
 
@@ -198,6 +215,29 @@ def load_checkpoint(args, model, optimizer=None, scheduler=None, path=None):
     return step
 
 
+def load_training_state(args, path):
+    """Restore epoch / in-epoch position / RNG states saved by
+    save_checkpoint.  Returns (epoch, batches_in_epoch); (0, 0) when the
+    checkpoint predates training_state.pt."""
+    f = os.path.join(path, "training_state.pt")
+    if not os.path.exists(f):
+        return 0, 0
+    state = torch.load(f, map_location="cpu", weights_only=False)
+    rng = state.get("rng", {})
+    if "torch" in rng:
+        torch.set_rng_state(rng["torch"])
+    if "numpy" in rng:
+        np.random.set_state(rng["numpy"])
+    if "random" in rng:
+        random.setstate(rng["random"])
+    if "cuda" in rng and torch.cuda.is_available() and args.n_gpu > 0:
+        try:
+            torch.cuda.set_rng_state_all(rng["cuda"])
+        except Exception:  # device count mismatch across restarts
+            pass
+    return state.get("epoch", 0), state.get("batches_in_epoch", 0)
+
+
 class LossScaler:
     """fp16 loss scaling: static (``--loss_scale N``) or dynamic (N == 0,
     the reference's apex-style default — ddp.py:174-180 intent, natively)."""
@@ -240,15 +280,35 @@ def _cast_model(args, model):
 
 
 def evaluate(args, model, dataset=None, max_batches=None):
-    """Evaluation loop (the reference left this as a stub, ddp.py:123-124)."""
+    """Evaluation loop (the reference left this as a stub, ddp.py:123-124).
+
+    Distributed-correct: each rank evaluates its deterministic shard
+    (ShardedSampler, shuffle off, drop_last so no padded duplicates skew the
+    metrics), per-rank [loss_sum, n_samples, n_correct] are all-reduced, and
+    accuracy uses the TRUE sample count as denominator (round 1 divided by
+    n_batches * batch_size, wrong on a partial final batch).
+    """
     if dataset is None:
         dataset = build_dataset(args.dataset, size=min(args.dataset_size, 10000))
+    distributed = args.local_rank != -1 and torch.distributed.is_initialized()
+    sampler = (
+        ShardedSampler(
+            dataset, num_replicas=args.world_size, rank=args.node_rank,
+            shuffle=False, drop_last=True,
+        )
+        if distributed
+        else None
+    )
     loader = DataLoader(
-        dataset, batch_size=args.train_batch_size, pin_memory=args.device.type == "cuda"
+        dataset,
+        sampler=sampler,
+        batch_size=args.train_batch_size,
+        pin_memory=args.device.type == "cuda",
     )
     criterion = _criterion_for(args)
     model.eval()
-    total, n, correct = 0.0, 0, 0
+    # [loss_sum (sample-weighted), n_samples, n_correct] — one all-reduce
+    totals = torch.zeros(3, dtype=torch.float64)
     with torch.no_grad():
         for i, (x, y) in enumerate(loader):
             if max_batches is not None and i >= max_batches:
@@ -259,15 +319,20 @@ def evaluate(args, model, dataset=None, max_batches=None):
                 x = x.to(next(model.parameters()).dtype)
             out = model(x)
             loss = criterion(out.float() if out.dtype != torch.float32 else out, y)
-            total += float(loss)
-            n += 1
+            bs = x.shape[0]
+            totals[0] += float(loss) * bs
+            totals[1] += bs
             if y.dtype == torch.long:
-                correct += int((out.argmax(-1) == y).sum())
+                totals[2] += int((out.argmax(-1) == y).sum())
+    if distributed:
+        t = totals.to(args.device) if args.device.type == "cuda" else totals
+        torch.distributed.all_reduce(t)
+        totals = t.cpu()
     model.train()
-    avg = total / max(1, n)
-    result = {"eval_loss": avg}
-    if correct:
-        result["eval_acc"] = correct / (n * args.train_batch_size)
+    n_samples = max(1.0, float(totals[1]))
+    result = {"eval_loss": float(totals[0]) / n_samples}
+    if float(totals[2]) > 0:
+        result["eval_acc"] = float(totals[2]) / n_samples
     logger.info("Evaluation complete.", dict(**result))
     return result
 
@@ -295,7 +360,13 @@ def train(args, model):
             seed=args.seed,
         )
     else:
-        train_sampler = RandomSampler(dataset)
+        # epoch-seeded shuffle even single-process (world 1 shard = the whole
+        # set): unlike the reference's RandomSampler (global-RNG order,
+        # ddp.py:144-145) the data order is a pure function of (seed, epoch),
+        # which is what makes mid-epoch resume replay-free and exact.
+        train_sampler = ShardedSampler(
+            dataset, num_replicas=1, rank=0, seed=args.seed
+        )
     loader = DataLoader(
         dataset,
         sampler=train_sampler,
@@ -337,6 +408,7 @@ def train(args, model):
         )
 
     global_step = 0
+    resume_epoch, resume_batches = 0, 0
     if args.resume_from or args.global_step:
         path = args.resume_from
         if path is None and args.global_step:
@@ -345,6 +417,11 @@ def train(args, model):
                 path = None  # fall back to latest
         # optimizer/scheduler state (incl. last_epoch) comes from the files
         global_step = load_checkpoint(args, model, optimizer, scheduler, path)
+        if path is None:
+            path = find_latest_checkpoint(args.output_dir)
+        if path is not None and global_step > 0:
+            # mid-epoch position + RNG states (true resume, not data replay)
+            resume_epoch, resume_batches = load_training_state(args, path)
 
     logger.info(
         "***** Running training *****",
@@ -390,17 +467,36 @@ def train(args, model):
     epoch_iter = trange(
         int(args.num_train_epochs), desc="Epoch", disable=not show_bars
     )
+    # Double-buffered H2D prefetch on a dedicated copy stream (GPU): the
+    # next batch's pinned-host copy overlaps this batch's fwd/bwd.  CPU and
+    # --no_prefetch fall back to the reference's in-loop .to(device).
+    model_dtype = next(model.parameters()).dtype
+    use_prefetch = args.prefetch and args.device.type == "cuda"
+
     for epoch in epoch_iter:
+        if epoch < resume_epoch:
+            continue  # scheduler/optimizer state already restored
         if isinstance(train_sampler, ShardedSampler):
             train_sampler.set_epoch(epoch)  # reference ddp.py:213-214
-        step_iter = tqdm(loader, desc="Iteration", disable=not show_bars)
+        skip_batches = resume_batches if epoch == resume_epoch else 0
+        batches = (
+            CudaPrefetcher(loader, args.device, model_dtype)
+            if use_prefetch
+            else loader
+        )
+        step_iter = tqdm(
+            batches, desc="Iteration", total=len(loader), disable=not show_bars
+        )
         for step, (x, y) in enumerate(step_iter):
-            x = x.to(args.device, non_blocking=True)
-            y = y.to(args.device, non_blocking=True)
-            if x.dtype.is_floating_point:
-                x = x.to(next(model.parameters()).dtype)
-            if y.dtype.is_floating_point:
-                y = y.to(next(model.parameters()).dtype)
+            if step < skip_batches:
+                continue  # fast-forward to the mid-epoch resume point
+            if not use_prefetch:
+                x = x.to(args.device, non_blocking=True)
+                y = y.to(args.device, non_blocking=True)
+                if x.dtype.is_floating_point:
+                    x = x.to(model_dtype)
+                if y.dtype.is_floating_point:
+                    y = y.to(model_dtype)
 
             accum_boundary = (step + 1) % args.gradient_accumulation_steps == 0
             sync_ctx = (
@@ -479,11 +575,24 @@ def train(args, model):
                         step_iter.set_postfix(loss=window)
 
                 if (
+                    args.eval_steps > 0
+                    and global_step % args.eval_steps == 0
+                ):
+                    # all ranks participate (sharded eval + all-reduce)
+                    result = evaluate(args, model, max_batches=args.eval_max_batches)
+                    if tb_writer is not None:
+                        for k, v in result.items():
+                            tb_writer.add_scalar(k, v, global_step)
+
+                if (
                     is_main_process()
                     and args.save_steps > 0
                     and global_step % args.save_steps == 0
                 ):
-                    save_checkpoint(args, model, optimizer, scheduler, global_step)
+                    save_checkpoint(
+                        args, model, optimizer, scheduler, global_step,
+                        epoch=epoch, batches_in_epoch=step + 1,
+                    )
 
                 if args.max_steps > 0 and global_step >= args.max_steps:
                     done = True  # exact stop (reference's ddp.py:280 overran by one)
@@ -542,6 +651,15 @@ def build_parser():
     parser.add_argument("--num_workers", type=int, default=0)
     parser.add_argument("--no_tensorboard", action="store_true")
     parser.add_argument("--no_progress_bar", action="store_true")
+    parser.add_argument("--eval_steps", type=int, default=0,
+                        help="run evaluate() every N optimizer steps (0 = off; "
+                             "the reference's evaluate was an empty stub)")
+    parser.add_argument("--eval_max_batches", type=int, default=None)
+    parser.add_argument("--prefetch", dest="prefetch", action="store_true",
+                        default=True,
+                        help="double-buffered pinned-host H2D prefetch on a "
+                             "dedicated copy stream (GPU only)")
+    parser.add_argument("--no_prefetch", dest="prefetch", action="store_false")
     return parser
 
 

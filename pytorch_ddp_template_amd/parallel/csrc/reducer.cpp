@@ -18,22 +18,28 @@
 //    ROCm; gloo for the CPU plumbing rung) and overlaps with the rest of
 //    backward.  finalize() waits on all outstanding work and divides by the
 //    world size.
-//  * Collectives are issued directly from C++ through the c10d
-//    ProcessGroup, so the hot loop never re-enters Python between the
-//    autograd hook and the collective launch.
+//  * The per-parameter ready hooks are registered in C++ directly on the
+//    AccumulateGrad autograd nodes (install_hooks): the entire
+//    grad-ready → bucket-countdown → collective-launch path runs inside the
+//    autograd engine thread with no Python re-entry and no GIL.
 //
 // The reference implements none of this in its own code — it relies on
 // torch's C++ reducer; this file is the from-scratch native replacement.
 
 #include <torch/extension.h>
 
+#include <torch/csrc/autograd/function.h>
+#include <torch/csrc/autograd/utils/lambda_post_hook.h>
+#include <torch/csrc/autograd/variable.h>
 #include <torch/csrc/distributed/c10d/ProcessGroup.hpp>
 #include <torch/csrc/distributed/c10d/Work.hpp>
 
 #include <algorithm>
 #include <cstdint>
+#include <map>
 #include <memory>
 #include <string>
+#include <utility>
 #include <vector>
 
 namespace ddp_amd {
@@ -47,7 +53,7 @@ struct Bucket {
   c10::intrusive_ptr<c10d::Work> work;  // in-flight all-reduce
 };
 
-class Reducer {
+class Reducer : public std::enable_shared_from_this<Reducer> {
  public:
   Reducer(std::vector<at::Tensor> params,
           c10::intrusive_ptr<c10d::ProcessGroup> pg,
@@ -145,6 +151,31 @@ class Reducer {
   }
   bool sync() const { return sync_; }
 
+  // Register the per-parameter ready hooks directly on the AccumulateGrad
+  // autograd nodes — pure C++ path from grad-ready to collective launch
+  // (the round-1 version registered ~1 Python closure per parameter:
+  // 60–150 interpreter hops per backward).
+  void install_hooks() {
+    TORCH_CHECK(grad_accumulators_.empty(), "hooks already installed");
+    std::weak_ptr<Reducer> self = weak_from_this();
+    for (size_t i = 0; i < params_.size(); ++i) {
+      auto acc = torch::autograd::impl::grad_accumulator(params_[i]);
+      TORCH_CHECK(acc, "parameter ", i, " has no grad accumulator");
+      acc->add_post_hook(
+          std::make_unique<torch::autograd::utils::LambdaPostHook>(
+              [self, i](const torch::autograd::variable_list& outputs,
+                        const torch::autograd::variable_list& /*inputs*/)
+                  -> torch::autograd::variable_list {
+                if (auto r = self.lock())
+                  r->mark_ready(static_cast<int64_t>(i));
+                return outputs;
+              }));
+      // keep the nodes alive for the hooks' lifetime (same ownership
+      // pattern as torch's own reducer)
+      grad_accumulators_.push_back(std::move(acc));
+    }
+  }
+
   // Called from the per-parameter post-accumulate-grad hook.
   void mark_ready(int64_t param_index) {
     if (!sync_) return;  // grad-accumulation micro-batch (no_sync)
@@ -153,8 +184,21 @@ class Reducer {
                 "bad param index");
     const int64_t bi = bucket_of_[param_index];
     Bucket& b = buckets_[bi];
-    TORCH_CHECK(b.pending > 0, "gradient marked ready twice for bucket ", bi,
-                " (param ", param_index, ")");
+    if (b.pending == 0) {
+      // A second backward pass inside one sync window (shared-trunk /
+      // multi-loss graph).  If the bucket's all-reduce has NOT launched yet
+      // (strict-order launch can hold a ready bucket back), the extra
+      // contribution already accumulated into the flat view before any
+      // collective reads it — nothing to do.  Once the collective is in
+      // flight a late gradient would race it, so that stays a hard error
+      // with an actionable message.
+      TORCH_CHECK(
+          !b.work, "bucket ", bi, " received a gradient (param ", param_index,
+          ") after its all-reduce launched. For multiple backward passes per "
+          "optimizer step, run all but the last under "
+          "DistributedModel.no_sync() so gradients accumulate locally first.");
+      return;
+    }
     if (--b.pending == 0 && sync_) {
       // RCCL/NCCL requires every rank to issue collectives in the SAME
       // order.  Buckets are built in reverse registration order (the usual
@@ -208,17 +252,49 @@ class Reducer {
   }
 
   // Broadcast all parameters (and any extra tensors, e.g. buffers) from
-  // rank 0 — DDP-wrap-time semantics (reference ddp.py:194).
+  // rank 0 — DDP-wrap-time semantics (reference ddp.py:194).  Tensors are
+  // coalesced into ONE flat buffer per (dtype, device) group and the group
+  // broadcasts are issued asynchronously, then unflattened — the round-1
+  // version issued one BLOCKING broadcast per tensor (O(#params) latency at
+  // wrap time on 8 ranks).
   void broadcast_state(std::vector<at::Tensor> extra) {
     torch::NoGradGuard no_grad;
     std::vector<at::Tensor> all;
     for (auto& t : params_) all.push_back(t.detach());
     for (auto& t : extra) all.push_back(t.detach());
-    for (auto& t : all) {
-      std::vector<at::Tensor> one{t};
+    // group indices by (scalar type, device), preserving order
+    std::map<std::pair<int, std::string>, std::vector<size_t>> groups;
+    for (size_t i = 0; i < all.size(); ++i) {
+      groups[{static_cast<int>(all[i].scalar_type()), all[i].device().str()}]
+          .push_back(i);
+    }
+    struct InFlight {
+      at::Tensor flat;
+      std::vector<size_t> idx;
+      c10::intrusive_ptr<c10d::Work> work;
+    };
+    std::vector<InFlight> inflight;
+    for (auto& [key, idx] : groups) {
+      std::vector<at::Tensor> views;
+      views.reserve(idx.size());
+      for (size_t i : idx) views.push_back(all[i].reshape({-1}));
+      InFlight f;
+      f.flat = at::cat(views);
+      f.idx = idx;
+      std::vector<at::Tensor> one{f.flat};
       c10d::BroadcastOptions o;
       o.rootRank = 0;
-      pg_->broadcast(one, o)->wait();
+      f.work = pg_->broadcast(one, o);
+      inflight.push_back(std::move(f));
+    }
+    for (auto& f : inflight) {
+      f.work->wait();
+      int64_t off = 0;
+      for (size_t i : f.idx) {
+        auto& dst = all[i];
+        dst.copy_(f.flat.narrow(0, off, dst.numel()).view(dst.sizes()));
+        off += dst.numel();
+      }
     }
   }
 
@@ -251,6 +327,7 @@ class Reducer {
 
   std::vector<at::Tensor> params_;
   c10::intrusive_ptr<c10d::ProcessGroup> pg_;
+  std::vector<std::shared_ptr<torch::autograd::Node>> grad_accumulators_;
   std::vector<Bucket> buckets_;
   std::vector<int64_t> bucket_of_;
   std::vector<int64_t> slot_of_;
@@ -267,6 +344,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("params"), py::arg("process_group"),
            py::arg("first_bucket_bytes") = 1 << 20,
            py::arg("bucket_bytes") = 50 << 20)
+      .def("install_hooks", &ddp_amd::Reducer::install_hooks)
       .def("mark_ready", &ddp_amd::Reducer::mark_ready)
       .def("mark_unused", &ddp_amd::Reducer::mark_unused)
       .def("finalize", &ddp_amd::Reducer::finalize,

@@ -5,9 +5,12 @@ Wraps an nn.Module for one-process-per-GPU data parallelism over RCCL/xGMI
 machinery is the native C++ reducer (``parallel/csrc/reducer.cpp``); this
 wrapper only:
 
-* registers one post-accumulate-grad hook per parameter that calls into the
-  C++ reducer (gradients live as views into the reducer's flat buckets, so
-  there is no copy between autograd and the collective),
+* has the C++ reducer register its own per-parameter hooks directly on the
+  AccumulateGrad autograd nodes (``Reducer.install_hooks``) — grad-ready →
+  bucket countdown → collective launch all happens inside the autograd
+  engine thread with no Python re-entry (gradients live as views into the
+  reducer's flat buckets, so there is also no copy between autograd and the
+  collective),
 * optionally walks the autograd graph after forward to find parameters that
   did not participate (``find_unused_parameters`` — reference ddp.py:195
   passes True) and marks them ready,
@@ -75,21 +78,19 @@ class DistributedModel(nn.Module):
             int_buffers = [b for b in module.buffers() if not b.is_floating_point()]
             self.reducer.broadcast_state(buffers + int_buffers)
 
-        self._hooks = []
-        for i, p in enumerate(self._params):
-            self._hooks.append(
-                p.register_post_accumulate_grad_hook(self._make_hook(i))
-            )
-
-    def _make_hook(self, index: int):
-        def hook(param):  # noqa: ARG001
-            self.reducer.mark_ready(index)
-
-        return hook
+        # C++-side hooks on the AccumulateGrad nodes (no Python per-grad hops)
+        self.reducer.install_hooks()
 
     def forward(self, *args, **kwargs):
         out = self.module(*args, **kwargs)
-        if self.find_unused_parameters and self.reducer.sync():
+        # Only walk when a backward can actually follow: under no_grad()
+        # (evaluation) every param would look unused and the walk would
+        # launch collectives from an eval pass.
+        if (
+            self.find_unused_parameters
+            and self.reducer.sync()
+            and torch.is_grad_enabled()
+        ):
             self._mark_unused(out)
         return out
 

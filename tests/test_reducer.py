@@ -146,3 +146,55 @@ def test_second_iteration_works(pg):
         dm(torch.randn(4, 10)).sum().backward()
         dm.finish_gradient_sync()
         dm.zero_grad()
+
+
+def test_multi_loss_no_sync_pattern(pg):
+    """Shared-trunk / multi-loss graphs: all backward passes but the last run
+    under no_sync(); grads accumulate into the bucket views and the final
+    pass launches one all-reduce per bucket.  (Round-2 reducer hardening —
+    the round-1 reducer hard-crashed on any second backward in a window.)"""
+    torch.manual_seed(3)
+    trunk = nn.Linear(10, 10)
+    head_a = nn.Linear(10, 5)
+    head_b = nn.Linear(10, 5)
+
+    class TwoHead(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.trunk, self.a, self.b = trunk, head_a, head_b
+
+        def forward(self, x):
+            t = torch.relu(self.trunk(x))
+            return self.a(t), self.b(t)
+
+    m = TwoHead()
+    ref = TwoHead()
+    ref.load_state_dict(m.state_dict())
+    dm = DistributedModel(m)
+    x = torch.randn(8, 10)
+    ya, yb = dm(x)
+    with dm.no_sync():
+        ya.sum().backward(retain_graph=True)
+    yb.sum().backward()
+    dm.finish_gradient_sync()
+    ra, rb = ref(x)
+    (ra.sum() + rb.sum()).backward()
+    for p, q in zip(m.parameters(), ref.parameters()):
+        torch.testing.assert_close(p.grad, q.grad)
+
+
+def test_double_backward_after_launch_raises_actionable(pg):
+    """A late gradient for a bucket whose all-reduce already launched is a
+    loud error pointing at no_sync(), not a cryptic crash."""
+    torch.manual_seed(4)
+    m = small_model()
+    dm = DistributedModel(m)
+    x = torch.randn(4, 10)
+    out = dm(x)
+    out.sum().backward(retain_graph=True)
+    with pytest.raises(RuntimeError, match="no_sync"):
+        out.sum().backward()
+    # the reducer stays usable after the failed window
+    dm.zero_grad()
+    dm(x).sum().backward()
+    dm.finish_gradient_sync()
