@@ -123,11 +123,15 @@ def main():
     torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
 
-    # max over ranks (slowest rank defines the whole-job time)
+    # max over ranks (slowest rank defines the whole-job time); keep every
+    # rank's own wall time for the scale-run skew diagnosis
+    rank_times = [elapsed]
     if distributed:
         t = torch.tensor([elapsed], device=dev, dtype=torch.float64)
-        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
-        elapsed = float(t.item())
+        gathered = [torch.zeros_like(t) for _ in range(world_size)]
+        torch.distributed.all_gather(gathered, t)
+        rank_times = [float(g.item()) for g in gathered]
+        elapsed = max(rank_times)
 
     n_gpus = world_size if distributed else 1
     global_batch = args.batch * n_gpus
@@ -149,7 +153,10 @@ def main():
                     "vs_baseline": None,
                     "dtype": "bf16",
                     "data": "synthetic",
-                    "loss": float(last_loss) if last_loss is not None else None,
+                    "loss": (
+                        float(last_loss.detach()) if last_loss is not None else None
+                    ),
+                    "rank_times_s": [round(t, 4) for t in rank_times],
                     "config": {
                         "model": cfg_model,
                         "global_batch": global_batch,
