@@ -367,14 +367,32 @@ def train(args, model):
         train_sampler = ShardedSampler(
             dataset, num_replicas=1, rank=0, seed=args.seed
         )
-    loader = DataLoader(
-        dataset,
-        sampler=train_sampler,
-        batch_size=args.train_batch_size,
-        pin_memory=args.device.type == "cuda",
-        num_workers=args.num_workers,
-        drop_last=True,
-    )
+    if getattr(dataset, "batched_indexing", False):
+        # vectorized batch fetch: the BatchSampler hands the dataset a LIST
+        # of indices and __getitem__ gathers the whole batch in one indexing
+        # op — no per-sample fetch, no torch.stack collation.  At the MI355X
+        # operating point (8192 samples/GPU/step) the per-sample path is
+        # host-bound; this one keeps the DataLoader off the critical path.
+        from torch.utils.data import BatchSampler
+
+        loader = DataLoader(
+            dataset,
+            sampler=BatchSampler(
+                train_sampler, args.train_batch_size, drop_last=True
+            ),
+            batch_size=None,
+            pin_memory=args.device.type == "cuda",
+            num_workers=args.num_workers,
+        )
+    else:
+        loader = DataLoader(
+            dataset,
+            sampler=train_sampler,
+            batch_size=args.train_batch_size,
+            pin_memory=args.device.type == "cuda",
+            num_workers=args.num_workers,
+            drop_last=True,
+        )
 
     if args.max_steps > 0:
         t_total = args.max_steps
