@@ -73,6 +73,9 @@ struct AddReluOp {
     return fmaxf(a + b, 0.f);
   }
 };
+struct AddOp {
+  DEV_INLINE float operator()(float a, float b) const { return a + b; }
+};
 struct ReluBwdOp {  // (dy, y) -> dy * (y > 0)
   DEV_INLINE float operator()(float dy, float y) const {
     return y > 0.f ? dy : 0.f;
@@ -341,6 +344,21 @@ torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {
     launch_binary<scalar_t>(dy, y, dx, ew::ReluBwdOp{});
   });
   return dx;
+}
+
+// a += b, in place (residual skip-grad accumulation in the fused blocks —
+// one read fewer than the eager out-of-place a + b the round-1 path used,
+// and the whole r18 backward's kernel time stays in-tree)
+torch::Tensor add_(torch::Tensor a, torch::Tensor b) {
+  CHECK_GPU(a);
+  CHECK_GPU(b);
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous() && a.numel() == b.numel(),
+              "add_: contiguous same-numel tensors only");
+  TORCH_CHECK(a.scalar_type() == b.scalar_type(), "add_: dtype mismatch");
+  DDP_DISPATCH_FLOAT(a.scalar_type(), "add_", [&] {
+    launch_binary<scalar_t>(a, b, a, ew::AddOp{});
+  });
+  return a;
 }
 
 torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b) {

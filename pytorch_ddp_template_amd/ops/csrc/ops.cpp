@@ -15,6 +15,7 @@
 torch::Tensor relu_fwd(torch::Tensor x);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
 torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b);
+torch::Tensor add_(torch::Tensor a, torch::Tensor b);
 torch::Tensor gelu_fwd(torch::Tensor x);
 torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x);
 torch::Tensor col_sum(torch::Tensor dy);
@@ -291,6 +292,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_fwd", &relu_fwd);
   m.def("relu_bwd", &relu_bwd);
   m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("add_", &add_);
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
   m.def("col_sum", &col_sum);

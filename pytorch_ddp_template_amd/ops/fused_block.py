@@ -129,7 +129,9 @@ class _FusedBasicBlockFn(torch.autograd.Function):
             dw1 = ext.conv2d_wgrad(dt1, x, stride, 1, 3, 3).to(w1.dtype)
 
         if wd is None:
-            dx = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W) + g
+            # in-place in-house add (the skip grad rides the dgrad output
+            # buffer; round 1 used an eager out-of-place torch add here)
+            dx = ext.add_(ext.conv2d_dgrad(dt1, w1, stride, 1, H, W), g)
             dwd = dgd = dbd = None
         else:
             dxa = ext.conv2d_dgrad(dt1, w1, stride, 1, H, W)
@@ -141,7 +143,7 @@ class _FusedBasicBlockFn(torch.autograd.Function):
                     dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
             else:
                 dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
-            dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W) + dxa
+            dx = ext.add_(ext.conv2d_dgrad(dtd, wd, stride, 0, H, W), dxa)
         if use_side:
             # weight grads are consumed (bucket accumulation) on the main
             # stream after this node returns
@@ -236,14 +238,14 @@ class _FusedBottleneckFn(torch.autograd.Function):
         dw1 = ext.conv2d_wgrad(dt1, x, 1, 0, 1, 1).to(w1.dtype)
 
         if wd is None:
-            dx = ext.conv2d_dgrad(dt1, w1, 1, 0, H, W) + g
+            dx = ext.add_(ext.conv2d_dgrad(dt1, w1, 1, 0, H, W), g)
             dwd = dgd = dbd = None
         else:
             dxa = ext.conv2d_dgrad(dt1, w1, 1, 0, H, W)
             dtd, dgd, dbd = ext.bn_bwd(g2d, _flat(td), gd, meand, rstdd, None)
             dtd = dtd.reshape(td.shape)
             dwd = ext.conv2d_wgrad(dtd, x, stride, 0, 1, 1).to(wd.dtype)
-            dx = ext.conv2d_dgrad(dtd, wd, stride, 0, H, W) + dxa
+            dx = ext.add_(ext.conv2d_dgrad(dtd, wd, stride, 0, H, W), dxa)
         return (dx, dw1, dg1, db1, dw2, dg2, db2, dw3, dg3, db3,
                 dwd, dgd, dbd, None, None, None, None, None)
 

@@ -553,3 +553,16 @@ def test_conv2d_cpad_fallback_backward_shapes():
         Fn.use_native = orig
     assert w.grad.shape == w.shape, f"dw shape {w.grad.shape} vs {w.shape}"
     assert x.grad.shape == x.shape, f"dx shape {x.grad.shape} vs {x.shape}"
+
+
+def test_add_inplace():
+    """ew::add_ (in-house in-place add — the residual skip-grad accumulation
+    path in the fused blocks)."""
+    for n in (64, 1000, 8192 * 33 + 5):
+        a = t32(n, seed=70).to(torch.bfloat16)
+        b = t32(n, seed=71).to(torch.bfloat16)
+        ref = (a.float() + b.float()).to(torch.bfloat16)
+        ad = a.to(DEV)
+        out = EXT.add_(ad, b.to(DEV))
+        assert out.data_ptr() == ad.data_ptr()
+        assert torch.equal(out.cpu(), ref)
