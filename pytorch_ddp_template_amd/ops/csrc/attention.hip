@@ -644,6 +644,191 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_kernel(
 #undef LDSB
 }
 
+// ===================== flash-style attention FORWARD ======================
+//
+// The round-1 forward (attn_fwd_kernel above) gives one workgroup a whole
+// (n, head) and loops 32-row q-tiles with two-pass softmax: PMC showed it
+// wait-bound at 4.3x SQ_BUSY (~45 barriers/block gating ~16 MFMAs each).
+// This version is built in the backward kernel's mold — grid over 64-row
+// q-tiles (4x the blocks), looping 64-row K/V tiles with ONLINE softmax
+// (running row max + rescaled PV accumulator, flash style), ~5 barriers
+// per k-tile.  Also removes the S <= 224 cap (K/V are tiled, not staged
+// whole) and writes the same stats the flash backward consumes.
+// Wave w owns the (w>>1, w&1) 32x32 quadrant of each 64x64 product.
+template <typename T16>
+__global__ __launch_bounds__(256, 2) void attn_fwd_flash_kernel(
+    const T16* __restrict__ qkv, T16* __restrict__ out,
+    float* __restrict__ stats, const T16* __restrict__ zpad, int N, int S,
+    int H, float scale) {
+  __shared__ __attribute__((aligned(16))) char smem[4 * kImg * 2 +
+                                                    2 * 128 * 4];
+  T16* ldsQs = reinterpret_cast<T16*>(smem);
+  T16* ldsKs = ldsQs + kImg;
+  T16* ldsVn = ldsKs + kImg;
+  T16* ldsPs = ldsVn + kImg;
+  float* redM = reinterpret_cast<float*>(ldsPs + kImg);  // [64 rows][2 ki]
+  float* redS = redM + 128;                              // [64 rows][2 ki]
+
+  using vec16 = typename MM<T16>::vec;
+  const int q0 = blockIdx.x * kBT;
+  const int n = blockIdx.z;
+  const int hh = blockIdx.y;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int D = H * kDh;
+  const long long bh = (long long)n * H + hh;
+  const int r32 = lane & 31;
+  const int ks = lane >> 5;
+  const int r8 = lane >> 3;
+  const int p16 = (lane & 7) * 8;
+  const int qi = wave >> 1;  // q half (rows)
+  const int ki = wave & 1;   // k half for S; dh half for PV
+
+  const unsigned tr_off =
+      (unsigned)((ks * 8 + ((lane & 15) >> 2)) * (kDh * 2) + (lane & 3) * 8);
+  const int img_sel = (lane >> 4) & 1;
+#define LDSB(p)                                               \
+  ((unsigned)(unsigned long long)(__attribute__((            \
+      address_space(3))) const T16*)(p))
+  auto tr_frag = [&](const T16* img, int sum0, int c0) {
+    const T16* p = img + sum0 * kDh + (c0 + img_sel * 16);
+    const unsigned b0 = LDSB(p) + tr_off;
+    v4s l0, h0;
+    asm volatile(
+        "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+        "ds_read_b64_tr_b16 %1, %2 offset:512\n\t"
+        "s_waitcnt lgkmcnt(0)"
+        : "=&v"(l0), "=&v"(h0)
+        : "v"(b0));
+    vec16 b;
+    reinterpret_cast<v4s*>(&b)[0] = l0;
+    reinterpret_cast<v4s*>(&b)[1] = h0;
+    return b;
+  };
+  auto sw_frag = [&](const T16* img, int row, int kc) {
+    return *reinterpret_cast<const vec16*>(img + row * kDh +
+                                           kqswz(kc * 2 + ks, row));
+  };
+
+  // ---- stage the Q tile (swizzled) once ----
+  for (int u = wave; u < kBT / 8; u += 4) {
+    const int row = q0 + u * 8 + r8;
+    const T16* src = zpad;
+    if (row < S)
+      src = qkv + (((long long)n * S + row) * 3 + 0) * D + hh * kDh +
+            kqswz(lane, r8);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(ldsQs + u * 8 * kDh),
+        16, 0, 0);
+  }
+
+  f32x16 oacc = {};
+  float mrow[16], srow[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    mrow[r] = -3.0e38f;
+    srow[r] = 0.f;
+  }
+
+  const int n_kt = (S + kBT - 1) / kBT;
+  for (int kt = 0; kt < n_kt; ++kt) {
+    __syncthreads();  // previous iteration's K/V/P readers done
+    const int k0 = kt * kBT;
+    for (int u = wave; u < 2 * (kBT / 8); u += 4) {
+      const int img = u / (kBT / 8);  // 0=Ks 1=Vn
+      const int r0 = (u % (kBT / 8)) * 8;
+      const int row = k0 + r0 + r8;
+      const int which = img == 0 ? 1 : 2;
+      const int chunk = img == 0 ? kqswz(lane, r8) : p16;
+      const T16* src = zpad;
+      if (row < S)
+        src = qkv + (((long long)n * S + row) * 3 + which) * D + hh * kDh +
+              chunk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (img == 0 ? ldsKs : ldsVn) + r0 * kDh),
+          16, 0, 0);
+    }
+    __syncthreads();  // staging visible (barrier drains glds)
+
+    // ---- S quadrant (qi, ki) ----
+    f32x16 accS = {};
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {
+      vec16 a = sw_frag(ldsQs, qi * 32 + r32, kc);
+      vec16 b = sw_frag(ldsKs, ki * 32 + r32, kc);
+      accS = MM<T16>::mma32(a, b, accS);
+    }
+
+    // ---- online softmax: per-row tile max within this ki half (32-lane
+    // butterfly), cross-ki combine via LDS, exp + tile sums, rescale ----
+    const int colk = k0 + ki * 32 + r32;
+    const bool colok = colk < S;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      float v = colok ? accS[r] * scale : -3.0e38f;
+      accS[r] = v;
+#pragma unroll
+      for (int w = 16; w >= 1; w >>= 1) v = fmaxf(v, __shfl_xor(v, w));
+      if (r32 == 0) {
+        const int rowl = qi * 32 + (r & 3) + 8 * (r >> 2) + 4 * ks;
+        redM[rowl * 2 + ki] = v;
+      }
+    }
+    __syncthreads();  // both ki halves' tile maxes visible
+    float alpha[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int rowl = qi * 32 + (r & 3) + 8 * (r >> 2) + 4 * ks;
+      const float mnew =
+          fmaxf(mrow[r], fmaxf(redM[rowl * 2], redM[rowl * 2 + 1]));
+      alpha[r] = mrow[r] == mnew ? 1.f : __expf(mrow[r] - mnew);
+      mrow[r] = mnew;
+      const float p = colok ? __expf(accS[r] - mnew) : 0.f;
+      accS[r] = p;  // now holds this half's P values
+      float s = p;
+#pragma unroll
+      for (int w = 16; w >= 1; w >>= 1) s += __shfl_xor(s, w);
+      if (r32 == 0) redS[rowl * 2 + ki] = s;  // this half's TILE sum
+      // P tile into LDS (swizzled) for the PV A-operand
+      const int c = ki * 32 + r32;
+      ldsPs[rowl * kDh + kqswz(c >> 3, rowl) + (c & 7)] = to_t<T16>(p);
+    }
+    __syncthreads();  // P + both sum halves visible
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int rowl = qi * 32 + (r & 3) + 8 * (r >> 2) + 4 * ks;
+      srow[r] = srow[r] * alpha[r] + redS[rowl * 2] + redS[rowl * 2 + 1];
+      oacc[r] *= alpha[r];
+    }
+    // ---- PV quadrant (qi, di = ki): oacc += P · V (unnormalized) ----
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {
+      vec16 a = sw_frag(ldsPs, qi * 32 + r32, kc);
+      vec16 b = tr_frag(ldsVn, kc * 16, ki * 32);
+      oacc = MM<T16>::mma32(a, b, oacc);
+    }
+  }
+
+  // ---- finalize: out = oacc / srow, stats = (m, 1/sum) per row ----
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int rowl = qi * 32 + (r & 3) + 8 * (r >> 2) + 4 * ks;
+    const int row = q0 + rowl;
+    if (row >= S) continue;
+    const float inv = 1.f / srow[r];
+    out[((long long)n * S + row) * D + hh * kDh + ki * 32 + r32] =
+        to_t<T16>(oacc[r] * inv);
+    if (ki == 0 && r32 == 0) {
+      stats[bh * S + row] = mrow[r];
+      stats[((long long)N * H + bh) * S + row] = inv;
+    }
+  }
+#undef LDSB
+}
+
 // D = rowsum(dO ∘ O) per (n, head, s) — one wave per (n, s, head) row.
 // dout/out are model-layout [N, S, H*dh]; Dsum is [N*H][S] (the backward's
 // per-(n,head) indexing).
@@ -793,7 +978,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
   TORCH_CHECK(D3 % (3 * H) == 0);
   const int dh = D3 / (3 * H);
   TORCH_CHECK(dh == attn::kDh, "fused attention supports head dim 64");
-  TORCH_CHECK(S <= attn::kMaxSP, "S too large for the fused kernel");
+  // the flash forward tiles K/V (no S cap); only the P-materializing
+  // two-pass kernel stages the whole K/V and is capped
+  TORCH_CHECK(!want_p || S <= attn::kMaxSP,
+              "S too large for the P-materializing kernel");
   auto out = torch::empty({N, S, (long long)H * dh}, qkv.options());
   auto P = want_p
                ? torch::empty({(long long)N * H, S, S}, qkv.options())
@@ -803,19 +991,30 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t heads,
   static torch::Tensor zp;
   if (!zp.defined() || zp.device() != qkv.device())
     zp = torch::zeros({64}, qkv.options());
-  dim3 grid(1, H, N);  // q-tiles looped in-kernel (K/V staged once)
   auto stream = c10::hip::getCurrentHIPStream();
   DDP_DISPATCH_FLOAT(qkv.scalar_type(), "attn_fwd", [&] {
     if constexpr (!std::is_same_v<scalar_t, float>) {
-      hipLaunchKernelGGL((attn::attn_fwd_kernel<scalar_t>), grid, dim3(256),
-                         0, stream,
-                         reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
-                         reinterpret_cast<scalar_t*>(out.data_ptr()),
-                         want_p ? reinterpret_cast<scalar_t*>(P.data_ptr())
-                                : nullptr,
-                         stats.data_ptr<float>(),
-                         reinterpret_cast<const scalar_t*>(zp.data_ptr()), N,
-                         S, H, (float)scale);
+      if (!want_p) {
+        // flash forward: grid over 64-row q-tiles, online softmax, no P
+        dim3 grid((unsigned)((S + attn::kBT - 1) / attn::kBT), H, N);
+        hipLaunchKernelGGL((attn::attn_fwd_flash_kernel<scalar_t>), grid,
+                           dim3(256), 0, stream,
+                           reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
+                           reinterpret_cast<scalar_t*>(out.data_ptr()),
+                           stats.data_ptr<float>(),
+                           reinterpret_cast<const scalar_t*>(zp.data_ptr()),
+                           N, S, H, (float)scale);
+      } else {
+        dim3 grid(1, H, N);  // q-tiles looped in-kernel (K/V staged once)
+        hipLaunchKernelGGL((attn::attn_fwd_kernel<scalar_t>), grid, dim3(256),
+                           0, stream,
+                           reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
+                           reinterpret_cast<scalar_t*>(out.data_ptr()),
+                           reinterpret_cast<scalar_t*>(P.data_ptr()),
+                           stats.data_ptr<float>(),
+                           reinterpret_cast<const scalar_t*>(zp.data_ptr()),
+                           N, S, H, (float)scale);
+      }
     } else {
       TORCH_CHECK(false, "attn_fwd: bf16/f16 only");
     }

@@ -594,7 +594,10 @@ def test_attn_flash_bwd_matches_composed():
         b = qkv.clone().requires_grad_(True)
         out_f = attention_qkv(a, h, 1.0 / 8.0)
         out_c = attention_qkv_composed(b, h, 1.0 / 8.0)
-        assert torch.equal(out_f.cpu(), out_c.cpu()), "fwd paths must agree"
+        # flash fwd (online softmax, k-tiled PV accumulation order) vs the
+        # two-pass fwd: same math, different summation order
+        fd = (out_f.float() - out_c.float()).abs().max()
+        assert fd < 0.02, f"fwd paths diverge: {float(fd)}"
         g = torch.randn_like(out_f)
         out_f.backward(g)
         out_c.backward(g)
