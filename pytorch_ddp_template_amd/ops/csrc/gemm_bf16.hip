@@ -549,13 +549,19 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 // batch, % zsplit the split-M slice) — ViT attention backward runs ~1.5K
 // small TN GEMMs per call and a host-side per-batch launch loop was 70%%
 // of its step time.
-// TI/TJ: MFMA fragments per wave along I/J — 2 gives a 64x64 block tile
+// TI/TJ: MFMA fragments along I/J for the whole BLOCK tile — 2 gives 64x64
 // (the conv wgrad default; TAPS=9 already fills the accumulators), 4 gives
 // 128x128 (TAPS=1 large plain GEMMs, e.g. ViT linear weight grads, where
-// 64x64 tiles were instruction-bound at 4 MFMAs per staged chunk).
+// 64x64 tiles were instruction-bound at 4 MFMAs per staged chunk), 8 with
+// WI=2/WJ=4 gives 256x256 on 8 waves (512 threads) — the r2 route for the
+// big ViT shapes: at 128x128 each operand is re-read ~6x from HBM and the
+// kernel sits on that traffic roofline (~450-510 TF measured, and rocBLAS
+// lands in the same band); 256x256 halves the re-reads.
+// WI/WJ: the wave grid (WI*WJ waves per block) splitting the block tile.
 template <typename T16, int TAPS, int MODE = MODE_CONV, int TI = 2,
-          int TJ = 2>
-__global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
+          int TJ = 2, int WI = 2, int WJ = 2>
+__global__ __launch_bounds__(WI * WJ * 64, WI * WJ > 4 ? 1 : 2)
+void gemm_wgrad_tr_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
     float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
     int I /*Kout*/, int J, long long ldc, ConvMeta cm, long long sA,
@@ -584,15 +590,16 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   const int ch1 = min(n_chunks, ch0 + per_z);
   if (ch0 >= ch1) return;  // uniform per block
 
+  constexpr int NW = WI * WJ;  // waves per block
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = (wave >> 1) * (BI / 2), wn = (wave & 1) * (BJ / 2);
+  const int wm = (wave / WJ) * (BI / WI), wn = (wave % WJ) * (BJ / WJ);
   const int fr = lane & 15;
   const int fs = lane >> 4;
   const int sm = lane >> 1;        // staged m row this lane covers
   const int sh8 = (lane & 1) * 8;  // 8-channel half within the 32-B row
 
-  f32x16 acc[TI / 2][TJ / 2][TAPS] = {};  // 32x32 tiles
+  f32x16 acc[TI / WI][TJ / WJ][TAPS] = {};  // 32x32 tiles
 
   // magic reciprocal for the CONVJ tap -> (r,s) decode (S <= 256)
   const int rcpS = MODE == MODE_PLAIN ? 0 : 65536 / cm.S + 1;
@@ -612,8 +619,8 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
       wb = wo * cm.stride - cm.pad;
     }
     T16* base = lds + buf * (TILE_A + TAPS * TILE_B);
-    // wave w issues units u = w, w+4, ... of the NUNITS image stages
-    for (int u = wave; u < NUNITS; u += 4) {
+    // wave w issues units u = w, w+NW, ... of the NUNITS image stages
+    for (int u = wave; u < NUNITS; u += NW) {
       const T16* src = zpad;
       T16* dst;
       if (u < AIMGS) {
@@ -674,7 +681,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   stage(0, ch0);
   __syncthreads();
 
-  constexpr int TI32 = TI / 2, TJ32 = TJ / 2;
+  constexpr int TI32 = TI / WI, TJ32 = TJ / WJ;
   float bsum[TI32] = {};
   for (int ch = ch0; ch < ch1; ++ch) {
     const int buf = (ch - ch0) & 1;
@@ -705,7 +712,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
       reinterpret_cast<v4s*>(&af[mi][0])[1] = h0;
       reinterpret_cast<v4s*>(&af[mi][1])[0] = l1;
       reinterpret_cast<v4s*>(&af[mi][1])[1] = h1;
-      if (colsum && blockIdx.x == 0 && (wave & 1) == 0) {
+      if (colsum && blockIdx.x == 0 && wn == 0) {
         // bias grad rides along: the staged dy tile is already in
         // registers; out-of-range m/i lanes were zero-page staged.
         // Only one J-tile column of blocks and only wn==0 waves contribute
@@ -756,7 +763,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   // ---- writeback: f32 atomics into dw[I][taps][Cin] slices ----
   // (static tap order: a runtime-rotated index into acc would demote the
   // accumulators to scratch)
-  if (colsum && blockIdx.x == 0 && (wave & 1) == 0) {
+  if (colsum && blockIdx.x == 0 && wn == 0) {
 #pragma unroll
     for (int mi = 0; mi < TI32; ++mi) {
       float v = bsum[mi] + __shfl_xor(bsum[mi], 32);  // join the ks pair
@@ -770,11 +777,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_tr_kernel(
   for (int tap = 0; tap < TAPS; ++tap) {
     const long long coff = (long long)tap * Cin;
 #pragma unroll
-    for (int ni = 0; ni < TJ / 2; ++ni) {
+    for (int ni = 0; ni < TJ32; ++ni) {
       const int col = j0 + wn + ni * 32 + (lane & 31);
       if (col >= Cin) continue;
 #pragma unroll
-      for (int mi = 0; mi < TI / 2; ++mi)
+      for (int mi = 0; mi < TI32; ++mi)
 #pragma unroll
         for (int reg = 0; reg < 16; ++reg) {
           const int row =
@@ -1073,13 +1080,35 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   // backward has ~1.5K batches; a per-batch host loop dominated its step)
   if (bsz > 1) z = std::max(1, std::min(n_chunks, (int)(2048 / bsz) + 1));
   const bool wide = I >= 128 && J >= 128;  // 128x128 tiles for big GEMMs
+  // 256x256 tiles on 8 waves for the really big shapes: halves the per-
+  // operand HBM re-reads (the 128x128 traffic roofline measured ~450-510 TF
+  // on the ViT wgrad shapes — the same band rocBLAS hits)
+  const bool xwide = I >= 256 && J >= 256 && M >= 4096;
   if (wide) {
     tiles = ((J + 127) / 128) * ((I + 127) / 128);
     z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
     if (bsz > 1) z = std::max(1, std::min(n_chunks, (int)(2048 / bsz) + 1));
   }
+  if (xwide) {
+    // one 512-thread WG per CU: target ~256 resident blocks (one full round)
+    tiles = ((J + 255) / 256) * ((I + 255) / 256);
+    z = std::max(1, std::min(n_chunks, (256 + tiles - 1) / std::max(1, tiles)));
+    if (bsz > 1) z = std::max(1, std::min(n_chunks, (int)(1024 / bsz) + 1));
+  }
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
+    if (xwide) {
+      dim3 grid((J + 255) / 256, (I + 255) / 256, (unsigned)(bsz * z));
+      hipLaunchKernelGGL(
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 8, 8, 2, 4>),
+          grid, dim3(512), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()),
+          reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J,
+          /*ldc=*/J, cm, (long long)M * I, (long long)M * J,
+          (long long)I * J, z);
+      return;
+    }
     if (wide) {
       dim3 grid((J + 127) / 128, (I + 127) / 128, (unsigned)(bsz * z));
       hipLaunchKernelGGL(
@@ -1123,12 +1152,27 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
   int n_chunks = (M + 31) / 32;
   g16::ConvMeta cm{};
   auto& zp = zero_page(A.device(), A.scalar_type());
+  const bool xwide = I >= 256 && J >= 256 && M >= 4096;
   const bool wide = I >= 128 && J >= 128;
-  int tiles = wide ? ((J + 127) / 128) * ((I + 127) / 128)
-                   : ((J + 63) / 64) * ((I + 63) / 64);
-  int z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
+  int tiles = xwide ? ((J + 255) / 256) * ((I + 255) / 256)
+              : wide ? ((J + 127) / 128) * ((I + 127) / 128)
+                     : ((J + 63) / 64) * ((I + 63) / 64);
+  int z = std::max(
+      1, std::min(n_chunks,
+                  ((xwide ? 256 : 512) + tiles - 1) / std::max(1, tiles)));
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
+    if (xwide) {
+      dim3 grid((J + 255) / 256, (I + 255) / 256, (unsigned)z);
+      hipLaunchKernelGGL(
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 8, 8, 2, 4>),
+          grid, dim3(512), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()),
+          reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J, J, cm, 0, 0,
+          0, z, db.data_ptr<float>());
+      return;
+    }
     if (wide) {
       dim3 grid((J + 127) / 128, (I + 127) / 128, (unsigned)z);
       hipLaunchKernelGGL(

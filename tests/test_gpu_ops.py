@@ -79,7 +79,10 @@ def test_gemm_nt_f32(m, n, k):
 
 
 @pytest.mark.parametrize("m,i,j", [(256, 64, 64), (1000, 70, 33), (64, 10, 512),
-                                   (512, 256, 384), (2048, 768, 768)])
+                                   (512, 256, 384), (2048, 768, 768),
+                                   # 256x256-tile 8-wave route (xwide)
+                                   (8192, 768, 768), (8192, 2304, 768),
+                                   (5000, 300, 260)])
 def test_gemm_tn_bf16(m, i, j):
     a = t32(m, i, seed=8).to(torch.bfloat16)
     b = t32(m, j, seed=9).to(torch.bfloat16)
