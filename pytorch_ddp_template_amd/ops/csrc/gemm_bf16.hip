@@ -558,8 +558,14 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 // kernel sits on that traffic roofline (~450-510 TF measured, and rocBLAS
 // lands in the same band); 256x256 halves the re-reads.
 // WI/WJ: the wave grid (WI*WJ waves per block) splitting the block tile.
+// NBUF: LDS staging depth.  3 enables counted vmcnt waits (stage chunk n+2
+// while chunk n computes; the barrier only waits for loads issued one chunk
+// back) — essential at 1 workgroup/CU where __syncthreads()'s vmcnt(0)
+// would drain the just-prefetched chunk and expose full HBM latency every
+// 32 m-rows.  2 keeps the original drain-at-barrier behavior (fine for the
+// conv variants where 2 co-resident WGs hide each other's stalls).
 template <typename T16, int TAPS, int MODE = MODE_CONV, int TI = 2,
-          int TJ = 2, int WI = 2, int WJ = 2>
+          int TJ = 2, int WI = 2, int WJ = 2, int NBUF = 2>
 __global__ __launch_bounds__(WI * WJ * 64, WI * WJ > 4 ? 1 : 2)
 void gemm_wgrad_tr_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
@@ -572,9 +578,12 @@ void gemm_wgrad_tr_kernel(
   constexpr int AIMGS = BI / 16, BIMGS = BJ / 16;
   constexpr int TILE_A = AIMGS * IMG, TILE_B = BIMGS * IMG;
   constexpr int NUNITS = AIMGS + TAPS * BIMGS;  // glds image stages / chunk
+  static_assert(NBUF == 2 || NUNITS % (WI * WJ) == 0,
+                "counted vmcnt waits need a uniform per-wave glds count");
   typedef short v4s __attribute__((ext_vector_type(4)));
   using vec16 = typename M16<T16>::vec;
-  __shared__ __attribute__((aligned(16))) T16 lds[2 * (TILE_A + TAPS * TILE_B)];
+  __shared__ __attribute__((aligned(16)))
+      T16 lds[NBUF * (TILE_A + TAPS * TILE_B)];
 
   const int Cin = J;
   const int i0 = blockIdx.y * BI;
@@ -678,15 +687,42 @@ void gemm_wgrad_tr_kernel(
   ((unsigned)(unsigned long long)(__attribute__((            \
       address_space(3))) const T16*)(p))
 
+  constexpr int PER_WAVE_GLDS = (NUNITS + WI * WJ - 1) / (WI * WJ);
+#define WG_WAIT_STAGE()                                        \
+  do {                                                         \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");         \
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PER_WAVE_GLDS)    \
+                 : "memory");                                  \
+  } while (0)
+#define WG_WAIT_ALL()                                          \
+  do {                                                         \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");         \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");           \
+  } while (0)
+
   stage(0, ch0);
-  __syncthreads();
+  if (NBUF == 3) {
+    if (ch0 + 1 < ch1) {
+      stage(1, ch0 + 1);
+      WG_WAIT_STAGE();
+    } else {
+      WG_WAIT_ALL();
+    }
+    asm volatile("s_barrier" ::: "memory");
+  } else {
+    __syncthreads();
+  }
 
   constexpr int TI32 = TI / WI, TJ32 = TJ / WJ;
   float bsum[TI32] = {};
   for (int ch = ch0; ch < ch1; ++ch) {
-    const int buf = (ch - ch0) & 1;
+    const int buf = (ch - ch0) % NBUF;
     const bool more = ch + 1 < ch1;
-    if (more) stage(buf ^ 1, ch + 1);  // glds latency hides under the MFMAs
+    if (NBUF == 3) {
+      if (ch + 2 < ch1) stage((ch - ch0 + 2) % 3, ch + 2);
+    } else if (more) {
+      stage(buf ^ 1, ch + 1);  // glds latency hides under the MFMAs
+    }
     const T16* base = lds + buf * (TILE_A + TAPS * TILE_B);
 
     // A fragments (dy): per 32-col tile mi, per 16-m half kh
@@ -755,9 +791,21 @@ void gemm_wgrad_tr_kernel(
         }
       }
     }
-    if (more) __syncthreads();  // drains the glds (vmcnt) + publishes buf^1
+    if (more) {
+      if (NBUF == 3) {
+        // only wait for the loads issued one chunk back; the chunk+2
+        // prefetch keeps flying through the barrier
+        if (ch + 2 < ch1) WG_WAIT_STAGE();
+        else WG_WAIT_ALL();
+        asm volatile("s_barrier" ::: "memory");
+      } else {
+        __syncthreads();  // drains the glds (vmcnt) + publishes buf^1
+      }
+    }
   }
 
+#undef WG_WAIT_STAGE
+#undef WG_WAIT_ALL
 #undef LDS_BYTE
 
   // ---- writeback: f32 atomics into dw[I][taps][Cin] slices ----
@@ -1082,8 +1130,12 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   const bool wide = I >= 128 && J >= 128;  // 128x128 tiles for big GEMMs
   // 256x256 tiles on 8 waves for the really big shapes: halves the per-
   // operand HBM re-reads (the 128x128 traffic roofline measured ~450-510 TF
-  // on the ViT wgrad shapes — the same band rocBLAS hits)
-  const bool xwide = I >= 256 && J >= 256 && M >= 4096;
+  // on the ViT wgrad shapes — the same band rocBLAS hits).
+  // PDT_TN_TILE=256 forces the 256 route (A/B measurements); default keeps
+  // the measured-faster 128 tiles.
+  static const char* tn_tile = getenv("PDT_TN_TILE");
+  const bool xwide = (tn_tile && tn_tile[0] == '2') && I >= 256 && J >= 256 &&
+                     M >= 4096;
   if (wide) {
     tiles = ((J + 127) / 128) * ((I + 127) / 128);
     z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
@@ -1100,7 +1152,7 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
     if (xwide) {
       dim3 grid((J + 255) / 256, (I + 255) / 256, (unsigned)(bsz * z));
       hipLaunchKernelGGL(
-          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 8, 8, 2, 4>),
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 8, 8, 2, 4, 3>),
           grid, dim3(512), 0, stream,
           reinterpret_cast<const t16*>(A.data_ptr()),
           reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
@@ -1112,7 +1164,7 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
     if (wide) {
       dim3 grid((J + 127) / 128, (I + 127) / 128, (unsigned)(bsz * z));
       hipLaunchKernelGGL(
-          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4>), grid,
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4, 2, 2, 3>), grid,
           dim3(g16::THREADS), 0, stream,
           reinterpret_cast<const t16*>(A.data_ptr()),
           reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
@@ -1152,7 +1204,9 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
   int n_chunks = (M + 31) / 32;
   g16::ConvMeta cm{};
   auto& zp = zero_page(A.device(), A.scalar_type());
-  const bool xwide = I >= 256 && J >= 256 && M >= 4096;
+  static const char* tn_tile2 = getenv("PDT_TN_TILE");
+  const bool xwide = (tn_tile2 && tn_tile2[0] == '2') && I >= 256 &&
+                     J >= 256 && M >= 4096;
   const bool wide = I >= 128 && J >= 128;
   int tiles = xwide ? ((J + 255) / 256) * ((I + 255) / 256)
               : wide ? ((J + 127) / 128) * ((I + 127) / 128)
@@ -1165,7 +1219,7 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
     if (xwide) {
       dim3 grid((J + 255) / 256, (I + 255) / 256, (unsigned)z);
       hipLaunchKernelGGL(
-          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 8, 8, 2, 4>),
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 8, 8, 2, 4, 3>),
           grid, dim3(512), 0, stream,
           reinterpret_cast<const t16*>(A.data_ptr()),
           reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
@@ -1176,7 +1230,7 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
     if (wide) {
       dim3 grid((J + 127) / 128, (I + 127) / 128, (unsigned)z);
       hipLaunchKernelGGL(
-          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4>), grid,
+          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4, 2, 2, 3>), grid,
           dim3(g16::THREADS), 0, stream,
           reinterpret_cast<const t16*>(A.data_ptr()),
           reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
