@@ -565,7 +565,7 @@ __global__ __launch_bounds__(THREADS) void gemm_tn_bf16_kernel(
 // 32 m-rows.  2 keeps the original drain-at-barrier behavior (fine for the
 // conv variants where 2 co-resident WGs hide each other's stalls).
 template <typename T16, int TAPS, int MODE = MODE_CONV, int TI = 2,
-          int TJ = 2, int WI = 2, int WJ = 2, int NBUF = 2>
+          int TJ = 2, int WI = 2, int WJ = 2, int NBUF = 2, int FUSE = 0>
 __global__ __launch_bounds__(WI * WJ * 64, WI * WJ > 4 ? 1 : 2)
 void gemm_wgrad_tr_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
@@ -585,11 +585,33 @@ void gemm_wgrad_tr_kernel(
   __shared__ __attribute__((aligned(16)))
       T16 lds[NBUF * (TILE_A + TAPS * TILE_B)];
 
+  // XCD-contiguous remap: the 8 XCDs have PRIVATE 4 MB L2s, and the default
+  // round-robin dispatch places the `tiles` blocks that share a zidx (same
+  // m-range → same A/B chunk reads) across all 8 — every shared chunk line
+  // is fetched from HBM once PER XCD.  Remapping logical tile ids so each
+  // XCD owns a contiguous x-fastest/z-slowest run keeps same-chunk readers
+  // on one L2.
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  {
+    const int gx = gridDim.x, gy = gridDim.y;
+    const long long nwg = (long long)gx * gy * gridDim.z;
+    long long id = ((long long)bz * gy + by) * gx + bx;
+    if (nwg >= 16) {
+      const long long q = nwg >> 3, r = nwg & 7;
+      const int xcd = (int)(id & 7);
+      const long long idx = id >> 3;
+      id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    bx = (int)(id % gx);
+    const long long t = id / gx;
+    by = (int)(t % gy);
+    bz = (int)(t / gy);
+  }
   const int Cin = J;
-  const int i0 = blockIdx.y * BI;
-  const int j0 = blockIdx.x * BJ;
-  const int batch = (int)blockIdx.z / zsplit;
-  const int zidx = (int)blockIdx.z % zsplit;
+  const int i0 = by * BI;
+  const int j0 = bx * BJ;
+  const int batch = bz / zsplit;
+  const int zidx = bz % zsplit;
   dy += (long long)batch * sA;
   x += (long long)batch * sB;
   dw += (long long)batch * sC;
@@ -688,26 +710,38 @@ void gemm_wgrad_tr_kernel(
       address_space(3))) const T16*)(p))
 
   constexpr int PER_WAVE_GLDS = (NUNITS + WI * WJ - 1) / (WI * WJ);
-#define WG_WAIT_STAGE()                                        \
-  do {                                                         \
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");         \
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PER_WAVE_GLDS)    \
-                 : "memory");                                  \
-  } while (0)
-#define WG_WAIT_ALL()                                          \
-  do {                                                         \
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");         \
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");           \
-  } while (0)
+  // wait until at most `n` STAGES are still in flight (exact count: waiting
+  // for more than are actually outstanding would let an older, unlanded
+  // stage through)
+  auto wg_wait_stages = [&](int n) {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    switch (n) {
+      case 0:
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        break;
+      case 1:
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(PER_WAVE_GLDS) : "memory");
+        break;
+      case 2:
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * PER_WAVE_GLDS)
+                     : "memory");
+        break;
+      default:
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(3 * PER_WAVE_GLDS)
+                     : "memory");
+        break;
+    }
+  };
 
   stage(0, ch0);
-  if (NBUF == 3) {
-    if (ch0 + 1 < ch1) {
-      stage(1, ch0 + 1);
-      WG_WAIT_STAGE();
-    } else {
-      WG_WAIT_ALL();
+  if (NBUF >= 3) {
+    // fill the prefetch pipeline NBUF-1 deep, then wait for stage 0 only
+    int issued = 1;
+    for (int p = 1; p < NBUF - 1 && ch0 + p < ch1; ++p) {
+      stage(p, ch0 + p);
+      ++issued;
     }
+    wg_wait_stages(issued - 1);
     asm volatile("s_barrier" ::: "memory");
   } else {
     __syncthreads();
@@ -718,12 +752,95 @@ void gemm_wgrad_tr_kernel(
   for (int ch = ch0; ch < ch1; ++ch) {
     const int buf = (ch - ch0) % NBUF;
     const bool more = ch + 1 < ch1;
-    if (NBUF == 3) {
-      if (ch + 2 < ch1) stage((ch - ch0 + 2) % 3, ch + 2);
+    if (NBUF >= 3) {
+      if (ch + NBUF - 1 < ch1)
+        stage((ch - ch0 + NBUF - 1) % NBUF, ch + NBUF - 1);
     } else if (more) {
       stage(buf ^ 1, ch + 1);  // glds latency hides under the MFMAs
     }
     const T16* base = lds + buf * (TILE_A + TAPS * TILE_B);
+
+    if constexpr (FUSE && TAPS == 1 && TI / WI == 2 && TJ / WJ == 2) {
+      // Fused-issue path: all 16 tr reads of the chunk go out in ONE asm
+      // block with a single lgkmcnt wait.  The generic path below pays a
+      // full lgkmcnt(0) drain per 4-read fragment block (4 serial LDS
+      // round-trips per chunk against only 8 MFMAs) — PMC showed the plain
+      // TN kernel wait-bound at ~6x SQ_BUSY with the MFMA pipe 16% busy.
+      const T16* tb = base + TILE_A;
+      const unsigned a0 =
+          LDS_BYTE(base + ((wm >> 4) + img_sel) * IMG) + tr_lane_off;
+      const unsigned a1 =
+          LDS_BYTE(base + ((wm >> 4) + 2 + img_sel) * IMG) + tr_lane_off;
+      const unsigned b0 =
+          LDS_BYTE(tb + ((wn >> 4) + img_sel) * IMG) + tr_lane_off;
+      const unsigned b1 =
+          LDS_BYTE(tb + ((wn >> 4) + 2 + img_sel) * IMG) + tr_lane_off;
+      v4s r[16];
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %16 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %16 offset:128\n\t"
+          "ds_read_b64_tr_b16 %2, %16 offset:512\n\t"
+          "ds_read_b64_tr_b16 %3, %16 offset:640\n\t"
+          "ds_read_b64_tr_b16 %4, %17 offset:0\n\t"
+          "ds_read_b64_tr_b16 %5, %17 offset:128\n\t"
+          "ds_read_b64_tr_b16 %6, %17 offset:512\n\t"
+          "ds_read_b64_tr_b16 %7, %17 offset:640\n\t"
+          "ds_read_b64_tr_b16 %8, %18 offset:0\n\t"
+          "ds_read_b64_tr_b16 %9, %18 offset:128\n\t"
+          "ds_read_b64_tr_b16 %10, %18 offset:512\n\t"
+          "ds_read_b64_tr_b16 %11, %18 offset:640\n\t"
+          "ds_read_b64_tr_b16 %12, %19 offset:0\n\t"
+          "ds_read_b64_tr_b16 %13, %19 offset:128\n\t"
+          "ds_read_b64_tr_b16 %14, %19 offset:512\n\t"
+          "ds_read_b64_tr_b16 %15, %19 offset:640\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(r[0]), "=&v"(r[1]), "=&v"(r[2]), "=&v"(r[3]), "=&v"(r[4]),
+            "=&v"(r[5]), "=&v"(r[6]), "=&v"(r[7]), "=&v"(r[8]), "=&v"(r[9]),
+            "=&v"(r[10]), "=&v"(r[11]), "=&v"(r[12]), "=&v"(r[13]),
+            "=&v"(r[14]), "=&v"(r[15])
+          : "v"(a0), "v"(a1), "v"(b0), "v"(b1));
+      vec16 af2[2][2], bf2[2][2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int kh = 0; kh < 2; ++kh) {
+          reinterpret_cast<v4s*>(&af2[mi][kh])[0] = r[mi * 4 + kh * 2];
+          reinterpret_cast<v4s*>(&af2[mi][kh])[1] = r[mi * 4 + kh * 2 + 1];
+          reinterpret_cast<v4s*>(&bf2[mi][kh])[0] = r[8 + mi * 4 + kh * 2];
+          reinterpret_cast<v4s*>(&bf2[mi][kh])[1] = r[8 + mi * 4 + kh * 2 + 1];
+        }
+      if (colsum && bx == 0 && wn == 0) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          float a = 0.f;
+#pragma unroll
+          for (int kh = 0; kh < 2; ++kh) {
+            const vec16 v = af2[mi][kh];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) a += (float)v[j];
+          }
+          bsum[mi] += a;
+        }
+      }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          acc[mi][ni][0] =
+              M16<T16>::mma32(af2[mi][0], bf2[ni][0], acc[mi][ni][0]);
+          acc[mi][ni][0] =
+              M16<T16>::mma32(af2[mi][1], bf2[ni][1], acc[mi][ni][0]);
+        }
+      if (more) {
+        if (NBUF >= 3) {
+          wg_wait_stages(min(ch + NBUF - 1, ch1 - 1) - (ch + 1));
+          asm volatile("s_barrier" ::: "memory");
+        } else {
+          __syncthreads();
+        }
+      }
+      continue;
+    }
 
     // A fragments (dy): per 32-col tile mi, per 16-m half kh
     vec16 af[TI32][2];
@@ -748,7 +865,7 @@ void gemm_wgrad_tr_kernel(
       reinterpret_cast<v4s*>(&af[mi][0])[1] = h0;
       reinterpret_cast<v4s*>(&af[mi][1])[0] = l1;
       reinterpret_cast<v4s*>(&af[mi][1])[1] = h1;
-      if (colsum && blockIdx.x == 0 && wn == 0) {
+      if (colsum && bx == 0 && wn == 0) {
         // bias grad rides along: the staged dy tile is already in
         // registers; out-of-range m/i lanes were zero-page staged.
         // Only one J-tile column of blocks and only wn==0 waves contribute
@@ -792,11 +909,10 @@ void gemm_wgrad_tr_kernel(
       }
     }
     if (more) {
-      if (NBUF == 3) {
-        // only wait for the loads issued one chunk back; the chunk+2
-        // prefetch keeps flying through the barrier
-        if (ch + 2 < ch1) WG_WAIT_STAGE();
-        else WG_WAIT_ALL();
+      if (NBUF >= 3) {
+        // only wait for the stage one chunk back; newer prefetches keep
+        // flying through the barrier
+        wg_wait_stages(min(ch + NBUF - 1, ch1 - 1) - (ch + 1));
         asm volatile("s_barrier" ::: "memory");
       } else {
         __syncthreads();  // drains the glds (vmcnt) + publishes buf^1
@@ -804,14 +920,12 @@ void gemm_wgrad_tr_kernel(
     }
   }
 
-#undef WG_WAIT_STAGE
-#undef WG_WAIT_ALL
 #undef LDS_BYTE
 
   // ---- writeback: f32 atomics into dw[I][taps][Cin] slices ----
   // (static tap order: a runtime-rotated index into acc would demote the
   // accumulators to scratch)
-  if (colsum && blockIdx.x == 0 && wn == 0) {
+  if (colsum && bx == 0 && wn == 0) {
 #pragma unroll
     for (int mi = 0; mi < TI32; ++mi) {
       float v = bsum[mi] + __shfl_xor(bsum[mi], 32);  // join the ks pair
@@ -842,6 +956,201 @@ void gemm_wgrad_tr_kernel(
   }
 }
 
+
+// ------------------------------------------ TN plain, 128-B-row images ----
+// Plain TN GEMM (C[I,J] += sum_m A[m,I] B[m,J]) with [32 m][64 ch] LDS
+// images (128-B rows).  The gemm_wgrad_tr_kernel above stages [32 m][16 ch]
+// images — 32 B per m-row per glds — so every staged line is a 32-B sliver
+// of a 64-B(+) memory line: TCC counters on the ViT wgrad shapes showed
+// 19M 64-B read requests per dispatch (4x the useful bytes) and the kernel
+// issue/latency-bound at ~425 TF.  128-B rows cut the request count 4x.
+// Everything else (tr-read fragments, f32 atomic writeback, zsplit, XCD
+// remap, optional fused bias grad) matches the kernel above.
+template <typename T16, int TI = 4, int TJ = 4, int WI = 2, int WJ = 2>
+__global__ __launch_bounds__(WI * WJ * 64, 2) void gemm_tn_plain_kernel(
+    const T16* __restrict__ dy, const T16* __restrict__ x,
+    float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot, int I,
+    int J, long long ldc, long long sA, long long sB, long long sC,
+    int zsplit, float* __restrict__ colsum = nullptr) {
+  constexpr int BI = 32 * TI, BJ = 32 * TJ, BMC = 32;
+  constexpr int IMG = 32 * 64;  // elements per [32 m][64 ch] image
+  constexpr int AIMGS = BI / 64, BIMGS = BJ / 64;
+  static_assert(BI % 64 == 0 && BJ % 64 == 0, "128-B-row images");
+  constexpr int TILE_A = AIMGS * IMG, TILE_B = BIMGS * IMG;
+  constexpr int NUNITS = (AIMGS + BIMGS) * 4;  // 8-row glds units per chunk
+  constexpr int NW = WI * WJ;
+  typedef short v4s __attribute__((ext_vector_type(4)));
+  using vec16 = typename M16<T16>::vec;
+  __shared__ __attribute__((aligned(16))) T16 lds[2 * (TILE_A + TILE_B)];
+
+  // XCD-contiguous remap (private per-XCD L2s; see gemm_wgrad_tr_kernel)
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  {
+    const int gx = gridDim.x, gy = gridDim.y;
+    const long long nwg = (long long)gx * gy * gridDim.z;
+    long long id = ((long long)bz * gy + by) * gx + bx;
+    if (nwg >= 16) {
+      const long long q = nwg >> 3, r = nwg & 7;
+      const int xcd = (int)(id & 7);
+      const long long idx = id >> 3;
+      id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    bx = (int)(id % gx);
+    const long long t = id / gx;
+    by = (int)(t % gy);
+    bz = (int)(t / gy);
+  }
+  const int i0 = by * BI;
+  const int j0 = bx * BJ;
+  const int batch = bz / zsplit;
+  const int zidx = bz % zsplit;
+  dy += (long long)batch * sA;
+  x += (long long)batch * sB;
+  dw += (long long)batch * sC;
+  const int n_chunks = (Mtot + BMC - 1) / BMC;
+  const int per_z = (n_chunks + zsplit - 1) / zsplit;
+  const int ch0 = zidx * per_z;
+  const int ch1 = min(n_chunks, ch0 + per_z);
+  if (ch0 >= ch1) return;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = (wave / WJ) * (BI / WI), wn = (wave % WJ) * (BJ / WJ);
+  const int srow = lane >> 3;        // staged m row within the 8-row unit
+  const int sch8 = (lane & 7) * 8;   // 8-ch group within the 128-B row
+
+  f32x16 acc[TI / WI][TJ / WJ] = {};
+
+  // stage chunk ch into buffer buf: every unit reads 8 rows x 128 B
+  // CONTIGUOUS from the operand (full memory lines)
+  auto stage = [&](int buf, int ch) {
+    const int gm0 = ch * BMC;
+    T16* base = lds + buf * (TILE_A + TILE_B);
+    for (int u = wave; u < NUNITS; u += NW) {
+      const int img = u >> 2;           // image index (A then B)
+      const int r0 = (u & 3) * 8;       // 8-row block
+      const int gm = gm0 + r0 + srow;
+      const T16* src = zpad;
+      T16* dst;
+      if (img < AIMGS) {
+        const int ii = i0 + img * 64 + sch8;
+        if (gm < Mtot && ii < I) src = dy + (long long)gm * I + ii;
+        dst = base + img * IMG + r0 * 64;
+      } else {
+        const int jj = j0 + (img - AIMGS) * 64 + sch8;
+        if (gm < Mtot && jj < J) src = x + (long long)gm * J + jj;
+        dst = base + TILE_A + (img - AIMGS) * IMG + r0 * 64;
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
+    }
+  };
+
+  // tr-read fragment addressing for 128-B rows: per-lane row term * 128 B;
+  // reads at +4 rows (512 B) and +16 rows (2048 B) walk the 32-m chunk
+  const int ks = lane >> 5;
+  const unsigned tr_lane_off =
+      (unsigned)((ks * 8 + ((lane & 15) >> 2)) * 128 + (lane & 3) * 8);
+  const int img_sel = (lane >> 4) & 1;
+#define LDS_BYTE(p)                                           \
+  ((unsigned)(unsigned long long)(__attribute__((            \
+      address_space(3))) const T16*)(p))
+
+  stage(0, ch0);
+  __syncthreads();
+
+  constexpr int TI32 = TI / WI, TJ32 = TJ / WJ;
+  float bsum[TI32] = {};
+  for (int ch = ch0; ch < ch1; ++ch) {
+    const int buf = (ch - ch0) & 1;
+    const bool more = ch + 1 < ch1;
+    if (more) stage(buf ^ 1, ch + 1);
+    const T16* base = lds + buf * (TILE_A + TILE_B);
+
+    vec16 af[TI32][2];
+#pragma unroll
+    for (int mi = 0; mi < TI32; ++mi) {
+      const int c = wm + mi * 32;
+      const unsigned a0 =
+          LDS_BYTE(base + (c >> 6) * IMG + (c & 63) + img_sel * 16) +
+          tr_lane_off;
+      v4s l0, h0, l1, h1;
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %4 offset:512\n\t"
+          "ds_read_b64_tr_b16 %2, %4 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %3, %4 offset:2560\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
+          : "v"(a0));
+      reinterpret_cast<v4s*>(&af[mi][0])[0] = l0;
+      reinterpret_cast<v4s*>(&af[mi][0])[1] = h0;
+      reinterpret_cast<v4s*>(&af[mi][1])[0] = l1;
+      reinterpret_cast<v4s*>(&af[mi][1])[1] = h1;
+      if (colsum && bx == 0 && wn == 0) {
+        float a = 0.f;
+#pragma unroll
+        for (int kh = 0; kh < 2; ++kh) {
+          const vec16 v = af[mi][kh];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) a += (float)v[j];
+        }
+        bsum[mi] += a;
+      }
+    }
+#pragma unroll
+    for (int ni = 0; ni < TJ32; ++ni) {
+      const int c = wn + ni * 32;
+      const unsigned b0 =
+          LDS_BYTE(base + TILE_A + (c >> 6) * IMG + (c & 63) + img_sel * 16) +
+          tr_lane_off;
+      v4s l0, h0, l1, h1;
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %4 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %4 offset:512\n\t"
+          "ds_read_b64_tr_b16 %2, %4 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %3, %4 offset:2560\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(l0), "=&v"(h0), "=&v"(l1), "=&v"(h1)
+          : "v"(b0));
+      vec16 bf0, bf1;
+      reinterpret_cast<v4s*>(&bf0)[0] = l0;
+      reinterpret_cast<v4s*>(&bf0)[1] = h0;
+      reinterpret_cast<v4s*>(&bf1)[0] = l1;
+      reinterpret_cast<v4s*>(&bf1)[1] = h1;
+#pragma unroll
+      for (int mi = 0; mi < TI32; ++mi) {
+        acc[mi][ni] = M16<T16>::mma32(af[mi][0], bf0, acc[mi][ni]);
+        acc[mi][ni] = M16<T16>::mma32(af[mi][1], bf1, acc[mi][ni]);
+      }
+    }
+    if (more) __syncthreads();
+  }
+#undef LDS_BYTE
+
+  if (colsum && bx == 0 && wn == 0) {
+#pragma unroll
+    for (int mi = 0; mi < TI32; ++mi) {
+      float v = bsum[mi] + __shfl_xor(bsum[mi], 32);
+      const int col = i0 + wm + mi * 32 + (lane & 31);
+      if (ks == 0 && col < I) atomicAdd(&colsum[col], v);
+    }
+  }
+#pragma unroll
+  for (int ni = 0; ni < TJ32; ++ni) {
+    const int col = j0 + wn + ni * 32 + (lane & 31);
+    if (col >= J) continue;
+#pragma unroll
+    for (int mi = 0; mi < TI32; ++mi)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row = i0 + wm + mi * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * ks;
+        if (row >= I) continue;
+        atomicAdd(&dw[(long long)row * ldc + col], acc[mi][ni][reg]);
+      }
+  }
+}
 
 }  // namespace g16
 
@@ -1138,8 +1447,15 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
                      M >= 4096;
   if (wide) {
     tiles = ((J + 127) / 128) * ((I + 127) / 128);
-    z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
+    // measured z sweep (ViT wgrad shapes, 1xMI355X): ~1024 blocks (= 2 full
+    // occupancy rounds) beats the old ~512 target by 18-27% — more
+    // independent m-sweeps per CU hide the staging latency; beyond ~3x the
+    // atomic fan-in starts to cost.
+    z = std::max(1,
+                 std::min(n_chunks, (1024 + tiles - 1) / std::max(1, tiles)));
     if (bsz > 1) z = std::max(1, std::min(n_chunks, (int)(2048 / bsz) + 1));
+    static const char* e_z = getenv("PDT_TN_Z");  // split-M sweep knob
+    if (e_z && bsz == 1) z = std::max(1, std::min(n_chunks, atoi(e_z)));
   }
   if (xwide) {
     // one 512-thread WG per CU: target ~256 resident blocks (one full round)
@@ -1163,14 +1479,27 @@ torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B) {
     }
     if (wide) {
       dim3 grid((J + 127) / 128, (I + 127) / 128, (unsigned)(bsz * z));
+      // PDT_TN_KERN=old falls back to the [32m][16ch]-image kernel (A/B)
+      static const char* e_kern = getenv("PDT_TN_KERN");
+      if (e_kern && e_kern[0] == 'o') {
+        hipLaunchKernelGGL(
+            (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4>), grid,
+            dim3(g16::THREADS), 0, stream,
+            reinterpret_cast<const t16*>(A.data_ptr()),
+            reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
+            reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J,
+            /*ldc=*/J, cm, (long long)M * I, (long long)M * J,
+            (long long)I * J, z);
+        return;
+      }
       hipLaunchKernelGGL(
-          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4, 2, 2, 3>), grid,
+          (g16::gemm_tn_plain_kernel<t16, 4, 4, 2, 2>), grid,
           dim3(g16::THREADS), 0, stream,
           reinterpret_cast<const t16*>(A.data_ptr()),
           reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
           reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J,
-          /*ldc=*/J, cm, (long long)M * I, (long long)M * J,
-          (long long)I * J, z);
+          /*ldc=*/J, (long long)M * I, (long long)M * J, (long long)I * J, z,
+          nullptr);
       return;
     }
     dim3 grid((J + 63) / 64, (I + 63) / 64, (unsigned)(bsz * z));
@@ -1213,7 +1542,7 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
                      : ((J + 63) / 64) * ((I + 63) / 64);
   int z = std::max(
       1, std::min(n_chunks,
-                  ((xwide ? 256 : 512) + tiles - 1) / std::max(1, tiles)));
+                  ((xwide ? 256 : 1024) + tiles - 1) / std::max(1, tiles)));
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
     if (xwide) {
@@ -1230,11 +1559,11 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
     if (wide) {
       dim3 grid((J + 127) / 128, (I + 127) / 128, (unsigned)z);
       hipLaunchKernelGGL(
-          (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_PLAIN, 4, 4, 2, 2, 3>), grid,
+          (g16::gemm_tn_plain_kernel<t16, 4, 4, 2, 2>), grid,
           dim3(g16::THREADS), 0, stream,
           reinterpret_cast<const t16*>(A.data_ptr()),
           reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
-          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J, J, cm, 0, 0,
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J, J, 0, 0,
           0, z, db.data_ptr<float>());
     } else {
       dim3 grid((J + 63) / 64, (I + 63) / 64, (unsigned)z);
