@@ -240,15 +240,52 @@ def load_training_state(args, path):
 
 class LossScaler:
     """fp16 loss scaling: static (``--loss_scale N``) or dynamic (N == 0,
-    the reference's apex-style default — ddp.py:174-180 intent, natively)."""
+    the reference's apex-style default — ddp.py:174-180 intent, natively).
 
-    def __init__(self, static_scale: float = 0.0):
+    On GPU the overflow check is fully device-side: the fused SGD kernel
+    skips the update when the (pre-unscale) grad norm is non-finite and
+    ticks a device counter; the host reads that counter once per
+    ``sync_interval`` optimizer steps to adjust the scale — NOT once per
+    accumulation step (the round-1 per-step ``isfinite().all()`` host sync
+    was the exact stall pattern the bf16 path avoids)."""
+
+    def __init__(self, static_scale: float = 0.0, sync_interval: int = 50):
         self.dynamic = static_scale == 0
         self.scale = static_scale if static_scale > 0 else 65536.0
         self.growth_interval = 2000
         self._good_steps = 0
+        self.sync_interval = sync_interval
+        self._since_sync = 0
+        self._skip_dev = None
+
+    def skip_counter(self, device):
+        if self._skip_dev is None or self._skip_dev.device != torch.device(device):
+            self._skip_dev = torch.zeros(
+                (), dtype=torch.float32, device=device
+            )
+        return self._skip_dev
+
+    def tick(self):
+        """Called once per optimizer step (GPU path); syncs the device skip
+        counter only every sync_interval steps (one readback)."""
+        self._since_sync += 1
+        self._good_steps += 1
+        if self._since_sync < self.sync_interval:
+            return
+        self._since_sync = 0
+        skips = float(self._skip_dev) if self._skip_dev is not None else 0.0
+        if skips > 0:
+            if self.dynamic:
+                self.scale = max(1.0, self.scale / (2.0 ** min(skips, 16.0)))
+            self._good_steps = 0
+            if self._skip_dev is not None:
+                self._skip_dev.zero_()
+        elif self.dynamic and self._good_steps >= self.growth_interval:
+            self.scale *= 2.0
+            self._good_steps = 0
 
     def step_ok(self, total_norm) -> bool:
+        """Synchronous variant (CPU path / tests): one host sync per call."""
         finite = bool(torch.isfinite(total_norm).all())
         if not self.dynamic:
             return finite
@@ -545,11 +582,23 @@ def train(args, model):
                 total_norm = clip_grad_norm_(
                     [p for p in model.parameters()], args.max_grad_norm
                 )
-                if scaler is not None and not scaler.step_ok(total_norm):
-                    # inf/nan grads: skip the step, shrink the scale
-                    model.zero_grad()
-                    continue
-                optimizer.step()
+                if scaler is not None and args.device.type == "cuda":
+                    # async overflow handling: the fused SGD kernel skips the
+                    # update ON DEVICE when total_norm is non-finite; the
+                    # host syncs the skip counter once per sync_interval
+                    optimizer.step(
+                        guard=total_norm,
+                        skip_count=scaler.skip_counter(args.device),
+                    )
+                    scaler.tick()
+                elif scaler is not None:
+                    if not scaler.step_ok(total_norm):
+                        # inf/nan grads: skip the step, shrink the scale
+                        model.zero_grad()
+                        continue
+                    optimizer.step()
+                else:
+                    optimizer.step()
                 scheduler.step()
                 model.zero_grad()
                 global_step += 1

@@ -1780,9 +1780,13 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   g16::ConvMeta cm{H, W, cl, (int)S, (int)R, (int)stride, (int)pad, HO, WO};
   int n_chunks = (M + 31) / 32;
   int tiles = ((Cin + 63) / 64) * ((Kout + 63) / 64);
-  // ~512 total blocks: fills 256 CUs 2 deep (the tr kernel runs 2 WGs/CU at
-  // its LDS size) while keeping the f32 atomic fan-in small.
-  int z = std::max(1, std::min(n_chunks, (512 + tiles - 1) / std::max(1, tiles)));
+  // block-count target: more split-M blocks hide the staging latency (the
+  // plain-TN z sweep measured +18-27% at ~1024 vs ~512); PDT_CONV_WGRAD_B
+  // overrides for sweeps.
+  static const char* e_cwb = getenv("PDT_CONV_WGRAD_B");
+  const int btarget = e_cwb ? atoi(e_cwb) : 512;
+  int z = std::max(1,
+                   std::min(n_chunks, (btarget + tiles - 1) / std::max(1, tiles)));
   const long long ldc = (long long)R * S * Cin;
   dim3 grid((Cin + 63) / 64, (Kout + 63) / 64, z);
   auto pow2l = [](int v) {
@@ -1816,7 +1820,8 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
       // dy grid.x times; halving grid.x halves that traffic
       int tiles2 = ((Cin + 127) / 128) * ((Kout + 127) / 128);
       int z2 = std::max(
-          1, std::min(n_chunks, (512 + tiles2 - 1) / std::max(1, tiles2)));
+          1,
+          std::min(n_chunks, (btarget + tiles2 - 1) / std::max(1, tiles2)));
       dim3 grid2((Cin + 127) / 128, (Kout + 127) / 128, (unsigned)z2);
       hipLaunchKernelGGL(
           (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_CONV, 4, 4>), grid2,
@@ -1840,7 +1845,8 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
       const int Jl = (int)ldc;  // = R*S*Cin
       int tiles2 = ((Jl + 127) / 128) * ((Kout + 63) / 64);
       int z2 = std::max(
-          1, std::min(n_chunks, (512 + tiles2 - 1) / std::max(1, tiles2)));
+          1,
+          std::min(n_chunks, (btarget + tiles2 - 1) / std::max(1, tiles2)));
       dim3 grid2((Jl + 127) / 128, (Kout + 63) / 64, (unsigned)z2);
       hipLaunchKernelGGL(
           (g16::gemm_wgrad_tr_kernel<t16, 1, g16::MODE_CONVJ, 2, 4>), grid2,

@@ -51,7 +51,18 @@ template <typename T>
 __global__ void sgd_kernel(const ChunkDesc* __restrict__ chunks,
                            const PtrTable* __restrict__ tab, float lr,
                            float momentum, float wd, float damp, int nesterov,
-                           int use_mom) {
+                           int use_mom,
+                           const float* __restrict__ guard = nullptr,
+                           float* __restrict__ skip_count = nullptr) {
+  // Device-side skip: when the (pre-clip) grad norm is non-finite the whole
+  // update is a no-op and a device counter ticks — the fp16 loss-scaler
+  // reads that counter ONCE per sync interval instead of forcing a host
+  // sync on every accumulation step.
+  if (guard != nullptr && !isfinite(*guard)) {
+    if (blockIdx.x == 0 && threadIdx.x == 0 && skip_count)
+      atomicAdd(skip_count, 1.f);
+    return;
+  }
   const ChunkDesc d = chunks[blockIdx.x];
   const long long n = tab->numel[d.tensor];
   const long long base = d.off;
@@ -208,9 +219,21 @@ std::vector<LaunchPlan> make_plans(const std::vector<torch::Tensor>& ref,
 void sgd_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
               std::vector<torch::Tensor> moms, std::vector<torch::Tensor> masters,
               double lr, double momentum, double wd, double damp,
-              bool nesterov) {
+              bool nesterov, c10::optional<torch::Tensor> guard,
+              c10::optional<torch::Tensor> skip_count) {
   TORCH_CHECK(!params.empty());
   auto dev = params[0].device();
+  const float* guard_p = nullptr;
+  float* skip_p = nullptr;
+  if (guard.has_value()) {
+    TORCH_CHECK(guard->scalar_type() == torch::kFloat32 && guard->numel() == 1);
+    guard_p = guard->data_ptr<float>();
+  }
+  if (skip_count.has_value()) {
+    TORCH_CHECK(skip_count->scalar_type() == torch::kFloat32 &&
+                skip_count->numel() == 1);
+    skip_p = skip_count->data_ptr<float>();
+  }
   const bool use_mom = momentum != 0.0;
   auto plans = make_plans(
       params, dev, [&](size_t i, int local, mt::PtrTable* tab) {
@@ -234,7 +257,7 @@ void sgd_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grad
           reinterpret_cast<const mt::ChunkDesc*>(plan.chunks_dev.data_ptr()),
           reinterpret_cast<const mt::PtrTable*>(plan.table_dev.data_ptr()),
           (float)lr, (float)momentum, (float)wd, (float)damp, nesterov ? 1 : 0,
-          use_mom ? 1 : 0);
+          use_mom ? 1 : 0, guard_p, skip_p);
     });
   }
 }

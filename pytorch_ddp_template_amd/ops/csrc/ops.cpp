@@ -108,7 +108,9 @@ void sgd_step(std::vector<torch::Tensor> params,
               std::vector<torch::Tensor> grads,
               std::vector<torch::Tensor> moms,
               std::vector<torch::Tensor> masters, double lr, double momentum,
-              double wd, double damp, bool nesterov);
+              double wd, double damp, bool nesterov,
+              c10::optional<torch::Tensor> guard,
+              c10::optional<torch::Tensor> skip_count);
 torch::Tensor l2norm_sq(std::vector<torch::Tensor> grads);
 void scale_(std::vector<torch::Tensor> ts, double s);
 void scale_by_tensor_(std::vector<torch::Tensor> ts, torch::Tensor s);
@@ -325,7 +327,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd);
   m.def("mse_fwd", &mse_fwd);
   m.def("mse_bwd", &mse_bwd);
-  m.def("sgd_step", &sgd_step);
+  m.def("sgd_step", &sgd_step, py::arg("params"), py::arg("grads"),
+        py::arg("moms"), py::arg("masters"), py::arg("lr"),
+        py::arg("momentum"), py::arg("wd"), py::arg("damp"),
+        py::arg("nesterov"), py::arg("guard") = py::none(),
+        py::arg("skip_count") = py::none());
   m.def("l2norm_sq", &l2norm_sq);
   m.def("scale_", &scale_);
   m.def("scale_by_tensor_", &scale_by_tensor_);

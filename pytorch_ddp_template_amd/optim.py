@@ -83,11 +83,15 @@ class SGD(torch.optim.Optimizer):
         super().__init__(params, defaults)
 
     @torch.no_grad()
-    def step(self, closure=None):
+    def step(self, closure=None, guard=None, skip_count=None):
+        """``guard`` (f32 device scalar, e.g. the pre-clip grad norm): when
+        non-finite the fused kernel skips the whole update ON DEVICE and
+        ticks ``skip_count`` — no host sync (the fp16 loss-scaler path)."""
         loss = None
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
+        self._guard, self._skip_count = guard, skip_count
         for group in self.param_groups:
             params, grads, moms, masters = [], [], [], []
             momentum = group["momentum"]
@@ -109,6 +113,7 @@ class SGD(torch.optim.Optimizer):
             if not params:
                 continue
             self._fused_step(group, params, grads, moms, masters)
+        self._guard = self._skip_count = None
         return loss
 
     def _fused_step(self, group, params, grads, moms, masters):
@@ -128,9 +133,17 @@ class SGD(torch.optim.Optimizer):
                 wd,
                 damp,
                 nesterov,
+                getattr(self, "_guard", None),
+                getattr(self, "_skip_count", None),
             )
             return
         # CPU reference path (also the numerics reference for the kernel)
+        guard = getattr(self, "_guard", None)
+        if guard is not None and not bool(torch.isfinite(guard).all()):
+            sc = getattr(self, "_skip_count", None)
+            if sc is not None:
+                sc.add_(1)
+            return
         for p, g, m, mw in zip(params, grads, moms, masters):
             work = mw if mw is not None else p
             gf = g.float()
