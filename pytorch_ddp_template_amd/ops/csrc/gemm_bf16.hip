@@ -203,22 +203,24 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)ldsA, 16, 0, 0);
-      // ---- B tile (always plain rows of [N,K]; only BNT rows) ----
-      // glds needs a wave-uniform LDS base (writes are lane-linear), so the
-      // guard is per 16-row chunk: chunk (wave*2+i) exists iff < BNT/16.
-      if ((wave * 2 + i) * 16 < BNT) {
-        const int rlb = rl_a[i];
-        const int gn = n0 + rlb;
-        const int swzb = kswz(rlb, kp);
-        const int gkb = k_base + swzb * 8;
-        const T16* srcb =
-            (gn < N && gkb < K) ? B + (long long)gn * K + gkb : zpad;
-        char* ldsB =
-            &smem[buf * BUF_BYTES + TILE_BYTES + (wave * 2 + i) * 1024];
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)srcb,
-            (__attribute__((address_space(3))) unsigned int*)ldsB, 16, 0, 0);
-      }
+    }
+    // ---- B tile (always plain rows of [N,K]; BNT rows, BNT/64 units per
+    // wave so BNT 64/128/256 all cover exactly — 256 is the wide-N route
+    // that halves the A-operand re-reads on big plain GEMMs) ----
+    constexpr int NBU = BNT >= 64 ? BNT / 64 : 1;
+#pragma unroll
+    for (int i = 0; i < NBU; ++i) {
+      const int rlb = (wave * NBU + i) * 16 + (lane >> 2);
+      const int gn = n0 + rlb;
+      const int swzb = kswz(rlb, kp);
+      const int gkb = k_base + swzb * 8;
+      const T16* srcb =
+          (gn < N && gkb < K) ? B + (long long)gn * K + gkb : zpad;
+      char* ldsB =
+          &smem[buf * BUF_BYTES + TILE_BYTES + (wave * NBU + i) * 1024];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcb,
+          (__attribute__((address_space(3))) unsigned int*)ldsB, 16, 0, 0);
     }
   };
 
@@ -229,15 +231,14 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   const int r32 = lane & 31;  // row (A) / col (B, D) within a 32-tile
   const int ks = lane >> 5;   // k half-slot (8 halfwords)
 
-  // per-wave glds count of one stage() call (the vmcnt budget at barriers)
-  const bool light = (BNT == 64) && (wave >= 2);  // no B chunks to stage
+  // per-wave glds count of one stage() call (the vmcnt budget at barriers):
+  // 2 A units + BNT/64 B units, uniform across waves
+  constexpr int STAGE_GLDS = 2 + (BNT >= 64 ? BNT / 64 : 1);
 #define NT_WAIT_STAGE()                                        \
   do {                                                         \
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");         \
-    if (light)                                                 \
-      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");         \
-    else                                                       \
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");         \
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(STAGE_GLDS)       \
+                 : "memory");                                  \
   } while (0)
 #define NT_WAIT_ALL()                                          \
   do {                                                         \
@@ -1190,6 +1191,18 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
   // zero-page columns (ResNet layer1 Kout=64 fwd, C=64 dgrads).
   const bool narrow = N <= 64;
   if (narrow) grid.x = (N + 63) / 64;
+  // wide-N variant: BNT=256 halves the A-operand HBM re-reads on big plain
+  // GEMMs (TCC: 706 MB read per mlp1 fwd dispatch vs 82 ideal at BNT=128)
+  // and doubles the MFMAs per barrier window — but measured NEUTRAL on the
+  // ViT shapes (510 vs 525 TF on mlp1): the 73.7 KB x3-buffer LDS drops
+  // workgroup co-residency and cancels the traffic win, the same pattern
+  // as every occupancy-reducing variant on the TN side.  OFF by default;
+  // PDT_NT_BN=256 enables for A/B runs.
+  static const char* e_ntbn = getenv("PDT_NT_BN");
+  const bool wide_n = (e_ntbn && e_ntbn[0] == '2') &&
+                      MODE == g16::MODE_PLAIN && !narrow && N >= 256 &&
+                      M >= 4096;
+  if (wide_n) grid.x = (N + 255) / 256;
   const bool extras = ex.stats_ws != nullptr || ex.addend != nullptr;
   if (extras) {
     TORCH_CHECK(!relu && !bias_p,
@@ -1230,6 +1243,17 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
                          K, sA, sB, sC, cm, ex.stats_ws, ex.ws_nblocks,       \
                          reinterpret_cast<const t16*>(ex.addend),             \
                          reinterpret_cast<const t16*>(ex.addend_mask));       \
+    else if (wide_n)                                                          \
+      hipLaunchKernelGGL(                                                     \
+          (g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB, 256>), grid,         \
+          dim3(g16::THREADS), 0, stream,                                      \
+          reinterpret_cast<const t16*>(A.data_ptr()),                         \
+          reinterpret_cast<const t16*>(B.data_ptr()),                         \
+          reinterpret_cast<t16*>(C.data_ptr()), bias_p,                       \
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, N, K, sA, sB, sC,   \
+          cm, ex.stats_ws, ex.ws_nblocks,                                     \
+          reinterpret_cast<const t16*>(ex.addend),                            \
+          reinterpret_cast<const t16*>(ex.addend_mask));                      \
     else                                                                      \
       hipLaunchKernelGGL((g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB>),     \
                          grid, dim3(g16::THREADS), 0, stream,                 \
