@@ -1199,9 +1199,8 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
   // as every occupancy-reducing variant on the TN side.  OFF by default;
   // PDT_NT_BN=256 enables for A/B runs.
   static const char* e_ntbn = getenv("PDT_NT_BN");
-  const bool wide_n = (e_ntbn && e_ntbn[0] == '2') &&
-                      MODE == g16::MODE_PLAIN && !narrow && N >= 256 &&
-                      M >= 4096;
+  const bool wide_n =
+      (e_ntbn && e_ntbn[0] == '2') && !narrow && N >= 256 && M >= 4096;
   if (wide_n) grid.x = (N + 255) / 256;
   const bool extras = ex.stats_ws != nullptr || ex.addend != nullptr;
   if (extras) {
