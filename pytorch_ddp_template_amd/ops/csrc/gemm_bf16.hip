@@ -1191,16 +1191,18 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
   // zero-page columns (ResNet layer1 Kout=64 fwd, C=64 dgrads).
   const bool narrow = N <= 64;
   if (narrow) grid.x = (N + 63) / 64;
-  // wide-N variant: BNT=256 halves the A-operand HBM re-reads on big plain
-  // GEMMs (TCC: 706 MB read per mlp1 fwd dispatch vs 82 ideal at BNT=128)
-  // and doubles the MFMAs per barrier window — but measured NEUTRAL on the
-  // ViT shapes (510 vs 525 TF on mlp1): the 73.7 KB x3-buffer LDS drops
-  // workgroup co-residency and cancels the traffic win, the same pattern
-  // as every occupancy-reducing variant on the TN side.  OFF by default;
-  // PDT_NT_BN=256 enables for A/B runs.
+  // wide-N variant: BNT=256 halves the A-operand re-reads and doubles the
+  // MFMAs per barrier window.  Measured: NEUTRAL on plain linear shapes
+  // (510 vs 525 TF on ViT mlp1 — the 73.7 KB LDS drops co-residency and
+  // cancels the traffic win) but a clear WIN for the conv modes where the
+  // A side is an im2col gather that is expensive to re-read: ResNet-18
+  // bench 106.2k -> 108.4k, ResNet-50 7621 -> 8006 samples/s.  Default:
+  // on for conv modes with Kout >= 256, off for MODE_PLAIN.
+  // PDT_NT_BN=256 / 128 forces on / off for A/B runs.
   static const char* e_ntbn = getenv("PDT_NT_BN");
   const bool wide_n =
-      (e_ntbn && e_ntbn[0] == '2') && !narrow && N >= 256 && M >= 4096;
+      (e_ntbn ? e_ntbn[0] == '2' : MODE != g16::MODE_PLAIN) && !narrow &&
+      N >= 256 && M >= 4096;
   if (wide_n) grid.x = (N + 255) / 256;
   const bool extras = ex.stats_ws != nullptr || ex.addend != nullptr;
   if (extras) {
