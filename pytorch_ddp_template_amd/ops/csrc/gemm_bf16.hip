@@ -1808,8 +1808,11 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   // block-count target: more split-M blocks hide the staging latency (the
   // plain-TN z sweep measured +18-27% at ~1024 vs ~512); PDT_CONV_WGRAD_B
   // overrides for sweeps.
+  // e2e bench sweep at the flagship batch sizes: 1024 beats 512 (r50
+  // 7994 -> 8342 samples/s, r18 neutral) — same more-split-M-latency-hiding
+  // result as the plain-TN z sweep.
   static const char* e_cwb = getenv("PDT_CONV_WGRAD_B");
-  const int btarget = e_cwb ? atoi(e_cwb) : 512;
+  const int btarget = e_cwb ? atoi(e_cwb) : 1024;
   int z = std::max(1,
                    std::min(n_chunks, (btarget + tiles - 1) / std::max(1, tiles)));
   const long long ldc = (long long)R * S * Cin;
