@@ -1634,12 +1634,30 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
   using vec16 = typename M16<T16>::vec;
   __shared__ __attribute__((aligned(16))) T16 lds[2 * (TILE_A + 3 * TILE_X)];
 
+  // XCD-contiguous remap (private per-XCD L2s — see gemm_wgrad_tr_kernel):
+  // blocks sharing a zidx read the same dy/x chunks; keep them on one L2.
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  {
+    const int gx = gridDim.x, gy = gridDim.y;
+    const long long nwg = (long long)gx * gy * gridDim.z;
+    long long id = ((long long)bz * gy + by) * gx + bx;
+    if (nwg >= 16) {
+      const long long q = nwg >> 3, r = nwg & 7;
+      const int xcd = (int)(id & 7);
+      const long long idx = id >> 3;
+      id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    bx = (int)(id % gx);
+    const long long t = id / gx;
+    by = (int)(t % gy);
+    bz = (int)(t / gy);
+  }
   const int Cin = J;
-  const int i0 = blockIdx.y * BI;
-  const int j0 = blockIdx.x * BJ;
+  const int i0 = by * BI;
+  const int j0 = bx * BJ;
   const int n_chunks = (Mtot + BMC - 1) / BMC;
   const int per_z = (n_chunks + zsplit - 1) / zsplit;
-  const int ch0 = (int)blockIdx.z * per_z;
+  const int ch0 = bz * per_z;
   const int ch1 = min(n_chunks, ch0 + per_z);
   if (ch0 >= ch1) return;
 
