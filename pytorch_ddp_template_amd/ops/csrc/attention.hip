@@ -443,9 +443,27 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_kernel(
   float* stD = stR + 64;
 
   using vec16 = typename MM<T16>::vec;
-  const int kt = blockIdx.x;           // k-tile (64 rows at kt*64)
-  const int n = blockIdx.z;
-  const int hh = blockIdx.y;
+  // XCD-contiguous remap (the k-tile blocks of one (n, head) read the same
+  // Q/dO stream — keep them on one XCD's private L2)
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  {
+    const int gx = gridDim.x, gy = gridDim.y;
+    const long long nwg = (long long)gx * gy * gridDim.z;
+    long long id = ((long long)bz * gy + by) * gx + bx;
+    if (nwg >= 16) {
+      const long long q = nwg >> 3, r = nwg & 7;
+      const int xcd = (int)(id & 7);
+      const long long idx = id >> 3;
+      id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    bx = (int)(id % gx);
+    const long long t = id / gx;
+    by = (int)(t % gy);
+    bz = (int)(t / gy);
+  }
+  const int kt = bx;                   // k-tile (64 rows at kt*64)
+  const int n = bz;
+  const int hh = by;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int D = H * kDh;
@@ -670,9 +688,27 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_flash_kernel(
   float* redS = redM + 128;                              // [64 rows][2 ki]
 
   using vec16 = typename MM<T16>::vec;
-  const int q0 = blockIdx.x * kBT;
-  const int n = blockIdx.z;
-  const int hh = blockIdx.y;
+  // XCD-contiguous remap: the q-tile blocks of one (n, head) read the same
+  // K/V; keep them on one XCD's private L2 (x-fastest logical runs).
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  {
+    const int gx = gridDim.x, gy = gridDim.y;
+    const long long nwg = (long long)gx * gy * gridDim.z;
+    long long id = ((long long)bz * gy + by) * gx + bx;
+    if (nwg >= 16) {
+      const long long q = nwg >> 3, r = nwg & 7;
+      const int xcd = (int)(id & 7);
+      const long long idx = id >> 3;
+      id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    bx = (int)(id % gx);
+    const long long t = id / gx;
+    by = (int)(t % gy);
+    bz = (int)(t / gy);
+  }
+  const int q0 = bx * kBT;
+  const int n = bz;
+  const int hh = by;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int D = H * kDh;
