@@ -48,6 +48,11 @@ def parse_args():
     # ResNet-18's grads are ~23 MB bf16; 8 MB buckets give ~3 in-flight
     # all-reduces to overlap with backward (one big bucket would serialize)
     p.add_argument("--bucket-mb", type=int, default=8)
+    # hipGraph-captured step (N=1 only): fwd+loss+bwd+clip+SGD captured once
+    # and replayed — removes the per-kernel launch tail that dominates
+    # small-batch/serving shapes (LR is frozen at capture; inputs are
+    # refreshed by a D2D copy before each replay).
+    p.add_argument("--graph", action="store_true")
     return p.parse_args()
 
 
@@ -95,18 +100,57 @@ def main():
         for _ in range(n_batches)
     ]
 
-    def step(i):
-        x, y = xs[i % n_batches], ys[i % n_batches]
-        out = model(x)
-        loss = crit(out, y)
-        loss.backward()
-        if distributed:
-            model.finish_gradient_sync()
-        clip_grad_norm_(list(model.parameters()), 1000.0)
-        opt.step()
-        sched.step()
-        model.zero_grad()
-        return loss
+    use_graph = args.graph and not distributed
+    if args.graph and distributed:
+        print("--graph is single-GPU only; running eager", file=sys.stderr)
+
+    if use_graph:
+        # hipGraph mode: stable grad/optimizer-state pointers are required
+        # (zero in place, not set_to_none), one static input pair that is
+        # refreshed by a D2D copy before each replay.
+        x_st = xs[0].clone()
+        y_st = ys[0].clone()
+        loss_st = None
+
+        def inner_step():
+            out = model(x_st)
+            loss = crit(out, y_st)
+            loss.backward()
+            clip_grad_norm_(list(model.parameters()), 1000.0)
+            opt.step()
+            model.zero_grad(set_to_none=False)
+            return loss
+
+        # eager warmup on a side stream (torch's capture recipe), then capture
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(max(3, args.warmup)):
+                loss_st = inner_step()
+        torch.cuda.current_stream().wait_stream(side)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            loss_st = inner_step()
+
+        def step(i):
+            x_st.copy_(xs[i % n_batches])
+            y_st.copy_(ys[i % n_batches])
+            graph.replay()
+            sched.step()  # host-side bookkeeping (captured lr stays fixed)
+            return loss_st
+    else:
+        def step(i):
+            x, y = xs[i % n_batches], ys[i % n_batches]
+            out = model(x)
+            loss = crit(out, y)
+            loss.backward()
+            if distributed:
+                model.finish_gradient_sync()
+            clip_grad_norm_(list(model.parameters()), 1000.0)
+            opt.step()
+            sched.step()
+            model.zero_grad()
+            return loss
 
     for i in range(args.warmup):
         step(i)
@@ -163,6 +207,7 @@ def main():
                         "image": img,
                         "num_classes": num_classes,
                         "parallelism": f"dp{n_gpus}",
+                        "hip_graph": bool(use_graph),
                         "optimizer": "sgd+momentum(master fp32)",
                     },
                 }

@@ -95,7 +95,7 @@ enum { MODE_PLAIN = 0, MODE_CONV = 1, MODE_CONVJ = 2 };
 // EXTRAS=false compiles the epilogue extras out entirely (they cost
 // registers/codegen even when the pointers are null).
 template <typename T16, int MODE, bool RELU, bool HAS_BIAS, int BNT = BN,
-          bool EXTRAS = false>
+          bool EXTRAS = false, int NBUF = 3>
 __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
     const T16* __restrict__ A, const T16* __restrict__ B,
     T16* __restrict__ C, const T16* __restrict__ bias,
@@ -109,7 +109,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   // with glds in flight emits vmcnt(0) and drains the NEXT k-step's loads,
   // collapsing the pipeline (guide §6: -16..20%% at GEMM scale).  With 3
   // buffers each barrier only waits for loads issued TWO steps back.
-  __shared__ __attribute__((aligned(16))) char smem[3 * BUF_BYTES];
+  __shared__ __attribute__((aligned(16))) char smem[NBUF * BUF_BYTES];
 
   // ----- block swizzle (bijective XCD remap over the x*y grid) -----
   int nwg = gridDim.x * gridDim.y;
@@ -247,12 +247,20 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
   } while (0)
 
   stage(0, 0);
-  if (KT > 1) stage(1, 1);
-  if (KT > 1) NT_WAIT_STAGE(); else NT_WAIT_ALL();
+  if (NBUF >= 3) {
+    if (KT > 1) stage(1, 1);
+    if (KT > 1) NT_WAIT_STAGE(); else NT_WAIT_ALL();
+  } else {
+    NT_WAIT_ALL();
+  }
   asm volatile("s_barrier" ::: "memory");
   for (int kt = 0; kt < KT; ++kt) {
-    const int buf = kt % 3;
-    if (kt + 2 < KT) stage((kt + 2) % 3, kt + 2);
+    const int buf = kt % NBUF;
+    if (NBUF >= 3) {
+      if (kt + 2 < KT) stage((kt + 2) % NBUF, kt + 2);
+    } else {
+      if (kt + 1 < KT) stage((kt + 1) & 1, kt + 1);
+    }
     const char* baseA = &smem[buf * BUF_BYTES];
     const char* baseB = baseA + TILE_BYTES;
     using vec16 = typename M16<T16>::vec;
@@ -279,7 +287,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_bf16_kernel(
           acc[mi][ni] = M16<T16>::mma32(af[mi], bf[ni], acc[mi][ni]);
     }
     if (kt + 1 < KT) {
-      if (kt + 2 < KT) NT_WAIT_STAGE(); else NT_WAIT_ALL();
+      if (NBUF >= 3 && kt + 2 < KT) NT_WAIT_STAGE(); else NT_WAIT_ALL();
       asm volatile("s_barrier" ::: "memory");
     }
   }
@@ -1204,6 +1212,10 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
       (e_ntbn ? e_ntbn[0] == '2' : MODE != g16::MODE_PLAIN) && !narrow &&
       N >= 256 && M >= 4096;
   if (wide_n) grid.x = (N + 255) / 256;
+  // 2-buffer wide-N: same 49 KB footprint as BNT=128x3buf (unchanged
+  // co-residency) with 2x the MFMAs per barrier window.  PDT_NT_NBUF2=1.
+  static const char* e_nb2 = getenv("PDT_NT_NBUF2");
+  const bool ntbuf2 = e_nb2 && e_nb2[0] == '1';
   const bool extras = ex.stats_ws != nullptr || ex.addend != nullptr;
   if (extras) {
     TORCH_CHECK(!relu && !bias_p,
@@ -1244,6 +1256,17 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
                          K, sA, sB, sC, cm, ex.stats_ws, ex.ws_nblocks,       \
                          reinterpret_cast<const t16*>(ex.addend),             \
                          reinterpret_cast<const t16*>(ex.addend_mask));       \
+    else if (wide_n && ntbuf2)                                                \
+      hipLaunchKernelGGL(                                                     \
+          (g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB, 256, false, 2>),     \
+          grid, dim3(g16::THREADS), 0, stream,                                \
+          reinterpret_cast<const t16*>(A.data_ptr()),                         \
+          reinterpret_cast<const t16*>(B.data_ptr()),                         \
+          reinterpret_cast<t16*>(C.data_ptr()), bias_p,                       \
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, N, K, sA, sB, sC,   \
+          cm, ex.stats_ws, ex.ws_nblocks,                                     \
+          reinterpret_cast<const t16*>(ex.addend),                            \
+          reinterpret_cast<const t16*>(ex.addend_mask));                      \
     else if (wide_n)                                                          \
       hipLaunchKernelGGL(                                                     \
           (g16::gemm_nt_bf16_kernel<t16, MODE, RELU, HB, 256>), grid,         \

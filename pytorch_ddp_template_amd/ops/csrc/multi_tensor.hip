@@ -180,6 +180,35 @@ struct LaunchPlan {
   int n_blocks;
 };
 
+// Plan cache: the descriptor tables depend only on the tensor PTRS/numels,
+// which are stable across steps when grads are zeroed in place (and must be
+// stable under hipGraph capture).  Caching (a) keeps the HOST staging
+// tensors alive so a captured H2D copy replays from live memory, and
+// (b) removes the per-step host build + H2D copy on the hot path.
+struct PlanKey {
+  std::vector<const void*> ptrs;
+  bool operator==(const PlanKey& o) const { return ptrs == o.ptrs; }
+};
+struct PlanEntry {
+  PlanKey key;
+  std::vector<LaunchPlan> plans;
+};
+
+std::vector<PlanEntry>& plan_cache() {
+  static std::vector<PlanEntry> c;
+  return c;
+}
+
+template <typename Build>
+const std::vector<LaunchPlan>& cached_plans(PlanKey&& key, Build build) {
+  auto& cache = plan_cache();
+  for (auto& e : cache)
+    if (e.key == key) return e.plans;
+  if (cache.size() >= 64) cache.clear();  // bound (param sets rarely churn)
+  cache.push_back(PlanEntry{std::move(key), build()});
+  return cache.back().plans;
+}
+
 // Build chunk descriptors + pointer table on host, copy to device once per
 // call (tiny: ~KBs).  Lists longer than kMaxTensors are processed in groups.
 template <typename FillPtrs>
@@ -235,19 +264,36 @@ void sgd_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grad
     skip_p = skip_count->data_ptr<float>();
   }
   const bool use_mom = momentum != 0.0;
-  auto plans = make_plans(
-      params, dev, [&](size_t i, int local, mt::PtrTable* tab) {
-        tab->grad[local] = grads[i].data_ptr();
-        tab->param[local] = params[i].data_ptr();
-        tab->mom[local] =
-            (use_mom && moms[i].defined() && moms[i].numel())
-                ? moms[i].data_ptr<float>()
-                : nullptr;
-        tab->master[local] =
-            (masters[i].defined() && masters[i].numel())
-                ? masters[i].data_ptr<float>()
-                : nullptr;
-      });
+  PlanKey key;
+  key.ptrs.reserve(4 * params.size() + 2);
+  key.ptrs.push_back((const void*)0x56D);  // op tag
+  key.ptrs.push_back((const void*)(uintptr_t)params.size());
+  for (size_t i = 0; i < params.size(); ++i) {
+    key.ptrs.push_back(params[i].data_ptr());
+    key.ptrs.push_back(grads[i].data_ptr());
+    key.ptrs.push_back((const void*)(uintptr_t)params[i].numel());
+    key.ptrs.push_back(
+        (use_mom && moms[i].defined() && moms[i].numel()) ? moms[i].data_ptr()
+                                                          : nullptr);
+    key.ptrs.push_back((masters[i].defined() && masters[i].numel())
+                           ? masters[i].data_ptr()
+                           : nullptr);
+  }
+  const auto& plans = cached_plans(std::move(key), [&] {
+    return make_plans(
+        params, dev, [&](size_t i, int local, mt::PtrTable* tab) {
+          tab->grad[local] = grads[i].data_ptr();
+          tab->param[local] = params[i].data_ptr();
+          tab->mom[local] =
+              (use_mom && moms[i].defined() && moms[i].numel())
+                  ? moms[i].data_ptr<float>()
+                  : nullptr;
+          tab->master[local] =
+              (masters[i].defined() && masters[i].numel())
+                  ? masters[i].data_ptr<float>()
+                  : nullptr;
+        });
+  });
   auto stream = c10::hip::getCurrentHIPStream();
   for (auto& plan : plans) {
     DDP_DISPATCH_FLOAT(params[0].scalar_type(), "sgd_step", [&] {
@@ -266,10 +312,20 @@ torch::Tensor l2norm_sq(std::vector<torch::Tensor> grads) {
   TORCH_CHECK(!grads.empty());
   auto dev = grads[0].device();
   auto out = torch::zeros({}, torch::dtype(torch::kFloat32).device(dev));
-  auto plans = make_plans(grads, dev,
-                          [&](size_t i, int local, mt::PtrTable* tab) {
-                            tab->grad[local] = grads[i].data_ptr();
-                          });
+  PlanKey key;
+  key.ptrs.reserve(2 * grads.size() + 2);
+  key.ptrs.push_back((const void*)0x127);  // op tag
+  key.ptrs.push_back((const void*)(uintptr_t)grads.size());
+  for (auto& g : grads) {
+    key.ptrs.push_back(g.data_ptr());
+    key.ptrs.push_back((const void*)(uintptr_t)g.numel());
+  }
+  const auto& plans = cached_plans(std::move(key), [&] {
+    return make_plans(grads, dev,
+                      [&](size_t i, int local, mt::PtrTable* tab) {
+                        tab->grad[local] = grads[i].data_ptr();
+                      });
+  });
   auto stream = c10::hip::getCurrentHIPStream();
   for (auto& plan : plans) {
     DDP_DISPATCH_FLOAT(grads[0].scalar_type(), "l2norm_sq", [&] {
@@ -287,8 +343,18 @@ torch::Tensor l2norm_sq(std::vector<torch::Tensor> grads) {
 static void scale_impl(std::vector<torch::Tensor>& ts, float s,
                        const torch::Tensor* dev_scale, bool clamp) {
   auto dev = ts[0].device();
-  auto plans = make_plans(ts, dev, [&](size_t i, int local, mt::PtrTable* tab) {
-    tab->param[local] = ts[i].data_ptr();
+  PlanKey key;
+  key.ptrs.reserve(2 * ts.size() + 2);
+  key.ptrs.push_back((const void*)0x5CA1E);  // op tag
+  key.ptrs.push_back((const void*)(uintptr_t)ts.size());
+  for (auto& t : ts) {
+    key.ptrs.push_back(t.data_ptr());
+    key.ptrs.push_back((const void*)(uintptr_t)t.numel());
+  }
+  const auto& plans = cached_plans(std::move(key), [&] {
+    return make_plans(ts, dev, [&](size_t i, int local, mt::PtrTable* tab) {
+      tab->param[local] = ts[i].data_ptr();
+    });
   });
   auto stream = c10::hip::getCurrentHIPStream();
   for (auto& plan : plans) {
