@@ -53,7 +53,11 @@ __global__ void sgd_kernel(const ChunkDesc* __restrict__ chunks,
                            float momentum, float wd, float damp, int nesterov,
                            int use_mom,
                            const float* __restrict__ guard = nullptr,
-                           float* __restrict__ skip_count = nullptr) {
+                           float* __restrict__ skip_count = nullptr,
+                           const float* __restrict__ lr_dev = nullptr) {
+  // lr can come from a device scalar so a hipGraph-captured step follows
+  // the LIVE schedule (the host writes the new lr before each replay)
+  if (lr_dev != nullptr) lr = *lr_dev;
   // Device-side skip: when the (pre-clip) grad norm is non-finite the whole
   // update is a no-op and a device counter ticks — the fp16 loss-scaler
   // reads that counter ONCE per sync interval instead of forcing a host
@@ -249,7 +253,8 @@ void sgd_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grad
               std::vector<torch::Tensor> moms, std::vector<torch::Tensor> masters,
               double lr, double momentum, double wd, double damp,
               bool nesterov, c10::optional<torch::Tensor> guard,
-              c10::optional<torch::Tensor> skip_count) {
+              c10::optional<torch::Tensor> skip_count,
+              c10::optional<torch::Tensor> lr_tensor) {
   TORCH_CHECK(!params.empty());
   auto dev = params[0].device();
   const float* guard_p = nullptr;
@@ -262,6 +267,12 @@ void sgd_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grad
     TORCH_CHECK(skip_count->scalar_type() == torch::kFloat32 &&
                 skip_count->numel() == 1);
     skip_p = skip_count->data_ptr<float>();
+  }
+  const float* lr_p = nullptr;
+  if (lr_tensor.has_value()) {
+    TORCH_CHECK(lr_tensor->scalar_type() == torch::kFloat32 &&
+                lr_tensor->numel() == 1);
+    lr_p = lr_tensor->data_ptr<float>();
   }
   const bool use_mom = momentum != 0.0;
   PlanKey key;
@@ -303,7 +314,7 @@ void sgd_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grad
           reinterpret_cast<const mt::ChunkDesc*>(plan.chunks_dev.data_ptr()),
           reinterpret_cast<const mt::PtrTable*>(plan.table_dev.data_ptr()),
           (float)lr, (float)momentum, (float)wd, (float)damp, nesterov ? 1 : 0,
-          use_mom ? 1 : 0, guard_p, skip_p);
+          use_mom ? 1 : 0, guard_p, skip_p, lr_p);
     });
   }
 }

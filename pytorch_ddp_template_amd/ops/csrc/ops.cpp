@@ -110,7 +110,8 @@ void sgd_step(std::vector<torch::Tensor> params,
               std::vector<torch::Tensor> masters, double lr, double momentum,
               double wd, double damp, bool nesterov,
               c10::optional<torch::Tensor> guard,
-              c10::optional<torch::Tensor> skip_count);
+              c10::optional<torch::Tensor> skip_count,
+              c10::optional<torch::Tensor> lr_tensor);
 torch::Tensor l2norm_sq(std::vector<torch::Tensor> grads);
 void scale_(std::vector<torch::Tensor> ts, double s);
 void scale_by_tensor_(std::vector<torch::Tensor> ts, torch::Tensor s);
@@ -331,7 +332,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("moms"), py::arg("masters"), py::arg("lr"),
         py::arg("momentum"), py::arg("wd"), py::arg("damp"),
         py::arg("nesterov"), py::arg("guard") = py::none(),
-        py::arg("skip_count") = py::none());
+        py::arg("skip_count") = py::none(),
+        py::arg("lr_tensor") = py::none());
   m.def("l2norm_sq", &l2norm_sq);
   m.def("scale_", &scale_);
   m.def("scale_by_tensor_", &scale_by_tensor_);

@@ -83,15 +83,18 @@ class SGD(torch.optim.Optimizer):
         super().__init__(params, defaults)
 
     @torch.no_grad()
-    def step(self, closure=None, guard=None, skip_count=None):
+    def step(self, closure=None, guard=None, skip_count=None, lr_tensor=None):
         """``guard`` (f32 device scalar, e.g. the pre-clip grad norm): when
         non-finite the fused kernel skips the whole update ON DEVICE and
-        ticks ``skip_count`` — no host sync (the fp16 loss-scaler path)."""
+        ticks ``skip_count`` — no host sync (the fp16 loss-scaler path).
+        ``lr_tensor`` (f32 device scalar): the kernel reads lr from the
+        device, so a hipGraph-captured step follows the live schedule."""
         loss = None
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
         self._guard, self._skip_count = guard, skip_count
+        self._lr_tensor = lr_tensor
         for group in self.param_groups:
             params, grads, moms, masters = [], [], [], []
             momentum = group["momentum"]
@@ -113,7 +116,7 @@ class SGD(torch.optim.Optimizer):
             if not params:
                 continue
             self._fused_step(group, params, grads, moms, masters)
-        self._guard = self._skip_count = None
+        self._guard = self._skip_count = self._lr_tensor = None
         return loss
 
     def _fused_step(self, group, params, grads, moms, masters):
@@ -135,9 +138,13 @@ class SGD(torch.optim.Optimizer):
                 nesterov,
                 getattr(self, "_guard", None),
                 getattr(self, "_skip_count", None),
+                getattr(self, "_lr_tensor", None),
             )
             return
         # CPU reference path (also the numerics reference for the kernel)
+        lrt = getattr(self, "_lr_tensor", None)
+        if lrt is not None:
+            lr = float(lrt)
         guard = getattr(self, "_guard", None)
         if guard is not None and not bool(torch.isfinite(guard).all()):
             sc = getattr(self, "_skip_count", None)

@@ -776,3 +776,18 @@ def test_sgd_device_guard_skips_nonfinite():
     torch.cuda.synchronize()
     assert not torch.equal(p, p0), "finite guard must step"
     assert float(skip) == 1.0
+
+
+def test_sgd_lr_from_device_tensor():
+    """sgd_step reads lr from a device scalar when given (the hipGraph path:
+    captured step follows the live schedule via one fill_ per step)."""
+    p = t32(512, seed=97).to(torch.float32).to(DEV)
+    p0 = p.clone()
+    g = torch.ones(512, dtype=torch.float32, device=DEV)
+    e = torch.Tensor()
+    lr_dev = torch.full((), 0.25, dtype=torch.float32, device=DEV)
+    # host lr says 99.0; device scalar must win
+    EXT.sgd_step([p], [g], [e], [e], 99.0, 0.0, 0.0, 0.0, False, None, None,
+                 lr_dev)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(p, p0 - 0.25, rtol=1e-6, atol=1e-6)
