@@ -126,15 +126,22 @@ def main():
             model.zero_grad(set_to_none=False)
             return loss
 
-        # eager warmup on a side stream (torch's capture recipe), then capture
+        # Capture recipe: grads/optimizer state/descriptor plans must all be
+        # pointer-stable BEFORE capture (in-place zeroing, not set_to_none —
+        # a multi-tensor plan rebuild inside capture would pin host memory,
+        # which is capture-illegal), warmup losses are not kept (stale
+        # autograd-graph references corrupt the capture), and the capture
+        # runs ON THE WARMUP STREAM so the AccumulateGrad nodes' recorded
+        # stream matches the capture stream.
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
             for _ in range(max(3, args.warmup)):
-                loss_st = inner_step()
+                inner_step()
         torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        with torch.cuda.graph(graph, stream=side):
             loss_st = inner_step()
 
         def step(i):

@@ -1208,9 +1208,15 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
   // on for conv modes with Kout >= 256, off for MODE_PLAIN.
   // PDT_NT_BN=256 / 128 forces on / off for A/B runs.
   static const char* e_ntbn = getenv("PDT_NT_BN");
+  // NOTE: the epilogue-extras instantiations (stats workspace / residual
+  // addend — the fused conv+BN path) are built at BNT=128 only, so wide-N
+  // must not halve their grid: doing so left half of C unwritten
+  // (uninitialized-memory NaNs from the second step on — found by the
+  // round-2 step-1 NaN bisect, tools/graph_dbg2.py).
+  const bool extras_early = ex.stats_ws != nullptr || ex.addend != nullptr;
   const bool wide_n =
       (e_ntbn ? e_ntbn[0] == '2' : MODE != g16::MODE_PLAIN) && !narrow &&
-      N >= 256 && M >= 4096;
+      !extras_early && N >= 256 && M >= 4096;
   if (wide_n) grid.x = (N + 255) / 256;
   // 2-buffer wide-N: same 49 KB footprint as BNT=128x3buf (unchanged
   // co-residency) with 2x the MFMAs per barrier window.  PDT_NT_NBUF2=1.

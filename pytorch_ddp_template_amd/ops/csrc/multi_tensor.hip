@@ -179,8 +179,10 @@ __global__ void scale_kernel(const ChunkDesc* __restrict__ chunks,
 namespace {
 
 struct LaunchPlan {
-  torch::Tensor chunks_dev;  // ChunkDesc[]
-  torch::Tensor table_dev;   // PtrTable
+  torch::Tensor chunks_dev;   // ChunkDesc[]
+  torch::Tensor table_dev;    // PtrTable
+  torch::Tensor chunks_host;  // PINNED sources kept alive: a hipGraph-
+  torch::Tensor table_host;   // captured H2D copy replays from this memory
   int n_blocks;
 };
 
@@ -222,8 +224,9 @@ std::vector<LaunchPlan> make_plans(const std::vector<torch::Tensor>& ref,
   size_t t = 0;
   while (t < ref.size()) {
     size_t t_end = std::min(ref.size(), t + (size_t)mt::kMaxTensors);
-    auto table_host = torch::empty({(long long)sizeof(mt::PtrTable)},
-                                   torch::dtype(torch::kUInt8));
+    auto table_host = torch::empty(
+        {(long long)sizeof(mt::PtrTable)},
+        torch::dtype(torch::kUInt8).pinned_memory(true));
     auto* tab = reinterpret_cast<mt::PtrTable*>(table_host.data_ptr());
     std::vector<mt::ChunkDesc> chunks;
     for (size_t i = t; i < t_end; ++i) {
@@ -234,12 +237,15 @@ std::vector<LaunchPlan> make_plans(const std::vector<torch::Tensor>& ref,
       for (long long off = 0; off < n; off += mt::kChunk)
         chunks.push_back({local, off});
     }
-    auto chunks_host = torch::from_blob(
-        chunks.data(), {(long long)(chunks.size() * sizeof(mt::ChunkDesc))},
-        torch::dtype(torch::kUInt8));
     LaunchPlan plan;
-    plan.chunks_dev = chunks_host.clone().to(dev, /*non_blocking=*/true);
-    plan.table_dev = table_host.to(dev, /*non_blocking=*/true);
+    plan.chunks_host = torch::from_blob(
+                           chunks.data(),
+                           {(long long)(chunks.size() * sizeof(mt::ChunkDesc))},
+                           torch::dtype(torch::kUInt8))
+                           .pin_memory();
+    plan.table_host = table_host;
+    plan.chunks_dev = plan.chunks_host.to(dev, /*non_blocking=*/true);
+    plan.table_dev = plan.table_host.to(dev, /*non_blocking=*/true);
     plan.n_blocks = (int)chunks.size();
     plans.push_back(plan);
     t = t_end;

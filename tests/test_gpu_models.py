@@ -75,6 +75,37 @@ def test_resnet18_bf16_training_loss_descends():
     assert all(l == l for l in losses), "NaN loss"
 
 
+def test_resnet18_training_grads_finite_under_allocator_churn():
+    """Regression for the round-2 wide-N bug (see test_gpu_ops.py
+    test_conv2d_fwd_stats_wide_n_shape): steady-state training loops recycle
+    the same allocator blocks each step, so a kernel that leaves part of its
+    output unwritten reads stale-but-plausible values and loss still
+    descends.  Alternate batch sizes to churn the allocator and assert every
+    grad/param stays finite — this fails loudly on unwritten-memory bugs."""
+    from pytorch_ddp_template_amd.models import resnet18
+    from pytorch_ddp_template_amd.ops import CrossEntropyLoss
+    from pytorch_ddp_template_amd.optim import SGD, clip_grad_norm_
+
+    torch.manual_seed(7)
+    m = resnet18(num_classes=10, stem="cifar").to(torch.bfloat16).to(DEV)
+    opt = SGD(m.parameters(), lr=0.01, momentum=0.9, master_weights=True)
+    crit = CrossEntropyLoss()
+    for i, bs in enumerate([256, 192, 320, 256, 64, 256]):
+        x = torch.randn(bs, 32, 32, 3).to(torch.bfloat16).to(DEV)
+        y = torch.randint(0, 10, (bs,)).to(DEV)
+        loss = crit(m(x), y)
+        loss.backward()
+        for n, p in m.named_parameters():
+            assert torch.isfinite(p.grad.float()).all(), f"step {i}: {n} grad"
+        clip_grad_norm_(list(m.parameters()), 1000.0)
+        opt.step()
+        m.zero_grad(set_to_none=False)
+        torch.cuda.synchronize()
+        for n, p in m.named_parameters():
+            assert torch.isfinite(p.float()).all(), f"step {i}: {n} param"
+        assert float(loss) < 6.0, f"step {i}: loss {float(loss)}"
+
+
 def test_foo_training_with_mse_descends():
     from pytorch_ddp_template_amd.models import FooModel
     from pytorch_ddp_template_amd.ops import MSELoss

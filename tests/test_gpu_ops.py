@@ -122,7 +122,16 @@ def test_transpose2d():
 # ---------------- conv2d ----------------
 
 
-def conv_ref(x, w, stride, pad):
+def _poison_allocator(numel):
+    """Fill a freed caching-allocator block of ~numel bf16 elements with NaN
+    so a kernel that leaves part of its freshly-allocated output unwritten
+    fails loudly instead of reading stale-but-plausible recycled values.
+    (The round-2 wide-N bug left half of C unwritten: first call ran on
+    driver-zeroed pages and every loop iteration recycled the same block, so
+    plain unit tests and short training loops both passed.)"""
+    junk = torch.full((numel,), float("nan"), dtype=torch.bfloat16, device=DEV)
+    del junk
+    torch.cuda.synchronize()
     return (
         F.conv2d(
             x.float().permute(0, 3, 1, 2),
@@ -144,11 +153,13 @@ def conv_ref(x, w, stride, pad):
         (2, 8, 128, 64, 1, 1, 0),     # 1x1
         (2, 32, 3, 64, 3, 1, 1),      # stem (im2col fallback)
         (1, 16, 64, 64, 1, 2, 0),     # 1x1 stride-2 (downsample)
+        (16, 16, 256, 256, 3, 1, 1),  # wide-N variant (M>=4096, N>=256)
     ],
 )
 def test_conv2d_fwd_bf16(n, h, c, k, r, stride, pad):
     x = t32(n, h, h, c, seed=16).to(torch.bfloat16)
     w = (t32(k, r, r, c, seed=17) * (2.0 / (r * r * c)) ** 0.5).to(torch.bfloat16)
+    _poison_allocator(n * h * h * k)
     out = EXT.conv2d_fwd(x.to(DEV), w.to(DEV), None, stride, pad, False)
     ref = conv_ref(x, w, stride, pad)
     close_bf16(out, ref, scale=ref.abs().max().clamp(min=0.5))
@@ -156,12 +167,14 @@ def test_conv2d_fwd_bf16(n, h, c, k, r, stride, pad):
 
 @pytest.mark.parametrize(
     "n,h,c,k,r,stride,pad",
-    [(2, 16, 64, 64, 3, 1, 1), (2, 16, 64, 128, 3, 2, 1), (1, 8, 64, 64, 1, 2, 0)],
+    [(2, 16, 64, 64, 3, 1, 1), (2, 16, 64, 128, 3, 2, 1), (1, 8, 64, 64, 1, 2, 0),
+     (16, 16, 256, 256, 3, 1, 1)],  # wide-N variant (M>=4096, N>=256)
 )
 def test_conv2d_dgrad_bf16(n, h, c, k, r, stride, pad):
     ho = (h + 2 * pad - r) // stride + 1
     dy = t32(n, ho, ho, k, seed=18).to(torch.bfloat16)
     w = (t32(k, r, r, c, seed=19) * 0.1).to(torch.bfloat16)
+    _poison_allocator(n * h * h * c)
     dx = EXT.conv2d_dgrad(dy.to(DEV), w.to(DEV), stride, pad, h, h)
     ref = torch.nn.grad.conv2d_input(
         (n, c, h, h),
@@ -545,6 +558,32 @@ def test_conv2d_fwd_stats_wrapped_workspace():
     var = yf.var(0, unbiased=False)
     close_f32(mean, mu.cpu(), tol=2e-3)
     close_f32(rstd, (var + 1e-5).rsqrt().cpu(), tol=2e-3)
+
+
+def test_conv2d_fwd_stats_wide_n_shape():
+    """Regression for the round-2 wide-N bug: conv2d_fwd_stats (the
+    epilogue-extras instantiation, built at BNT=128) was launched with a
+    wide-N (BNT=256) grid for Kout>=256 / M>=4096, leaving half of y
+    unwritten — ResNet-18 layer3/4 trained on recycled-allocator garbage
+    from the second step on (NaN weights, ReLU-dead network, loss pinned at
+    ln(10)).  Poison the allocator so unwritten columns fail loudly."""
+    N, H, C, K = 16, 16, 256, 256  # M = 16*16*16 = 4096, Kout = 256
+    x = t32(N, H, H, C, seed=80).to(torch.bfloat16)
+    w = (t32(K, 3, 3, C, seed=81) * (2.0 / (9 * C)) ** 0.5).to(torch.bfloat16)
+    _poison_allocator(N * H * H * K)
+    y, ws = EXT.conv2d_fwd_stats(x.to(DEV), w.to(DEV), None, 1, 1, False)
+    ref = conv_ref(x, w, 1, 1)
+    assert torch.isfinite(y.float()).all(), "unwritten output columns"
+    close_bf16(y, ref, scale=ref.abs().max().clamp(min=0.5))
+    # the stats workspace must describe ALL of y (mean/var over every column)
+    g = torch.ones(K, dtype=torch.bfloat16, device=DEV)
+    b = torch.zeros(K, dtype=torch.bfloat16, device=DEV)
+    out, mean, rstd = EXT.bn_fwd_ws(
+        y.reshape(-1, K), g, b, ws, None, None, 0.1, 1e-5, False, None
+    )
+    yf = y.reshape(-1, K).float()
+    close_f32(mean, yf.mean(0).cpu(), tol=2e-3)
+    close_f32(rstd, (yf.var(0, unbiased=False) + 1e-5).rsqrt().cpu(), tol=2e-3)
 
 
 def test_conv2d_cpad_fallback_backward_shapes():
