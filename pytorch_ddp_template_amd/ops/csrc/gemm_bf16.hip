@@ -1214,9 +1214,11 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
   // (uninitialized-memory NaNs from the second step on — found by the
   // round-2 step-1 NaN bisect, tools/graph_dbg2.py).
   const bool extras_early = ex.stats_ws != nullptr || ex.addend != nullptr;
-  const bool wide_n =
-      (e_ntbn ? e_ntbn[0] == '2' : MODE != g16::MODE_PLAIN) && !narrow &&
-      !extras_early && N >= 256 && M >= 4096;
+  // Default OFF everywhere: the round-2 "conv-mode win" (+2.1%/+5.1%) was
+  // measured against the unwritten-half-of-C bug above; with the grid
+  // corrected, wide-N LOSES e2e (r18 bench 103.0k wide vs 105.8k off).
+  const bool wide_n = (e_ntbn ? e_ntbn[0] == '2' : false) && !narrow &&
+                      !extras_early && N >= 256 && M >= 4096;
   if (wide_n) grid.x = (N + 255) / 256;
   // 2-buffer wide-N: same 49 KB footprint as BNT=128x3buf (unchanged
   // co-residency) with 2x the MFMAs per barrier window.  PDT_NT_NBUF2=1.
@@ -1855,11 +1857,12 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
   // block-count target: more split-M blocks hide the staging latency (the
   // plain-TN z sweep measured +18-27% at ~1024 vs ~512); PDT_CONV_WGRAD_B
   // overrides for sweeps.
-  // e2e bench sweep at the flagship batch sizes: 1024 beats 512 (r50
-  // 7994 -> 8342 samples/s, r18 neutral) — same more-split-M-latency-hiding
-  // result as the plain-TN z sweep.
+  // e2e bench sweep at the flagship batch sizes AFTER the wide-N fix:
+  // 512 edges 1024 on r18 (103.5k vs 103.0k wide-on; 106.1k vs 105.8k
+  // wide-off) — the earlier 1024-wins result was contaminated by the
+  // unwritten-half-of-C bug in the fused conv forwards.
   static const char* e_cwb = getenv("PDT_CONV_WGRAD_B");
-  const int btarget = e_cwb ? atoi(e_cwb) : 1024;
+  const int btarget = e_cwb ? atoi(e_cwb) : 512;
   int z = std::max(1,
                    std::min(n_chunks, (btarget + tiles - 1) / std::max(1, tiles)));
   const long long ldc = (long long)R * S * Cin;

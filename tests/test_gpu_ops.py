@@ -132,6 +132,9 @@ def _poison_allocator(numel):
     junk = torch.full((numel,), float("nan"), dtype=torch.bfloat16, device=DEV)
     del junk
     torch.cuda.synchronize()
+
+
+def conv_ref(x, w, stride, pad):
     return (
         F.conv2d(
             x.float().permute(0, 3, 1, 2),
