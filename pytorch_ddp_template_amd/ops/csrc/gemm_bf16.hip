@@ -1190,7 +1190,8 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
                  torch::Tensor& C, const c10::optional<torch::Tensor>& bias,
                  bool relu, const torch::Tensor& zp, dim3 grid, int M, int N,
                  int K, long long sA, long long sB, long long sC,
-                 g16::ConvMeta cm, NtExtras ex = {}) {
+                 g16::ConvMeta cm, NtExtras ex = {},
+                 bool wide_default = false) {
   auto stream = c10::hip::getCurrentHIPStream();
   const t16* bias_p =
       bias.has_value() ? reinterpret_cast<const t16*>(bias->data_ptr())
@@ -1214,11 +1215,13 @@ void launch_nt16(const torch::Tensor& A, const torch::Tensor& B,
   // (uninitialized-memory NaNs from the second step on — found by the
   // round-2 step-1 NaN bisect, tools/graph_dbg2.py).
   const bool extras_early = ex.stats_ws != nullptr || ex.addend != nullptr;
-  // Default OFF everywhere: the round-2 "conv-mode win" (+2.1%/+5.1%) was
-  // measured against the unwritten-half-of-C bug above; with the grid
-  // corrected, wide-N LOSES e2e (r18 bench 103.0k wide vs 105.8k off).
-  const bool wide_n = (e_ntbn ? e_ntbn[0] == '2' : false) && !narrow &&
-                      !extras_early && N >= 256 && M >= 4096;
+  // Post-fix defaults: ON only where the call site asks (conv FORWARD
+  // without epilogue extras — the im2col A-side re-read argument and the
+  // ViT patchify win are real there).  The round-2 blanket conv-mode
+  // "win" was an artifact of the unwritten-half-of-C bug above; with the
+  // grid corrected, wide-N on DGRADs loses e2e (r18 103.0k vs 105.8k).
+  const bool wide_n = (e_ntbn ? e_ntbn[0] == '2' : wide_default) &&
+                      !narrow && !extras_early && N >= 256 && M >= 4096;
   if (wide_n) grid.x = (N + 255) / 256;
   // 2-buffer wide-N: same 49 KB footprint as BNT=128x3buf (unchanged
   // co-residency) with 2x the MFMAs per barrier window.  PDT_NT_NBUF2=1.
@@ -1390,10 +1393,11 @@ std::vector<torch::Tensor> conv2d_fwd_bf16_impl(
   }
   if (x.scalar_type() == torch::kBFloat16)
     launch_nt16<bf16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M, Kout,
-                                      K, 0, 0, 0, cm, ex);
+                                      K, 0, 0, 0, cm, ex, /*wide*/ true);
   else
     launch_nt16<_Float16, g16::MODE_CONV>(x, w, y, bias, relu, zp, grid, M,
-                                          Kout, K, 0, 0, 0, cm, ex);
+                                          Kout, K, 0, 0, 0, cm, ex,
+                                          /*wide*/ true);
   if (stats) return {y, ws};
   return {y};
 }
