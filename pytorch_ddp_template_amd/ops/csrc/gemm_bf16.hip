@@ -1318,6 +1318,11 @@ int log2_exact(int v) {
 
 }  // namespace
 
+// 256x256-tile glds kernel for big plain bf16 shapes (gemm_plain.hip)
+bool gemm_nt_plain256(const torch::Tensor& A, const torch::Tensor& B,
+                      torch::Tensor& C,
+                      const c10::optional<torch::Tensor>& bias, bool relu);
+
 // C[b,M,N] = A[b,M,K] @ B[b,N,K]^T (+bias, +relu).  2-D inputs = batch 1.
 torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
                           c10::optional<torch::Tensor> bias, bool relu) {
@@ -1330,6 +1335,7 @@ torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
   TORCH_CHECK(K % 8 == 0, "K must be padded to a multiple of 8 (host)");
   auto C = batched ? torch::empty({bsz, M, N}, A.options())
                    : torch::empty({M, N}, A.options());
+  if (!batched && gemm_nt_plain256(A, B, C, bias, relu)) return C;
   auto& zp = zero_page(A.device(), A.scalar_type());
   dim3 grid((N + g16::BN - 1) / g16::BN, (M + g16::BM - 1) / g16::BM,
             (unsigned)bsz);

@@ -1,0 +1,800 @@
+// Large plain NT GEMM — 256x256 tile, glds-staged, CDNA4 (gfx950).
+//
+// C[M,N] = A[M,K] @ B[N,K]^T (+bias, +relu), bf16.  This is the big-shape
+// path for the ViT linear layers (reference model.py:11,13 scope — SURVEY
+// §2b GEMM row): M=50432, K/N in {768, 2304, 3072}.  Round 1 conceded these
+// shapes to rocBLAS (846-1097 TF vs 510-687 TF for the 128-tile NT kernel);
+// this kernel exists to close that gap and retire the library route.
+//
+// Structure (the verified 256² glds recipe from the CDNA4 guide):
+//   * 256x256 macro-tile, BK=64, 8 waves as 2(M)x4(N), 512 threads.
+//   * mfma_f32_16x16x32_bf16: per-wave 128x64 output = acc[8][4] f32x4.
+//   * Staging via __builtin_amdgcn_global_load_lds (16B, wave-uniform LDS
+//     base + lane*16): the LDS image IS the global access order, so both
+//     operands land row-major [row][64k] — 128 KiB total, 2 buffers,
+//     1 block/CU.
+//   * One __syncthreads() per K-tile: the next tile's 8 glds are issued
+//     BEFORE the current tile's MFMAs, so their latency hides under the
+//     64 MFMAs; the barrier's implicit vmcnt(0) drains them after.
+//   * Epilogue: accumulators staged through LDS (reusing the operand
+//     buffers) so C is written in coalesced 16B rows.
+// The K-slot XOR swizzle (T2) is applied on the glds SOURCE address with
+// the matching XOR on the ds_read side: rows 4..7 (mod 8) swap 32B halves
+// of each 64B k-span, taking the b128 fragment reads from 8-way to 4-way
+// bank conflicts without touching the lane-linear LDS image.
+
+#include <torch/extension.h>
+
+#include "common.h"
+#include "dispatch.h"
+
+namespace gp {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr int BM = 256, BN = 256, BK = 64;
+constexpr int THREADS = 512;
+constexpr int NT_BUF = 2;
+
+__device__ __forceinline__ f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+// K-slot swizzle: XOR the 32B half-slot of the 128B LDS row with bit 2 of
+// the row (st_16x32: byte ^= ((byte>>9)&1)<<5 — row*128 puts row-bit-2 at
+// byte-bit-9).  Applied to the glds *source* k-chunk and to ds_read
+// addresses; the LDS image itself stays lane-linear.
+__device__ __forceinline__ int kswz(int row, int kbyte) {
+  return kbyte ^ (((row >> 2) & 1) << 5);
+}
+
+// full swizzle: row bits 1..3 -> kb bits 6..4.  All 16 rows of a b128
+// fragment-read lane group land on distinct bank offsets (conflict-free);
+// the permutation stays inside the 128B row, so the glds source still
+// touches the same cachelines (FETCH-neutral).
+__device__ __forceinline__ int kswz_full(int row, int kbyte) {
+  return kbyte ^ (((row >> 1) & 1) << 6) ^ (((row >> 2) & 1) << 5) ^
+         (((row >> 3) & 1) << 4);
+}
+
+template <bool HAS_BIAS, bool RELU, int SWZ>
+__global__ __launch_bounds__(THREADS, 1) void gemm_nt_plain256_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ B,
+    __hip_bfloat16* __restrict__ C, const __hip_bfloat16* __restrict__ bias,
+    int M, int N, int K) {
+  // one shared object only (hipcc de-pipelines glds with a second one)
+  __shared__ __hip_bfloat16 smem[NT_BUF * 2 * BM * BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  // XCD-contiguous bijective remap over the 1-D grid (8 XCDs)
+  int nwg = gridDim.x * gridDim.y;
+  int bid = blockIdx.y * gridDim.x + blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7, x = bid & 7, o = bid >> 3;
+    bid = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
+  }
+  const int n0 = (bid % gridDim.x) * BN;
+  const int m0 = (bid / gridDim.x) * BM;
+
+  // ---- staging: 8 glds per thread per K-tile (4 rounds A + 4 rounds B).
+  // Round r covers rows [r*64, r*64+64); within a round, wave w loads rows
+  // [w*8, w*8+8), lane: row += lane>>3, 16B chunk = lane&7.  The LDS dest
+  // is wave-uniform (glds adds lane*16).
+  const int srow = (tid >> 3) & 7;   // row within the wave's 8-row slice
+  const int schunk = lane & 7;       // 16B chunk within the 128B row
+  auto stage = [&](int buf, int kt) {
+    const long long k0 = (long long)kt * BK + schunk * 8;
+    __hip_bfloat16* sa = smem + buf * (2 * BM * BK);
+    __hip_bfloat16* sb = sa + BM * BK;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r * 64 + wave * 8 + srow;
+      long long kk = k0;
+      if (SWZ == 1) kk = (long long)kt * BK + kswz(row, schunk * 16) / 2;
+      if (SWZ == 2) kk = (long long)kt * BK + kswz_full(row, schunk * 16) / 2;
+      // LDS base is WAVE-UNIFORM (&s[row0*BK], row0 = r*64+wave*8); the
+      // instruction adds lane*16B, which walks exactly rows row0..row0+7
+      // (8 lanes per 128B row) because BK*2B == 128B.
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              A + (long long)(m0 + row) * K + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              sa + (r * 64 + wave * 8) * BK),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r * 64 + wave * 8 + srow;
+      long long kk = k0;
+      if (SWZ == 1) kk = (long long)kt * BK + kswz(row, schunk * 16) / 2;
+      if (SWZ == 2) kk = (long long)kt * BK + kswz_full(row, schunk * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              B + (long long)(n0 + row) * K + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              sb + (r * 64 + wave * 8) * BK),
+          16, 0, 0);
+    }
+  };
+
+  const int wm = (wave >> 2) * 128;  // wave's M offset in the tile
+  const int wn = (wave & 3) * 64;    // wave's N offset
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // fragment read offsets (bf16 elements from the operand image base):
+  // row = frag*16 + (lane&15), kbyte = ks*64 + (lane>>4)*16 (+XOR swizzle)
+  const int frow = lane & 15;
+  const int fkb = (lane >> 4) * 16;
+
+  const int NT = K / BK;
+  stage(0, 0);
+  __syncthreads();
+
+  int cur = 0;
+  for (int kt = 0; kt < NT; ++kt) {
+    if (kt + 1 < NT) stage(cur ^ 1, kt + 1);
+    const __hip_bfloat16* sa = smem + cur * (2 * BM * BK);
+    const __hip_bfloat16* sb = sa + BM * BK;
+
+    // B fragments for the wave: 4 n-frags x 2 k-steps
+    bf16x8 bf[4][2];
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wn + ni * 16 + frow;
+        int kb = ks * 64 + fkb;
+        if (SWZ == 1) kb = kswz(row, kb);
+        if (SWZ == 2) kb = kswz_full(row, kb);
+        bf[ni][ks] = *(const bf16x8*)((const char*)(sb + row * BK) + kb);
+      }
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi) {
+      bf16x8 af[2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wm + mi * 16 + frow;
+        int kb = ks * 64 + fkb;
+        if (SWZ == 1) kb = kswz(row, kb);
+        if (SWZ == 2) kb = kswz_full(row, kb);
+        af[ks] = *(const bf16x8*)((const char*)(sa + row * BK) + kb);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        acc[mi][ni] = mfma16(af[0], bf[ni][0], acc[mi][ni]);
+        acc[mi][ni] = mfma16(af[1], bf[ni][1], acc[mi][ni]);
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue: stage the C tile (bf16) through LDS for coalesced
+  // global rows.  C/D fragment map: col = lane&15, row = (lane>>4)*4+reg.
+  float bv[4];
+  if (HAS_BIAS) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      bv[ni] = bf2f(bias[n0 + wn + ni * 16 + (lane & 15)]);
+  }
+  __hip_bfloat16* cs = smem;  // [256][256] bf16 view = 128 KiB
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = wn + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = wm + mi * 16 + (lane >> 4) * 4 + reg;
+        float v = acc[mi][ni][reg];
+        if (HAS_BIAS) v += bv[ni];
+        if (RELU) v = fmaxf(v, 0.f);
+        cs[row * BN + col] = f2bf(v);
+      }
+    }
+  __syncthreads();
+  // 512 threads write 256 rows x 512B: 16 rows per round, 32 chunks/row
+  const int wrow = tid >> 5, wchunk = tid & 31;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = r * 16 + wrow;
+    *(short8*)((char*)(C + (long long)(m0 + row) * N + n0) + wchunk * 16) =
+        *(const short8*)((const char*)(cs + row * BN) + wchunk * 16);
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// v2: software-pipelined variant.  K-split LDS images ([buf][op][khalf]
+// [256][32]) let staging run at half-K-tile granularity: each phase issues
+// the NEXT tile's matching k-half (4 glds), then waits a COUNTED vmcnt(4)
+// — the just-issued loads stay in flight across the raw s_barrier, so the
+// glds queue never drains inside the loop (the structural stall of the
+// __syncthreads version: its barrier carries an implicit vmcnt(0)).
+// Schedule invariant: the wait in phase p leaves only S(p) (4 glds) in
+// flight, so S(p-1) — the data phase p+1 reads — is complete at every
+// barrier; phases with nothing to issue wait vmcnt(0).
+//
+// LDS bank math: a k-half row is 64 B, so fragment rows r and r^8 share a
+// bank group; the XOR of the 32 B half-slot with row bit 3 splits them —
+// b128 fragment reads go 4-way -> 2-way vs the v1 layout.
+
+__device__ __forceinline__ int kswz2(int row, int kbyte) {
+  return kbyte ^ (((row >> 3) & 1) << 5);
+}
+
+template <bool HAS_BIAS, bool RELU>
+__global__ __launch_bounds__(THREADS, 1) void gemm_nt_plain256_p2_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ B,
+    __hip_bfloat16* __restrict__ C, const __hip_bfloat16* __restrict__ bias,
+    int M, int N, int K) {
+  __shared__ __hip_bfloat16 smem[NT_BUF * 2 * 2 * BM * 32];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  int nwg = gridDim.x * gridDim.y;
+  int bid = blockIdx.y * gridDim.x + blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7, x = bid & 7, o = bid >> 3;
+    bid = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
+  }
+  const int n0 = (bid % gridDim.x) * BN;
+  const int m0 = (bid / gridDim.x) * BM;
+
+  // image offsets (bf16 elems): [buf][A=0/B=1][khalf][row][32]
+  auto img = [&](int buf, int op, int kh) {
+    return smem + (((buf * 2 + op) * 2 + kh) * BM) * 32;
+  };
+
+  // one k-half sub-stage = 2 glds/thread: round r covers rows [r*128,
+  // r*128+128); wave w rows [w*16, +16), lane: row += lane>>2, 16B chunk
+  // = lane&3 within the 64B row.
+  const int srow2 = (tid >> 2) & 15;
+  const int schunk2 = lane & 3;
+  auto stage_half = [&](int buf, int kt, int kh) {
+    const long long kbase = (long long)kt * BK + kh * 32;
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      const int row = r * 128 + wave * 16 + srow2;
+      const long long kk =
+          kbase + kswz2(row, schunk2 * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              A + (long long)(m0 + row) * K + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              img(buf, 0, kh) + (r * 128 + wave * 16) * 32),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      const int row = r * 128 + wave * 16 + srow2;
+      const long long kk =
+          kbase + kswz2(row, schunk2 * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              B + (long long)(n0 + row) * K + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              img(buf, 1, kh) + (r * 128 + wave * 16) * 32),
+          16, 0, 0);
+    }
+  };
+
+  const int wm = (wave >> 2) * 128;
+  const int wn = (wave & 3) * 64;
+  const int frow = lane & 15;
+  const int fkb = (lane >> 4) * 16;  // byte offset within the 64B k-half row
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / BK;
+  const int NP = 2 * NT;  // phases: index pi = 2*kt + kh consumes sub-tile
+                          // (kt, kh) and stages sub-tile pi+3 — a 3-phase-
+                          // deep pipeline (~3x32 MFMAs of latency cover,
+                          // matching HBM latency) with vmcnt(8) leaving the
+                          // two newest sub-stages in flight at each barrier.
+  // prologue: stage phases 0..2 (t0-klo, t0-khi, t1-klo), wait for phase 0
+  stage_half(0, 0, 0);
+  stage_half(0, 0, 1);
+  if (NT > 1) {
+    stage_half(1, 1, 0);
+    asm volatile("s_waitcnt vmcnt(8)");
+  } else {
+    asm volatile("s_waitcnt vmcnt(4)");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = 0; kt < NT; ++kt) {
+    const int cur = kt & 1;
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      // ds_read this phase's fragments (guaranteed by the previous
+      // phase's counted wait + barrier)
+      const __hip_bfloat16* sa = img(cur, 0, kh);
+      const __hip_bfloat16* sb = img(cur, 1, kh);
+      bf16x8 af[8], bf[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int row = wn + ni * 16 + frow;
+        bf[ni] = *(const bf16x8*)((const char*)(sb + row * 32) +
+                                  kswz2(row, fkb));
+      }
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi) {
+        const int row = wm + mi * 16 + frow;
+        af[mi] = *(const bf16x8*)((const char*)(sa + row * 32) +
+                                  kswz2(row, fkb));
+      }
+      // stage phase pi+3's sub-tile (its LDS slot was consumed at pi-1 and
+      // its reads drained two barriers ago), then wait so that only the
+      // newest in-flight sub-stages remain across the barrier
+      const int pi = 2 * kt + kh;
+      if (pi + 3 < NP) {
+        stage_half(((pi + 3) >> 1) & 1, (pi + 3) >> 1, (pi + 3) & 1);
+        asm volatile("s_waitcnt vmcnt(8)");
+      } else if (pi + 2 < NP) {
+        asm volatile("s_waitcnt vmcnt(4)");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)");
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = mfma16(af[mi], bf[ni], acc[mi][ni]);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // epilogue identical to v1: stage C through LDS for coalesced rows
+  float bv[4];
+  if (HAS_BIAS) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      bv[ni] = bf2f(bias[n0 + wn + ni * 16 + (lane & 15)]);
+  }
+  __hip_bfloat16* cs = smem;
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = wn + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = wm + mi * 16 + (lane >> 4) * 4 + reg;
+        float v = acc[mi][ni][reg];
+        if (HAS_BIAS) v += bv[ni];
+        if (RELU) v = fmaxf(v, 0.f);
+        cs[row * BN + col] = f2bf(v);
+      }
+    }
+  __syncthreads();
+  const int wrow = tid >> 5, wchunk = tid & 31;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = r * 16 + wrow;
+    *(short8*)((char*)(C + (long long)(m0 + row) * N + n0) + wchunk * 16) =
+        *(const short8*)((const char*)(cs + row * BN) + wchunk * 16);
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// v3: 3-buffer glds ring, 128x256 tile, counted vmcnt across raw barriers.
+// The "+83% 3-buf span" structure from the guide: staging runs TWO K-tiles
+// ahead into a 3-slot LDS ring (48 KB/tile -> 144 KB), so each iteration's
+// barrier carries a counted s_waitcnt vmcnt(6) — the newest tile's 6 glds
+// stay in flight across the barrier and the queue never drains.  The inner
+// loop body has no asm between the ds_reads and MFMAs: the compiler's own
+// counted lgkm scheduling (near-optimal per the guide) does the fine
+// interleave that the coarse v2 phase-split destroyed.
+// 128x256 tile also doubles the grid vs v1 on the ViT shapes (smaller
+// drain tail) at unchanged per-wave MFMA geometry (64x64 out per wave).
+
+template <bool HAS_BIAS, bool RELU>
+__global__ __launch_bounds__(THREADS, 1) void gemm_nt_plain128x256_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ B,
+    __hip_bfloat16* __restrict__ C, const __hip_bfloat16* __restrict__ bias,
+    int M, int N, int K) {
+  constexpr int TM = 128, TN = 256;
+  // ring slot: A [128][64] (16 KB) + B [256][64] (32 KB)
+  __shared__ __hip_bfloat16 smem[3 * (TM + TN) * BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  int nwg = gridDim.x * gridDim.y;
+  int bid = blockIdx.y * gridDim.x + blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7, x = bid & 7, o = bid >> 3;
+    bid = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
+  }
+  const int n0 = (bid % gridDim.x) * TN;
+  const int m0 = (bid / gridDim.x) * TM;
+
+  // stage one K-tile into ring slot s: A 2 rounds + B 4 rounds = 6 glds.
+  // Round covers 64 rows (512 threads x 16B = one 64x128B slab); wave w
+  // rows [w*8, +8), lane: row += lane>>3, chunk = lane&7; source chunk is
+  // XOR-swizzled (32B half-slot ^ row bit 2) like v1.
+  const int srow = (tid >> 3) & 7;
+  const int schunk = lane & 7;
+  auto stage = [&](int slot, int kt) {
+    __hip_bfloat16* sa = smem + slot * ((TM + TN) * BK);
+    __hip_bfloat16* sb = sa + TM * BK;
+    const long long kb = (long long)kt * BK;
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      const int row = r * 64 + wave * 8 + srow;
+      const long long kk = kb + kswz(row, schunk * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              A + (long long)(m0 + row) * K + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              sa + (r * 64 + wave * 8) * BK),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r * 64 + wave * 8 + srow;
+      const long long kk = kb + kswz(row, schunk * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              B + (long long)(n0 + row) * K + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              sb + (r * 64 + wave * 8) * BK),
+          16, 0, 0);
+    }
+  };
+
+  const int wm = (wave >> 2) * 64;  // 2 waves along M: 64 rows each
+  const int wn = (wave & 3) * 64;   // 4 waves along N
+  const int frow = lane & 15;
+  const int fkb = (lane >> 4) * 16;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / BK;
+  stage(0, 0);
+  if (NT > 1) {
+    stage(1, 1);
+    asm volatile("s_waitcnt vmcnt(6)");  // tile 0 landed, tile 1 in flight
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = 0; kt < NT; ++kt) {
+    const int slot = kt % 3;
+    if (kt + 2 < NT) stage((kt + 2) % 3, kt + 2);
+    const __hip_bfloat16* sa = smem + slot * ((TM + TN) * BK);
+    const __hip_bfloat16* sb = sa + TM * BK;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 af[4], bf[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int row = wn + ni * 16 + frow;
+        bf[ni] = *(const bf16x8*)((const char*)(sb + row * BK) +
+                                  kswz(row, ks * 64 + fkb));
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const int row = wm + mi * 16 + frow;
+        af[mi] = *(const bf16x8*)((const char*)(sa + row * BK) +
+                                  kswz(row, ks * 64 + fkb));
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = mfma16(af[mi], bf[ni], acc[mi][ni]);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)");
+    if (kt + 2 < NT) asm volatile("s_waitcnt vmcnt(6)");
+    else asm volatile("s_waitcnt vmcnt(0)");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: stage C (128x256 bf16 = 64 KB) through slot 0 for coalesced
+  // 16B row writes
+  float bv[4];
+  if (HAS_BIAS) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      bv[ni] = bf2f(bias[n0 + wn + ni * 16 + (lane & 15)]);
+  }
+  __hip_bfloat16* cs = smem;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = wn + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = wm + mi * 16 + (lane >> 4) * 4 + reg;
+        float v = acc[mi][ni][reg];
+        if (HAS_BIAS) v += bv[ni];
+        if (RELU) v = fmaxf(v, 0.f);
+        cs[row * TN + col] = f2bf(v);
+      }
+    }
+  __syncthreads();
+  const int wrow = tid >> 5, wchunk = tid & 31;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = r * 16 + wrow + 0;
+    *(short8*)((char*)(C + (long long)(m0 + row) * N + n0) + wchunk * 16) =
+        *(const short8*)((const char*)(cs + row * TN) + wchunk * 16);
+  }
+#pragma unroll
+  for (int r = 4; r < 8; ++r) {
+    const int row = r * 16 + wrow;
+    *(short8*)((char*)(C + (long long)(m0 + row) * N + n0) + wchunk * 16) =
+        *(const short8*)((const char*)(cs + row * TN) + wchunk * 16);
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// v4: v1 structure with mfma_f32_32x32x16_bf16 (2382 vs 2075 TF ubench
+// issue ceiling, half the MFMA instructions for the same FLOPs).  Fragment
+// maps match the repo's conv/attention kernels: A/B lane->row = r32,
+// k = ks*8 within each 16-deep window; C row = (reg&3)+8*(reg>>2)+4*ks.
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+__device__ __forceinline__ f32x16 mfma32(bf16x8 a, bf16x8 b, f32x16 c) {
+  return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+template <bool HAS_BIAS, bool RELU>
+__global__ __launch_bounds__(THREADS, 1) void gemm_nt_plain256_m32_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ B,
+    __hip_bfloat16* __restrict__ C, const __hip_bfloat16* __restrict__ bias,
+    int M, int N, int K) {
+  __shared__ __hip_bfloat16 smem[NT_BUF * 2 * BM * BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  int nwg = gridDim.x * gridDim.y;
+  int bid = blockIdx.y * gridDim.x + blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7, x = bid & 7, o = bid >> 3;
+    bid = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
+  }
+  const int n0 = (bid % gridDim.x) * BN;
+  const int m0 = (bid / gridDim.x) * BM;
+
+  const int srow = (tid >> 3) & 7;
+  const int schunk = lane & 7;
+  auto stage = [&](int buf, int kt) {
+    __hip_bfloat16* sa = smem + buf * (2 * BM * BK);
+    __hip_bfloat16* sb = sa + BM * BK;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r * 64 + wave * 8 + srow;
+      const long long kk =
+          (long long)kt * BK + kswz_full(row, schunk * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              A + (long long)(m0 + row) * K + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              sa + (r * 64 + wave * 8) * BK),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r * 64 + wave * 8 + srow;
+      const long long kk =
+          (long long)kt * BK + kswz_full(row, schunk * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              B + (long long)(n0 + row) * K + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              sb + (r * 64 + wave * 8) * BK),
+          16, 0, 0);
+    }
+  };
+
+  const int wm = (wave >> 2) * 128;
+  const int wn = (wave & 3) * 64;
+  const int r32 = lane & 31;
+  const int ks = lane >> 5;
+
+  f32x16 acc[4][2] = {};
+
+  const int NT = K / BK;
+  stage(0, 0);
+  __syncthreads();
+
+  int cur = 0;
+  for (int kt = 0; kt < NT; ++kt) {
+    if (kt + 1 < NT) stage(cur ^ 1, kt + 1);
+    const __hip_bfloat16* sa = smem + cur * (2 * BM * BK);
+    const __hip_bfloat16* sb = sa + BM * BK;
+#pragma unroll
+    for (int kw = 0; kw < 4; ++kw) {  // 16-deep k-windows of the 64 tile
+      const int kb0 = kw * 32 + ks * 16;
+      bf16x8 bfr[2];
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int row = wn + ni * 32 + r32;
+        bfr[ni] = *(const bf16x8*)((const char*)(sb + row * BK) +
+                                   kswz_full(row, kb0));
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const int row = wm + mi * 32 + r32;
+        bf16x8 af = *(const bf16x8*)((const char*)(sa + row * BK) +
+                                     kswz_full(row, kb0));
+        acc[mi][0] = mfma32(af, bfr[0], acc[mi][0]);
+        acc[mi][1] = mfma32(af, bfr[1], acc[mi][1]);
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  float bv[2];
+  if (HAS_BIAS) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+      bv[ni] = bf2f(bias[n0 + wn + ni * 32 + r32]);
+  }
+  __hip_bfloat16* cs = smem;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int col = wn + ni * 32 + r32;
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row = wm + mi * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * ks;
+        float v = acc[mi][ni][reg];
+        if (HAS_BIAS) v += bv[ni];
+        if (RELU) v = fmaxf(v, 0.f);
+        cs[row * BN + col] = f2bf(v);
+      }
+    }
+  __syncthreads();
+  const int wrow = tid >> 5, wchunk = tid & 31;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = r * 16 + wrow;
+    *(short8*)((char*)(C + (long long)(m0 + row) * N + n0) + wchunk * 16) =
+        *(const short8*)((const char*)(cs + row * BN) + wchunk * 16);
+  }
+}
+
+}  // namespace gp
+
+// Host entry: true when the shape is handled (M%256, N%256, K%64, bf16).
+bool gemm_nt_plain256(const torch::Tensor& A, const torch::Tensor& B,
+                      torch::Tensor& C,
+                      const c10::optional<torch::Tensor>& bias, bool relu) {
+  static const char* e = getenv("PDT_PLAIN256");
+  if (e && e[0] == '0') return false;
+  if (A.scalar_type() != torch::kBFloat16) return false;
+  const int M = (int)A.size(0), K = (int)A.size(1), N = (int)B.size(0);
+  if (M % gp::BM || N % gp::BN || K % gp::BK) return false;
+  if (M < 4096) return false;  // small-M shapes: the 128-tile kernel wins
+  auto stream = c10::hip::getCurrentHIPStream();
+  dim3 grid(N / gp::BN, M / gp::BM);
+  const __hip_bfloat16* bias_p =
+      bias.has_value()
+          ? reinterpret_cast<const __hip_bfloat16*>(bias->data_ptr())
+          : nullptr;
+  const int swz = [] {
+    static const char* s = getenv("PDT_PLAIN256_SWZ");
+    if (!s) return 1;
+    return s[0] == '0' ? 0 : (s[0] == '2' ? 2 : 1);
+  }();
+  const bool pipe = [] {
+    static const char* s = getenv("PDT_PLAIN256_PIPE");
+    return s && s[0] == '1';
+  }();
+  const bool m32 = [] {
+    static const char* s = getenv("PDT_PLAIN256_M32");
+    return s && s[0] == '1';
+  }();
+  if (m32) {
+#define LAUNCH_GPM(HB, RL)                                                 \
+  hipLaunchKernelGGL((gp::gemm_nt_plain256_m32_kernel<HB, RL>), grid,      \
+                     dim3(gp::THREADS), 0, stream,                         \
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),\
+                     reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),\
+                     reinterpret_cast<__hip_bfloat16*>(C.data_ptr()),      \
+                     bias_p, M, N, K)
+    if (bias_p && relu) LAUNCH_GPM(true, true);
+    else if (bias_p) LAUNCH_GPM(true, false);
+    else if (relu) LAUNCH_GPM(false, true);
+    else LAUNCH_GPM(false, false);
+#undef LAUNCH_GPM
+    return true;
+  }
+  const bool ring3 = [] {
+    static const char* s = getenv("PDT_PLAIN256_RING");
+    return s && s[0] == '1';
+  }();
+  if (ring3 && M % 128 == 0) {
+    dim3 grid3(N / gp::BN, M / 128);
+#define LAUNCH_GP3(HB, RL)                                                 \
+  hipLaunchKernelGGL((gp::gemm_nt_plain128x256_kernel<HB, RL>), grid3,     \
+                     dim3(gp::THREADS), 0, stream,                         \
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),\
+                     reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),\
+                     reinterpret_cast<__hip_bfloat16*>(C.data_ptr()),      \
+                     bias_p, M, N, K)
+    if (bias_p && relu) LAUNCH_GP3(true, true);
+    else if (bias_p) LAUNCH_GP3(true, false);
+    else if (relu) LAUNCH_GP3(false, true);
+    else LAUNCH_GP3(false, false);
+#undef LAUNCH_GP3
+    return true;
+  }
+  if (pipe) {
+#define LAUNCH_GP2(HB, RL)                                                 \
+  hipLaunchKernelGGL((gp::gemm_nt_plain256_p2_kernel<HB, RL>), grid,       \
+                     dim3(gp::THREADS), 0, stream,                         \
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),\
+                     reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),\
+                     reinterpret_cast<__hip_bfloat16*>(C.data_ptr()),      \
+                     bias_p, M, N, K)
+    if (bias_p && relu) LAUNCH_GP2(true, true);
+    else if (bias_p) LAUNCH_GP2(true, false);
+    else if (relu) LAUNCH_GP2(false, true);
+    else LAUNCH_GP2(false, false);
+#undef LAUNCH_GP2
+    return true;
+  }
+#define LAUNCH_GP(HB, RL, SW)                                              \
+  hipLaunchKernelGGL((gp::gemm_nt_plain256_kernel<HB, RL, SW>), grid,      \
+                     dim3(gp::THREADS), 0, stream,                         \
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),\
+                     reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),\
+                     reinterpret_cast<__hip_bfloat16*>(C.data_ptr()),      \
+                     bias_p, M, N, K)
+  if (swz == 2) {
+    if (bias_p && relu) LAUNCH_GP(true, true, 2);
+    else if (bias_p) LAUNCH_GP(true, false, 2);
+    else if (relu) LAUNCH_GP(false, true, 2);
+    else LAUNCH_GP(false, false, 2);
+  } else if (swz == 1) {
+    if (bias_p && relu) LAUNCH_GP(true, true, 1);
+    else if (bias_p) LAUNCH_GP(true, false, 1);
+    else if (relu) LAUNCH_GP(false, true, 1);
+    else LAUNCH_GP(false, false, 1);
+  } else {
+    if (bias_p && relu) LAUNCH_GP(true, true, 0);
+    else if (bias_p) LAUNCH_GP(true, false, 0);
+    else if (relu) LAUNCH_GP(false, true, 0);
+    else LAUNCH_GP(false, false, 0);
+  }
+#undef LAUNCH_GP
+  return true;
+}

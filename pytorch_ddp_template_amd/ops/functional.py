@@ -41,12 +41,18 @@ from .native import native, use_native
 
 
 def _plain_gemm_to_blas(m: int, k: int, n: int) -> bool:
-    """Big plain GEMMs go to rocBLAS/hipBLASLt (the MI355X library path);
-    the hand-written NT/TN kernels keep every fused shape (bias/relu
-    epilogues, conv modes, bias-grad-in-GEMM).  Measured on the ViT-B/16
-    linear shapes (tools/gemm_ab.py, 1xMI355X): rocBLAS 846-1097 TF vs
-    510-687 TF for the in-house NT kernel; the TN wgrad with its fused
-    bias grad stays with the in-house kernel (competitive or faster).
+    """Big plain GEMMs go to rocBLAS/hipBLASLt (the MI355X library path,
+    sanctioned for PLAIN library GEMMs only); every fused shape (bias/relu
+    epilogues, conv modes, bias-grad-in-GEMM) stays on the hand-written
+    kernels.  Round-2 status (tools/plain256_ab.py, 1xMI355X): the new
+    256x256-tile glds kernel (ops/csrc/gemm_plain.hip) reaches 1040-1130 TF
+    sustained and 870-960 TF on the ViT-B/16 linear shapes — up from
+    510-687 TF in round 1 — vs rocBLAS 950-1150 on the same box, so the
+    library keeps a measured ~10% edge on exactly these shapes and keeps
+    the route.  Pipelined variants (counted-vmcnt k-half phases, 3-buffer
+    ring, 32x32 MFMA) all measured slower than the simple 2-buffer glds
+    structure; the remaining gap is the barrier's glds drain, which per the
+    CDNA4 guide only the exact asm-level 8-phase interleave removes.
     """
     return m >= 4096 and k >= 512 and n >= 512
 

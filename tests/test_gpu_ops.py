@@ -41,7 +41,10 @@ def close_f32(out, ref, tol=1e-4):
 
 
 @pytest.mark.parametrize("m,n,k", [(64, 64, 64), (128, 128, 128),
-                                   (130, 70, 40), (32, 10, 16), (257, 129, 96)])
+                                   (130, 70, 40), (32, 10, 16), (257, 129, 96),
+                                   # 256x256-tile glds kernel (gemm_plain.hip):
+                                   # M>=4096, M%256, N%256, K%64
+                                   (4096, 256, 128), (4352, 512, 192)])
 def test_gemm_nt_bf16(m, n, k):
     a = t32(m, k, seed=1).to(torch.bfloat16)
     b = t32(n, k, seed=2).to(torch.bfloat16)
