@@ -1783,7 +1783,8 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
     if (more) stage(buf ^ 1, ch + 1);
     const T16* base = lds + buf * (TILE_A + 3 * TILE_X);
 
-    // A fragments (dy), as in the generic tr kernel
+    // A fragments (dy): 4 tr reads issued, drained by the FIRST counted
+    // wait of the B pipeline below (not a full lgkmcnt(0) drain)
     vec16 af[2];
     {
       const unsigned a0 =
@@ -1802,30 +1803,52 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
       reinterpret_cast<v4s*>(&af[1])[0] = l1;
       reinterpret_cast<v4s*>(&af[1])[1] = h1;
     }
+    // B-pair software pipeline: 18 (2x tr-read, MFMA) pairs; pair i+1's
+    // reads are issued BEFORE pair i's MFMA and the wait is a counted
+    // lgkmcnt(2) (pair i done, pair i+1 pending) — the old full
+    // lgkmcnt(0) before every MFMA serialized LDS-read -> MFMA and kept
+    // the matrix pipe half-fed (round-1 PMC stopping point).
+    {
+      const unsigned lane_b =
+          (unsigned)(((lane & 15) >> 2) * 32 + (lane & 3) * 8);
+      // 3-slot rotation: the ds_read that REWRITES a slot is issued two
+      // MFMAs after the MFMA that sourced it (a 2-slot ping-pong put the
+      // async LDS write inside the in-flight MFMA's source-read window —
+      // WAR hazard, caught by the cold-launch stress test)
+      v4s pb[3][2];
+      auto issueB = [&](int slot, int idx) {
+        const int rt = idx / 6, rem = idx % 6, s2 = rem >> 1, kh = rem & 1;
+        const T16* timg =
+            base + TILE_A + rt * TILE_X + (img_sel + (wn >> 4)) * IMG_X;
+        const unsigned b0 =
+            LDS_BYTE(timg) + (unsigned)((qw[kh] + s2) * 32) + lane_b;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+            "ds_read_b64_tr_b16 %1, %3 offset:0"
+            : "=&v"(pb[slot][0]), "=&v"(pb[slot][1])
+            : "v"(b0), "v"(b0 + (unsigned)jump));
+      };
+      issueB(0, 0);
 #pragma unroll
-    for (int rt = 0; rt < 3; ++rt) {
-      const T16* timg =
-          base + TILE_A + rt * TILE_X + (img_sel + (wn >> 4)) * IMG_X;
-#pragma unroll
-      for (int s2 = 0; s2 < 3; ++s2) {
-#pragma unroll
-        for (int kh = 0; kh < 2; ++kh) {
-          const unsigned b0 =
-              LDS_BYTE(timg) + (unsigned)((qw[kh] + s2) * 32) +
-              (unsigned)(((lane & 15) >> 2) * 32 + (lane & 3) * 8);
-          const unsigned b1 = b0 + (unsigned)jump;
-          v4s l0, h0;
-          asm volatile(
-              "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
-              "ds_read_b64_tr_b16 %1, %3 offset:0\n\t"
-              "s_waitcnt lgkmcnt(0)"
-              : "=&v"(l0), "=&v"(h0)
-              : "v"(b0), "v"(b1));
-          vec16 bf;
-          reinterpret_cast<v4s*>(&bf)[0] = l0;
-          reinterpret_cast<v4s*>(&bf)[1] = h0;
-          acc[rt * 3 + s2] = M16<T16>::mma32(af[kh], bf, acc[rt * 3 + s2]);
+      for (int idx = 0; idx < 18; ++idx) {
+        const int slot = idx % 3;
+        // the wait carries the pair's registers as "+v" operands: the
+        // compiler must order the copies/MFMA that consume them AFTER the
+        // wait (a bare asm waitcnt does not stop it hoisting the v_movs
+        // above the wait — stale-register NaNs at some shapes)
+        if (idx + 1 < 18) {
+          issueB((idx + 1) % 3, idx + 1);
+          asm volatile("s_waitcnt lgkmcnt(2)"
+                       : "+v"(pb[slot][0]), "+v"(pb[slot][1]));
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(0)"
+                       : "+v"(pb[slot][0]), "+v"(pb[slot][1]));
         }
+        const int rt = idx / 6, rem = idx % 6, s2 = rem >> 1, kh = rem & 1;
+        vec16 bf;
+        reinterpret_cast<v4s*>(&bf)[0] = pb[slot][0];
+        reinterpret_cast<v4s*>(&bf)[1] = pb[slot][1];
+        acc[rt * 3 + s2] = M16<T16>::mma32(af[kh], bf, acc[rt * 3 + s2]);
       }
     }
     if (more) __syncthreads();
