@@ -1357,6 +1357,11 @@ torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
 bool conv2d_fwd_halo(torch::Tensor x, torch::Tensor w, torch::Tensor y,
                      torch::Tensor zp, int64_t stride, int64_t pad,
                      float* stats_ws, int ws_nblocks);
+// glds implicit-GEMM conv fwd (conv_glds.hip): pow2 spatial, C%64==0
+bool conv2d_fwd_glds(const torch::Tensor& x, const torch::Tensor& w,
+                     torch::Tensor& y, const torch::Tensor& zp,
+                     int64_t stride, int64_t pad, float* stats_ws,
+                     int ws_nblocks);
 
 // conv2d forward, NHWC x[N,H,W,C] * w[Kout,R,S,C] -> y[N,HO,WO,Kout]
 // stats=true additionally returns the BN partial workspace ws
@@ -1390,6 +1395,12 @@ std::vector<torch::Tensor> conv2d_fwd_bf16_impl(
                       x.options().dtype(torch::kFloat32));
     ex.stats_ws = ws.data_ptr<float>();
     ex.ws_nblocks = nblocks;
+  }
+  if (!relu && !bias.has_value() &&
+      conv2d_fwd_glds(x, w, y, zp, stride, pad, ex.stats_ws,
+                      ex.ws_nblocks)) {
+    if (stats) return {y, ws};
+    return {y};
   }
   if (!relu && !bias.has_value() &&
       conv2d_fwd_halo(x, w, y, zp, stride, pad, ex.stats_ws,
@@ -1452,6 +1463,9 @@ torch::Tensor conv2d_dgrad_bf16(torch::Tensor dy, torch::Tensor wr,
                    (int)stride, HOs, WOs};
   // dense (stride-1) dgrad is exactly a 3x3 stride-1 pad-1 conv of dy with
   // the rotated weight: take the LDS-halo fast path when it applies
+  if ((int)stride == 1 && !addend.has_value() &&
+      conv2d_fwd_glds(dy, wr, dx, zp, 1, dpad, nullptr, 0))
+    return dx;
   if ((int)stride == 1 && !addend.has_value() &&
       conv2d_fwd_halo(dy, wr, dx, zp, 1, dpad, nullptr, 0))
     return dx;
