@@ -44,6 +44,11 @@ __device__ __forceinline__ int kswz(int row, int kbyte) {
 
 struct Geom {
   int H, W, C, lgC, lgHO, lgWO, stride, pad, S, RS;  // S = kernel width
+  // transposed-conv (stride-s dgrad) gather: the logical image is the
+  // zero-STUFFED dy — coordinates valid only on the stride grid, source
+  // indexed by the un-stuffed position.  lgSt=0 -> plain conv (H,W = real
+  // source dims).
+  int lgSt = 0;
 };
 
 template <int BM, int BN, bool STATS>
@@ -80,7 +85,7 @@ __global__ __launch_bounds__(THREADS, 1) void conv_fwd_glds_kernel(
   // Per-row gather bases hoisted out of the K-loop: this thread stages the
   // same M-rows every tile.  hi/wi validity is per-tap, so keep the
   // unshifted coordinates.
-  long long abase[RA];
+  long long nbase[RA];
   int hv[RA], wv[RA];
 #pragma unroll
   for (int r = 0; r < RA; ++r) {
@@ -91,8 +96,9 @@ __global__ __launch_bounds__(THREADS, 1) void conv_fwd_glds_kernel(
     const long long n = t >> g.lgHO;
     hv[r] = ho * g.stride - g.pad;
     wv[r] = wo * g.stride - g.pad;
-    abase[r] = ((n * g.H + hv[r]) * (long long)g.W + wv[r]) << g.lgC;
+    nbase[r] = n * g.H * (long long)g.W;
   }
+  const int smask = (1 << g.lgSt) - 1;
 
   auto stage = [&](int buf, int kt) {
     __hip_bfloat16* sa = smem + buf * ((BM + BN) * BK);
@@ -100,15 +106,19 @@ __global__ __launch_bounds__(THREADS, 1) void conv_fwd_glds_kernel(
     const int tap = (kt << 6) >> g.lgC;
     const int c0 = (kt << 6) & (g.C - 1);
     const int rt = tap / g.S, st = tap - rt * g.S;
-    const long long toff = ((long long)rt * g.W + st) << g.lgC;
 #pragma unroll
     for (int r = 0; r < RA; ++r) {
       const int row = r * 64 + wave * 8 + srow;
       const int kk = c0 + kswz(row, schunk * 16) / 2;
       const __hip_bfloat16* src = zpad;
-      if ((unsigned)(hv[r] + rt) < (unsigned)g.H &&
-          (unsigned)(wv[r] + st) < (unsigned)g.W)
-        src = x + abase[r] + toff + kk;
+      const int hs = hv[r] + rt, ws = wv[r] + st;
+      if ((unsigned)(hs >> g.lgSt) < (unsigned)g.H &&
+          (unsigned)(ws >> g.lgSt) < (unsigned)g.W &&
+          ((hs | ws) & smask) == 0)
+        src = x + ((nbase[r] + (long long)(hs >> g.lgSt) * g.W +
+                    (ws >> g.lgSt))
+                   << g.lgC) +
+              kk;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)(
@@ -240,10 +250,10 @@ __global__ __launch_bounds__(THREADS, 1) void conv_fwd_glds_kernel(
 
 // Host entry: returns true when handled.  y must be pre-allocated
 // [N,HO,WO,Kout]; stats_ws (optional) pre-zeroed [2*ws_nblocks, Kout].
-bool conv2d_fwd_glds(const torch::Tensor& x, const torch::Tensor& w,
-                     torch::Tensor& y, const torch::Tensor& zp,
-                     int64_t stride, int64_t pad, float* stats_ws,
-                     int ws_nblocks) {
+bool conv2d_fwd_glds_ex(const torch::Tensor& x, const torch::Tensor& w,
+                        torch::Tensor& y, const torch::Tensor& zp,
+                        int64_t stride, int64_t pad, float* stats_ws,
+                        int ws_nblocks, int lgSt) {
   static const char* e = getenv("PDT_CONV_GLDS");
   if (e && e[0] == '0') return false;
   if (x.scalar_type() != torch::kBFloat16) return false;
@@ -269,7 +279,8 @@ bool conv2d_fwd_glds(const torch::Tensor& x, const torch::Tensor& w,
     else return false;
   }
   if (M / BM > 2147483647LL / 8) return false;
-  cg::Geom g{H, W, C, lgC, lgHO, lgWO, (int)stride, (int)pad, S, R * S};
+  cg::Geom g{H, W, C, lgC, lgHO, lgWO, (int)stride, (int)pad, S, R * S,
+             lgSt};
   auto stream = c10::hip::getCurrentHIPStream();
   dim3 grid(Kout / BN, (unsigned)(M / BM));
   const auto* xp = reinterpret_cast<const __hip_bfloat16*>(x.data_ptr());
@@ -293,4 +304,12 @@ bool conv2d_fwd_glds(const torch::Tensor& x, const torch::Tensor& w,
   else LAUNCH_CG(256, 64);
 #undef LAUNCH_CG
   return true;
+}
+
+bool conv2d_fwd_glds(const torch::Tensor& x, const torch::Tensor& w,
+                     torch::Tensor& y, const torch::Tensor& zp,
+                     int64_t stride, int64_t pad, float* stats_ws,
+                     int ws_nblocks) {
+  return conv2d_fwd_glds_ex(x, w, y, zp, stride, pad, stats_ws, ws_nblocks,
+                            0);
 }

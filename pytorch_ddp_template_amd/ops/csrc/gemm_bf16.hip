@@ -1362,6 +1362,10 @@ bool conv2d_fwd_glds(const torch::Tensor& x, const torch::Tensor& w,
                      torch::Tensor& y, const torch::Tensor& zp,
                      int64_t stride, int64_t pad, float* stats_ws,
                      int ws_nblocks);
+bool conv2d_fwd_glds_ex(const torch::Tensor& x, const torch::Tensor& w,
+                        torch::Tensor& y, const torch::Tensor& zp,
+                        int64_t stride, int64_t pad, float* stats_ws,
+                        int ws_nblocks, int lgSt);
 
 // conv2d forward, NHWC x[N,H,W,C] * w[Kout,R,S,C] -> y[N,HO,WO,Kout]
 // stats=true additionally returns the BN partial workspace ws
@@ -1463,9 +1467,18 @@ torch::Tensor conv2d_dgrad_bf16(torch::Tensor dy, torch::Tensor wr,
                    (int)stride, HOs, WOs};
   // dense (stride-1) dgrad is exactly a 3x3 stride-1 pad-1 conv of dy with
   // the rotated weight: take the LDS-halo fast path when it applies
-  if ((int)stride == 1 && !addend.has_value() &&
-      conv2d_fwd_glds(dy, wr, dx, zp, 1, dpad, nullptr, 0))
-    return dx;
+  // glds path: stride-1 dgrad is a plain conv of dy with the rotated
+  // weight; pow2-stride dgrad uses the stuffed-coordinate gather (valid
+  // only on the stride grid, indexed by the un-stuffed position)
+  static const char* e_gs2 = getenv("PDT_CONV_GLDS_S2");
+  const bool glds_s2 = !(e_gs2 && e_gs2[0] == '0');
+  if (!addend.has_value() && ((int)stride & ((int)stride - 1)) == 0 &&
+      ((int)stride == 1 || glds_s2)) {
+    const int lgSt = log2_exact((int)stride);
+    if (lgSt >= 0 &&
+        conv2d_fwd_glds_ex(dy, wr, dx, zp, 1, dpad, nullptr, 0, lgSt))
+      return dx;
+  }
   if ((int)stride == 1 && !addend.has_value() &&
       conv2d_fwd_halo(dy, wr, dx, zp, 1, dpad, nullptr, 0))
     return dx;
