@@ -43,7 +43,7 @@ __device__ __forceinline__ int kswz(int row, int kbyte) {
 }
 
 struct Geom {
-  int H, W, C, lgC, lgHO, lgWO, stride, pad, S, RS;  // S = kernel width
+  int H, W, C, lgC, HO, WO, stride, pad, S, RS;  // S = kernel width
   // transposed-conv (stride-s dgrad) gather: the logical image is the
   // zero-STUFFED dy — coordinates valid only on the stride grid, source
   // indexed by the un-stuffed position.  lgSt=0 -> plain conv (H,W = real
@@ -89,11 +89,13 @@ __global__ __launch_bounds__(THREADS, 1) void conv_fwd_glds_kernel(
   int hv[RA], wv[RA];
 #pragma unroll
   for (int r = 0; r < RA; ++r) {
+    // hoisted out of the K-loop, so plain div/mod (non-pow2 spatial: the
+    // ResNet-50 224-input pyramid is 56/28/14/7) costs nothing hot
     const long long m = m0 + r * 64 + wave * 8 + srow;
-    const int wo = (int)(m & ((1 << g.lgWO) - 1));
-    const long long t = m >> g.lgWO;
-    const int ho = (int)(t & ((1 << g.lgHO) - 1));
-    const long long n = t >> g.lgHO;
+    const int wo = (int)(m % g.WO);
+    const long long t = m / g.WO;
+    const int ho = (int)(t % g.HO);
+    const long long n = t / g.HO;
     hv[r] = ho * g.stride - g.pad;
     wv[r] = wo * g.stride - g.pad;
     nbase[r] = n * g.H * (long long)g.W;
@@ -267,8 +269,7 @@ bool conv2d_fwd_glds_ex(const torch::Tensor& x, const torch::Tensor& w,
     while ((1 << l) < v) ++l;
     return ((1 << l) == v) ? l : -1;
   };
-  const int lgC = pow2l(C), lgHO = pow2l(HO), lgWO = pow2l(WO);
-  if (lgHO < 0 || lgWO < 0) return false;
+  const int lgC = pow2l(C);
   const long long M = (long long)N_ * HO * WO;
   const int K = R * S * C;
   int BN = Kout % 256 == 0 ? 256 : Kout % 128 == 0 ? 128 : 64;
@@ -279,8 +280,7 @@ bool conv2d_fwd_glds_ex(const torch::Tensor& x, const torch::Tensor& w,
     else return false;
   }
   if (M / BM > 2147483647LL / 8) return false;
-  cg::Geom g{H, W, C, lgC, lgHO, lgWO, (int)stride, (int)pad, S, R * S,
-             lgSt};
+  cg::Geom g{H, W, C, lgC, HO, WO, (int)stride, (int)pad, S, R * S, lgSt};
   auto stream = c10::hip::getCurrentHIPStream();
   dim3 grid(Kout / BN, (unsigned)(M / BM));
   const auto* xp = reinterpret_cast<const __hip_bfloat16*>(x.data_ptr());

@@ -160,6 +160,7 @@ def conv_ref(x, w, stride, pad):
         (2, 32, 3, 64, 3, 1, 1),      # stem (im2col fallback)
         (1, 16, 64, 64, 1, 2, 0),     # 1x1 stride-2 (downsample)
         (16, 16, 256, 256, 3, 1, 1),  # wide-N variant (M>=4096, N>=256)
+        (64, 14, 64, 64, 3, 1, 1),    # glds path, non-pow2 spatial (r50)
     ],
 )
 def test_conv2d_fwd_bf16(n, h, c, k, r, stride, pad):
@@ -174,7 +175,9 @@ def test_conv2d_fwd_bf16(n, h, c, k, r, stride, pad):
 @pytest.mark.parametrize(
     "n,h,c,k,r,stride,pad",
     [(2, 16, 64, 64, 3, 1, 1), (2, 16, 64, 128, 3, 2, 1), (1, 8, 64, 64, 1, 2, 0),
-     (16, 16, 256, 256, 3, 1, 1)],  # wide-N variant (M>=4096, N>=256)
+     (16, 16, 256, 256, 3, 1, 1),   # wide-N variant (M>=4096, N>=256)
+     (64, 14, 64, 64, 3, 1, 1),     # glds path, non-pow2 spatial (r50)
+     (64, 28, 64, 128, 3, 2, 1)],   # glds stuffed dgrad, non-pow2
 )
 def test_conv2d_dgrad_bf16(n, h, c, k, r, stride, pad):
     ho = (h + 2 * pad - r) // stride + 1
