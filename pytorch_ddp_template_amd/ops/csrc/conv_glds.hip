@@ -51,8 +51,8 @@ struct Geom {
   int lgSt = 0;
 };
 
-template <int BM, int BN, bool STATS>
-__global__ __launch_bounds__(THREADS, 1) void conv_fwd_glds_kernel(
+template <int BM, int BN, bool STATS, int MINB = 1>
+__global__ __launch_bounds__(THREADS, MINB) void conv_fwd_glds_kernel(
     const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
     __hip_bfloat16* __restrict__ y, const __hip_bfloat16* __restrict__ zpad,
     int M, int N /*Kout*/, int K /*RSC*/, Geom g,
@@ -287,21 +287,30 @@ bool conv2d_fwd_glds_ex(const torch::Tensor& x, const torch::Tensor& w,
   const auto* wp = reinterpret_cast<const __hip_bfloat16*>(w.data_ptr());
   auto* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr());
   const auto* zpp = reinterpret_cast<const __hip_bfloat16*>(zp.data_ptr());
-#define LAUNCH_CG(BMv, BNv)                                                  \
+#define LAUNCH_CG(BMv, BNv, MB)                                              \
   do {                                                                       \
     if (stats_ws)                                                            \
-      hipLaunchKernelGGL((cg::conv_fwd_glds_kernel<BMv, BNv, true>), grid,   \
-                         dim3(cg::THREADS), 0, stream, xp, wp, yp, zpp,      \
-                         (int)M, Kout, K, g, stats_ws, ws_nblocks);          \
+      hipLaunchKernelGGL((cg::conv_fwd_glds_kernel<BMv, BNv, true, MB>),     \
+                         grid, dim3(cg::THREADS), 0, stream, xp, wp, yp,     \
+                         zpp, (int)M, Kout, K, g, stats_ws, ws_nblocks);     \
     else                                                                     \
-      hipLaunchKernelGGL((cg::conv_fwd_glds_kernel<BMv, BNv, false>), grid,  \
-                         dim3(cg::THREADS), 0, stream, xp, wp, yp, zpp,      \
-                         (int)M, Kout, K, g, nullptr, 0);                    \
+      hipLaunchKernelGGL((cg::conv_fwd_glds_kernel<BMv, BNv, false, MB>),    \
+                         grid, dim3(cg::THREADS), 0, stream, xp, wp, yp,     \
+                         zpp, (int)M, Kout, K, g, nullptr, 0);               \
   } while (0)
-  if (BN == 256) LAUNCH_CG(256, 256);
-  else if (BN == 128) LAUNCH_CG(256, 128);
-  else if (BM == 512) LAUNCH_CG(512, 64);
-  else LAUNCH_CG(256, 64);
+  // Kout=64: A/B between one 512x64 block/CU and two co-resident 256x64
+  // blocks (80 KB LDS each — exactly 2/CU; better glds latency hiding)
+  static const char* e_l1 = getenv("PDT_CG_N64");
+  const bool n64_2b = !(e_l1 && e_l1[0] == '5');  // "5" -> 512x64 route
+  if (BN == 64 && n64_2b && M % 256 == 0) {
+    dim3 grid64(Kout / 64, (unsigned)(M / 256));
+    grid = grid64;
+    BM = 256;
+  }
+  if (BN == 256) LAUNCH_CG(256, 256, 1);
+  else if (BN == 128) LAUNCH_CG(256, 128, 1);
+  else if (BM == 512) LAUNCH_CG(512, 64, 1);
+  else LAUNCH_CG(256, 64, 2);
 #undef LAUNCH_CG
   return true;
 }
