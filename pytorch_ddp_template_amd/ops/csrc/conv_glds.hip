@@ -272,7 +272,9 @@ bool conv2d_fwd_glds_ex(const torch::Tensor& x, const torch::Tensor& w,
   const int lgC = pow2l(C);
   const long long M = (long long)N_ * HO * WO;
   const int K = R * S * C;
+  static const char* e_bn = getenv("PDT_CG_BN");
   int BN = Kout % 256 == 0 ? 256 : Kout % 128 == 0 ? 128 : 64;
+  if (e_bn && e_bn[0] == '6') BN = 64;  // force the 2-block 64-wide route
   if (Kout % 64) return false;
   int BM = BN == 64 ? 512 : 256;
   if (M % BM) {
