@@ -300,6 +300,70 @@ class LossScaler:
         return False
 
 
+class GraphStep:
+    """hipGraph-captured training step for the ws=1 CLI engine.
+
+    Same recipe the flagship bench uses (bench.py --graph): three eager
+    warmup iterations on a side stream (so AccumulateGrad stream metadata
+    matches the capture stream), then one capture of
+    fwd+loss+bwd+clip+fused-SGD+in-place-zero with pointer-stable
+    grads/optimizer state, replayed per step with a D2D input refresh and
+    the LR read from a device scalar updated from the LIVE schedule.
+    Removes the per-kernel launch tail; no work is skipped."""
+
+    WARMUP = 3
+
+    def __init__(self, model, criterion, optimizer, max_grad_norm):
+        self.model, self.criterion = model, criterion
+        self.optimizer, self.max_grad_norm = optimizer, max_grad_norm
+        self.side = torch.cuda.Stream()
+        self.graph = None
+        self.loss_st = None
+        self.x_st = self.y_st = None
+        self.lr_dev = None
+        self._warm = 0
+
+    def _inner(self):
+        out = self.model(self.x_st)
+        loss = self.criterion(out, self.y_st)
+        loss.backward()
+        clip_grad_norm_(list(self.model.parameters()), self.max_grad_norm)
+        self.optimizer.step(lr_tensor=self.lr_dev)
+        self.model.zero_grad(set_to_none=False)
+        return loss
+
+    @property
+    def active(self):
+        return self.graph is not None
+
+    def step(self, x, y):
+        """Run one full optimizer step (captured once ready); returns the
+        loss tensor (device scalar, no host sync)."""
+        if self.x_st is None:
+            self.x_st, self.y_st = x.clone(), y.clone()
+            self.lr_dev = torch.zeros((), dtype=torch.float32, device=x.device)
+        else:
+            self.x_st.copy_(x)
+            self.y_st.copy_(y)
+        self.lr_dev.fill_(self.optimizer.param_groups[0]["lr"])
+        if self.graph is None:
+            if self._warm < self.WARMUP:
+                self.side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(self.side):
+                    loss = self._inner()
+                torch.cuda.current_stream().wait_stream(self.side)
+                self._warm += 1
+                return loss.detach()
+            torch.cuda.synchronize()
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph, stream=self.side):
+                self.loss_st = self._inner()
+            # capture RECORDS the work without executing it: replay now so
+            # this batch still trains
+        self.graph.replay()
+        return self.loss_st.detach()
+
+
 def _criterion_for(args):
     if args.model in ("foo", "foomodel", "mlp"):
         return MSELoss()  # reference ddp.py:164
@@ -498,6 +562,24 @@ def train(args, model):
         LossScaler(args.loss_scale) if (args.fp16 and not args.bf16) else None
     )
 
+    # hipGraph-captured step (opt-in): ws=1, GPU, no accumulation, no fp16
+    # scaler (the distributed reducer's in-backward RCCL launches and the
+    # scaler's control flow stay eager)
+    graph_step = None
+    if (
+        args.hip_graph
+        and args.device.type == "cuda"
+        and args.local_rank == -1
+        and args.gradient_accumulation_steps == 1
+        and scaler is None
+    ):
+        graph_step = GraphStep(model, criterion, optimizer, args.max_grad_norm)
+    elif args.hip_graph and is_main_process():
+        logger.warning(
+            "--hip_graph ignored (needs single-process CUDA, accumulation=1,"
+            " no fp16 scaler)"
+        )
+
     model.train()
     model.zero_grad()
     device_loss = torch.zeros((), device=args.device)
@@ -559,7 +641,14 @@ def train(args, model):
                 if (not accum_boundary and isinstance(model, DistributedModel))
                 else _nullcontext()
             )
-            with sync_ctx:
+            if graph_step is not None:
+                # captured step: fwd+bwd+clip+opt+zero in one replay
+                device_loss += graph_step.step(x, y)
+                steps_since_log += 1
+                scheduler.step()
+                global_step += 1
+            else:
+              with sync_ctx:
                 outputs = model(x)
                 loss = criterion(outputs, y)
                 if args.gradient_accumulation_steps > 1:
@@ -568,10 +657,10 @@ def train(args, model):
                     (loss * scaler.scale).backward()
                 else:
                     loss.backward()
-            device_loss += loss.detach()
-            steps_since_log += 1
+              device_loss += loss.detach()
+              steps_since_log += 1
 
-            if accum_boundary:
+            if accum_boundary and graph_step is None:
                 if isinstance(model, DistributedModel):
                     model.finish_gradient_sync()
                 if scaler is not None:
@@ -602,6 +691,8 @@ def train(args, model):
                 scheduler.step()
                 model.zero_grad()
                 global_step += 1
+
+            if accum_boundary:
 
                 if (
                     is_main_process()
@@ -722,6 +813,10 @@ def build_parser():
                         help="run evaluate() every N optimizer steps (0 = off; "
                              "the reference's evaluate was an empty stub)")
     parser.add_argument("--eval_max_batches", type=int, default=None)
+    parser.add_argument("--hip_graph", action="store_true",
+                        help="capture the ws=1 training step in a hipGraph "
+                             "(fwd+bwd+clip+fused SGD replayed per step; "
+                             "live LR via device scalar)")
     parser.add_argument("--prefetch", dest="prefetch", action="store_true",
                         default=True,
                         help="double-buffered pinned-host H2D prefetch on a "

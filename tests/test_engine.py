@@ -242,3 +242,17 @@ def test_evaluate_partial_batch_denominator(tmp_path):
     res = evaluate(args, model)
     assert "eval_acc" in res
     assert 0.0 <= res["eval_acc"] <= 1.0
+
+
+def test_hip_graph_flag_ignored_on_cpu(tmp_path):
+    """--hip_graph on CPU must fall back to the eager path (warn, not
+    crash)."""
+    from pytorch_ddp_template_amd.ddp import main
+
+    main([
+        "--model", "foo", "--dataset_size", "64",
+        "--per_gpu_train_batch_size", "16", "--max_steps", "2",
+        "--no_cuda", "--hip_graph", "--output_dir", str(tmp_path / "o"),
+        "--num_workers", "0", "--no_tensorboard", "--no_progress_bar",
+        "--logging_steps", "1", "--save_steps", "0",
+    ])

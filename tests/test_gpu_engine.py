@@ -37,3 +37,27 @@ def test_engine_bf16_e2e(tmp_path):
 def test_engine_fp16_scaler_e2e(tmp_path):
     out = _run(tmp_path, ["--fp16", "--loss_scale", "0"])  # dynamic scale
     assert (out / "checkpoint-2" / "model.bin").exists()
+
+
+def test_engine_hip_graph_e2e(tmp_path):
+    """--hip_graph: the ws=1 train loop's captured step (GraphStep) must
+    train through warmup + capture + replays, keep a sane loss, and still
+    checkpoint.  Weights must actually CHANGE across replays (a frozen
+    capture — e.g. stale plan pointers — would leave them fixed)."""
+    from pytorch_ddp_template_amd.models import build_model
+
+    out = _run(tmp_path, ["--bf16", "--hip_graph", "--max_steps", "8",
+                          "--save_steps", "8", "--learning_rate", "0.05",
+                          "--momentum", "0.9"])
+    ck = out / "checkpoint-8"
+    assert (ck / "model.bin").exists()
+    sd = torch.load(ck / "model.bin", map_location="cpu", weights_only=True)
+    init = build_model("resnet18", 10).to(torch.bfloat16).state_dict()
+    moved = sum(
+        1 for k in sd
+        if k in init and sd[k].shape == init[k].shape
+        and not torch.equal(sd[k], init[k])
+    )
+    assert moved > 10, f"only {moved} tensors changed — frozen capture?"
+    for v in sd.values():
+        assert torch.isfinite(v.float()).all()
