@@ -1687,8 +1687,14 @@ namespace g16 {
 // (E = G*(L+2) <= 48 rows, image padded to 64 so each is exactly 2 glds).
 // A fragment m-window of 4 always sits inside one segment (L >= 4, windows
 // 4-aligned); the second tr read's row jump is L>=8 ? +4 : +6 (uniform).
-template <typename T16>
-__global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
+// TSPLIT=3: the 9-tap accumulator set is split by kernel ROW across a
+// 12-wave block (wave = rt * 4 + quadrant) — 48 acc VGPRs per wave instead
+// of 144 lifts occupancy 2 -> 3 waves/SIMD, and each wave reads only its
+// own rt's x tiles (per-CU tr-read traffic -40%).  dw writes were already
+// atomicAdd, so the rt-partials need no extra reduction.
+template <typename T16, int TSPLIT = 1>
+__global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
+                             TSPLIT == 3 ? 1 : 2) void gemm_wgrad_seg_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
     float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
     int I /*Kout*/, int J /*Cin*/, long long ldc, ConvMeta cm, int lgWO,
@@ -1731,7 +1737,9 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
+  const int w4 = TSPLIT == 3 ? (wave & 3) : wave;
+  const int rtw = TSPLIT == 3 ? (wave >> 2) : 0;  // this wave's kernel row
+  const int wm = (w4 >> 1) * 32, wn = (w4 & 1) * 32;
   const int sm = lane >> 1;
   const int sh8 = (lane & 1) * 8;
   const int L = min(32, 1 << lgWO);  // seg length (pow2)
@@ -1740,13 +1748,13 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
   const int E = segs * (L + 2);
   const int rcpL2 = 65536 / (L + 2) + 1;  // magic recip (q <= 63)
 
-  f32x16 acc[9] = {};
+  f32x16 acc[TSPLIT == 3 ? 3 : 9] = {};
 
   // ---- stage chunk: dy (4 glds) + 3 x row-tiles (8 glds each) ----
   auto stage = [&](int buf, int ch) {
     const int m0 = ch * BMC;
     T16* base = lds + buf * (TILE_A + 3 * TILE_X);
-    for (int u = wave; u < 4 + 24; u += 4) {
+    for (int u = wave; u < 4 + 24; u += (TSPLIT == 3 ? 12 : 4)) {
       const T16* src = zpad;
       T16* dst;
       if (u < 4) {  // dy image u
@@ -1843,8 +1851,10 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
       // async LDS write inside the in-flight MFMA's source-read window —
       // WAR hazard, caught by the cold-launch stress test)
       v4s pb[3][2];
+      constexpr int NPAIR = TSPLIT == 3 ? 6 : 18;
       auto issueB = [&](int slot, int idx) {
-        const int rt = idx / 6, rem = idx % 6, s2 = rem >> 1, kh = rem & 1;
+        const int rt = TSPLIT == 3 ? rtw : idx / 6;
+        const int rem = idx % 6, s2 = rem >> 1, kh = rem & 1;
         const T16* timg =
             base + TILE_A + rt * TILE_X + (img_sel + (wn >> 4)) * IMG_X;
         const unsigned b0 =
@@ -1857,13 +1867,13 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
       };
       issueB(0, 0);
 #pragma unroll
-      for (int idx = 0; idx < 18; ++idx) {
+      for (int idx = 0; idx < NPAIR; ++idx) {
         const int slot = idx % 3;
         // the wait carries the pair's registers as "+v" operands: the
         // compiler must order the copies/MFMA that consume them AFTER the
         // wait (a bare asm waitcnt does not stop it hoisting the v_movs
         // above the wait — stale-register NaNs at some shapes)
-        if (idx + 1 < 18) {
+        if (idx + 1 < NPAIR) {
           issueB((idx + 1) % 3, idx + 1);
           asm volatile("s_waitcnt lgkmcnt(2)"
                        : "+v"(pb[slot][0]), "+v"(pb[slot][1]));
@@ -1871,7 +1881,8 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
           asm volatile("s_waitcnt lgkmcnt(0)"
                        : "+v"(pb[slot][0]), "+v"(pb[slot][1]));
         }
-        const int rt = idx / 6, rem = idx % 6, s2 = rem >> 1, kh = rem & 1;
+        const int rt = TSPLIT == 3 ? 0 : idx / 6;
+        const int rem = idx % 6, s2 = rem >> 1, kh = rem & 1;
         vec16 bf;
         reinterpret_cast<v4s*>(&bf)[0] = pb[slot][0];
         reinterpret_cast<v4s*>(&bf)[1] = pb[slot][1];
@@ -1881,9 +1892,10 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
     if (more) __syncthreads();
   }
 
-  // ---- writeback (32x32x16 C/D layout) ----
+  // ---- writeback (32x32x16 C/D layout; TSPLIT: only this wave's taps) ----
 #pragma unroll
-  for (int tap = 0; tap < 9; ++tap) {
+  for (int t3 = 0; t3 < (TSPLIT == 3 ? 3 : 9); ++t3) {
+    const int tap = TSPLIT == 3 ? rtw * 3 + t3 : t3;
     const long long coff = (long long)tap * Cin;
     const int col = j0 + wn + (lane & 31);
     if (col >= Cin) continue;
@@ -1891,7 +1903,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_wgrad_seg_kernel(
     for (int reg = 0; reg < 16; ++reg) {
       const int row = i0 + wm + (reg & 3) + 8 * (reg >> 2) + 4 * ks;
       if (row >= I) continue;
-      atomicAdd(&dw[(long long)row * ldc + coff + col], acc[tap][reg]);
+      atomicAdd(&dw[(long long)row * ldc + coff + col], acc[t3][reg]);
     }
   }
 }
@@ -1937,14 +1949,30 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
     using t16 = decltype(tag);
     if (R == 3 && S == 3 && stride == 1 && lgWO >= 2 && lgHO >= 2) {
       // segment-staged fast path (stride-1, pow2 spatial): 28 glds/chunk
-      // instead of 40, x gathered once per kernel ROW not per tap
-      hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16>), grid,
-                         dim3(g16::THREADS), 0, stream,
-                         reinterpret_cast<const t16*>(dy.data_ptr()),
-                         reinterpret_cast<const t16*>(x.data_ptr()),
-                         dw.data_ptr<float>(),
-                         reinterpret_cast<const t16*>(zp.data_ptr()), M, Kout,
-                         Cin, ldc, cm, lgWO, lgHO, (int)grid.z);
+      // instead of 40, x gathered once per kernel ROW not per tap.
+      // Measured: the 12-wave rt-split variant (92 VGPR, 3 waves/SIMD,
+      // -40% per-CU tr traffic) LOSES to the 4-wave form (19.5 vs 18.4 ms
+      // summed wgrad; bench 113.8k vs 115.7k) — one 12-wave block
+      // locksteps every wave at the per-chunk barrier, while two 4-wave
+      // blocks interleave around each other's glds drains.  Kept behind
+      // PDT_WGRAD_TSPLIT=3 as a documented negative result.
+      static const char* e_ts = getenv("PDT_WGRAD_TSPLIT");
+      if (e_ts && e_ts[0] == '3')
+        hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16, 3>), grid,
+                           dim3(768), 0, stream,
+                           reinterpret_cast<const t16*>(dy.data_ptr()),
+                           reinterpret_cast<const t16*>(x.data_ptr()),
+                           dw.data_ptr<float>(),
+                           reinterpret_cast<const t16*>(zp.data_ptr()), M,
+                           Kout, Cin, ldc, cm, lgWO, lgHO, (int)grid.z);
+      else
+        hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16>), grid,
+                           dim3(g16::THREADS), 0, stream,
+                           reinterpret_cast<const t16*>(dy.data_ptr()),
+                           reinterpret_cast<const t16*>(x.data_ptr()),
+                           dw.data_ptr<float>(),
+                           reinterpret_cast<const t16*>(zp.data_ptr()), M,
+                           Kout, Cin, ldc, cm, lgWO, lgHO, (int)grid.z);
     } else if (R == 3 && S == 3) {
       hipLaunchKernelGGL((g16::gemm_wgrad_tr_kernel<t16, 9>), grid,
                          dim3(g16::THREADS), 0, stream,
