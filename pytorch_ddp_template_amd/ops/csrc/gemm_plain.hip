@@ -691,6 +691,133 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_nt_plain256_m32_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Split-K variant for TRANSPOSED weight-gradient GEMMs.
+//
+// dw[N][K] = dy^T @ x is a TN GEMM (both operands M-major) whose transpose
+// reads cap the TN kernels near ~450-530 TF.  Transposing BOTH operands
+// once (dyT [N][M], xT [K][M] — two HBM passes each over tensors the wgrad
+// reads anyway) turns it into THIS kernel: a plain NT GEMM with the huge M
+// as the reduction dim — the glds structure's best regime.  The tiny
+// output (e.g. 768x2304) parallelizes over grid.z K-chunks; each block
+// accumulates its chunk and atomicAdds fp32 fragments into the pre-zeroed
+// output (atomics land on a ~few-MB tensor: L2-resident).
+__global__ __launch_bounds__(THREADS, 1) void gemm_nt_plain256_splitk_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ B,
+    float* __restrict__ C, int M, int N, int K, int ckt) {
+  __shared__ __hip_bfloat16 smem[NT_BUF * 2 * BM * BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  long long nwg = (long long)gridDim.x * gridDim.y * gridDim.z;
+  long long bid = ((long long)blockIdx.z * gridDim.y + blockIdx.y) *
+                      gridDim.x + blockIdx.x;
+  {
+    const long long q = nwg >> 3, r = nwg & 7;
+    const int xcd = (int)(bid & 7);
+    const long long o = bid >> 3;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + o;
+  }
+  const int n0 = (int)(bid % gridDim.x) * BN;
+  long long t = bid / gridDim.x;
+  const int m0 = (int)(t % gridDim.y) * BM;
+  const int zi = (int)(t / gridDim.y);
+  const int NTt = K / BK;
+  const int kt0 = zi * ckt, kt1 = min(NTt, kt0 + ckt);
+  if (kt0 >= kt1) return;
+
+  const int srow = (tid >> 3) & 7;
+  const int schunk = lane & 7;
+  auto stage = [&](int buf, int kt) {
+    __hip_bfloat16* sa = smem + buf * (2 * BM * BK);
+    __hip_bfloat16* sb = sa + BM * BK;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r * 64 + wave * 8 + srow;
+      const long long kk = (long long)kt * BK + kswz(row, schunk * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              A + (long long)(m0 + row) * M /*lda = reduction dim*/ + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              sa + (r * 64 + wave * 8) * BK),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r * 64 + wave * 8 + srow;
+      const long long kk = (long long)kt * BK + kswz(row, schunk * 16) / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              B + (long long)(n0 + row) * M + kk),
+          (__attribute__((address_space(3))) unsigned int*)(
+              sb + (r * 64 + wave * 8) * BK),
+          16, 0, 0);
+    }
+  };
+
+  const int wm = (wave >> 2) * 128;
+  const int wn = (wave & 3) * 64;
+  const int frow = lane & 15;
+  const int fkb = (lane >> 4) * 16;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  stage(0, kt0);
+  __syncthreads();
+  int cur = 0;
+  for (int kt = kt0; kt < kt1; ++kt) {
+    if (kt + 1 < kt1) stage(cur ^ 1, kt + 1);
+    const __hip_bfloat16* sa = smem + cur * (2 * BM * BK);
+    const __hip_bfloat16* sb = sa + BM * BK;
+    bf16x8 bf[4][2];
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wn + ni * 16 + frow;
+        bf[ni][ks] = *(const bf16x8*)((const char*)(sb + row * BK) +
+                                      kswz(row, ks * 64 + fkb));
+      }
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi) {
+      bf16x8 af[2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wm + mi * 16 + frow;
+        af[ks] = *(const bf16x8*)((const char*)(sa + row * BK) +
+                                  kswz(row, ks * 64 + fkb));
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        acc[mi][ni] = mfma16(af[0], bf[ni][0], acc[mi][ni]);
+        acc[mi][ni] = mfma16(af[1], bf[ni][1], acc[mi][ni]);
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // fp32 fragment atomics (output is a few MB: L2-resident)
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = n0 + wn + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
+        atomicAdd(&C[(long long)row * N + col], acc[mi][ni][reg]);
+      }
+    }
+}
+
 }  // namespace gp
 
 // Host entry: true when the shape is handled (M%256, N%256, K%64, bf16).
@@ -797,4 +924,34 @@ bool gemm_nt_plain256(const torch::Tensor& A, const torch::Tensor& B,
   }
 #undef LAUNCH_GP
   return true;
+}
+
+// dw[I][J] (fp32, pre-zeroed by caller semantics: allocated here) from
+// TRANSPOSED operands: A=[I][M] (dy^T), B=[J][M] (x^T), both M-contiguous.
+torch::Tensor gemm_nt_splitk_f32(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.dim() == 2 && B.dim() == 2);
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              B.scalar_type() == torch::kBFloat16);
+  const int I = (int)A.size(0);
+  const long long M = A.size(1);
+  const int J = (int)B.size(0);
+  TORCH_CHECK(B.size(1) == M, "reduction dim mismatch");
+  TORCH_CHECK(I % gp::BM == 0 && J % gp::BN == 0 && M % gp::BK == 0,
+              "gemm_nt_splitk_f32 needs I%256, J%256, M%64");
+  auto C = torch::zeros({I, J}, A.options().dtype(torch::kFloat32));
+  const int NT = (int)(M / gp::BK);
+  const int tiles = (I / gp::BM) * (J / gp::BN);
+  // target ~1536 blocks (6/CU): enough z to fill the chip on tiny outputs
+  int z = std::max(1, std::min(NT, (1536 + tiles - 1) / tiles));
+  int ckt = (NT + z - 1) / z;
+  z = (NT + ckt - 1) / ckt;
+  auto stream = c10::hip::getCurrentHIPStream();
+  dim3 grid(J / gp::BN, I / gp::BM, (unsigned)z);
+  hipLaunchKernelGGL(gp::gemm_nt_plain256_splitk_kernel, grid,
+                     dim3(gp::THREADS), 0, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),
+                     C.data_ptr<float>(), (int)M, J, (int)M, ckt);
+  return C;
 }

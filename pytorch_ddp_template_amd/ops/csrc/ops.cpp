@@ -34,6 +34,9 @@ torch::Tensor im2col(torch::Tensor x, int64_t R, int64_t S, int64_t stride,
 // gemm_bf16.hip
 torch::Tensor bmm_nt_bf16(torch::Tensor A, torch::Tensor B,
                           c10::optional<torch::Tensor> bias, bool relu);
+// transposed-operand split-K NT (gemm_plain.hip): dw = A[I,M] @ B[J,M]^T,
+// fp32 out, for linear/1x1-conv weight grads routed via explicit transposes
+torch::Tensor gemm_nt_splitk_f32(torch::Tensor A, torch::Tensor B);
 torch::Tensor bmm_tn_bf16(torch::Tensor A, torch::Tensor B);
 std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B);
 torch::Tensor conv2d_fwd_bf16(torch::Tensor x, torch::Tensor w,
@@ -274,6 +277,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias") = c10::nullopt, py::arg("relu") = false,
         py::arg("out_f32") = false);
   m.def("gemm_tn", &gemm_tn);
+  m.def("gemm_nt_splitk_f32", &gemm_nt_splitk_f32);
   m.def("gemm_tn_bias", [](torch::Tensor A, torch::Tensor B) {
     TORCH_CHECK(A.scalar_type() == torch::kBFloat16 ||
                 A.scalar_type() == torch::kHalf,

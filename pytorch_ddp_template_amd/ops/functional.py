@@ -30,6 +30,8 @@ ResNet/ViT benchmark configs.
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -38,6 +40,26 @@ from .native import native, use_native
 # ---------------------------------------------------------------------------
 # Linear (+ optional fused ReLU epilogue)
 # ---------------------------------------------------------------------------
+
+
+# Transposed-operand wgrad route: measured SLOWER than the TN kernels on
+# the ViT shapes (NT-T 271-338 TF vs TN 425-496; ViT bench 3,116 vs 3,987
+# samples/s) — the two explicit transposes plus the split-K z-chunking
+# (each block runs only ~14 K-tiles, back in the pipeline-fill-dominated
+# regime) cost more than the transpose-free staging saves.  Kept behind
+# PDT_WGRAD_NT=1 as a documented negative result.
+_WGRAD_NT = os.environ.get("PDT_WGRAD_NT", "0") == "1"
+
+
+def _wgrad_via_nt(dys, xs) -> bool:
+    return (
+        _WGRAD_NT
+        and dys.dtype == torch.bfloat16
+        and dys.shape[0] >= 4096
+        and dys.shape[0] % 64 == 0
+        and dys.shape[1] % 256 == 0
+        and xs.shape[1] % 256 == 0
+    )
 
 
 def _plain_gemm_to_blas(m: int, k: int, n: int) -> bool:
@@ -96,7 +118,18 @@ class _LinearFn(torch.autograd.Function):
             else:
                 wt = ext.transpose2d(w)  # (K, N)
                 dx = ext.gemm_nt(dys, wt, None, False, False)
-            if ctx.has_bias and dys.dtype in (torch.bfloat16, torch.float16):
+            if _wgrad_via_nt(dys, xs):
+                # transposed-operand route (gemm_plain.hip split-K NT):
+                # transposing dy and x once (two HBM passes over tensors the
+                # wgrad reads anyway) turns the TN wgrad into a plain NT
+                # GEMM with the huge M as reduction — the glds kernel's
+                # best regime (TN kernels cap at ~450-530 TF on these
+                # shapes; ViT wgrad was 29% of the step)
+                dyT = ext.transpose2d(dys)
+                xT = ext.transpose2d(xs)
+                dw = ext.gemm_nt_splitk_f32(dyT, xT).to(w.dtype)
+                db = ext.col_sum(dys).to(w.dtype) if ctx.has_bias else None
+            elif ctx.has_bias and dys.dtype in (torch.bfloat16, torch.float16):
                 # bias grad rides inside the TN GEMM (dy is staged anyway)
                 dwf, dbf = ext.gemm_tn_bias(dys, xs)
                 dw = dwf.to(w.dtype)

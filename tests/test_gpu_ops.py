@@ -839,3 +839,15 @@ def test_sgd_lr_from_device_tensor():
                  lr_dev)
     torch.cuda.synchronize()
     torch.testing.assert_close(p, p0 - 0.25, rtol=1e-6, atol=1e-6)
+
+
+def test_gemm_nt_splitk_f32():
+    """Transposed-operand split-K NT (the linear/1x1 wgrad route): fp32
+    atomic accumulation over grid.z K-chunks must match torch fp32."""
+    for i, j, m in [(256, 256, 4096), (768, 256, 8192), (256, 512, 12608)]:
+        a = t32(i, m, seed=90).to(torch.bfloat16).to(DEV)
+        b = t32(j, m, seed=91).to(torch.bfloat16).to(DEV)
+        out = EXT.gemm_nt_splitk_f32(a, b)
+        ref = a.float() @ b.float().t()
+        err = (out - ref).abs().max() / ref.abs().max().clamp(min=1.0)
+        assert float(err) < 0.03, f"{i}x{j}x{m}: rel err {float(err)}"
