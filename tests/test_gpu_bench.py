@@ -33,3 +33,27 @@ def test_bench_single_gpu_json():
     assert d["higher_is_better"] is True
     assert d["config"]["model"] == "resnet18-cifar"
     assert d["loss"] == d["loss"]  # not NaN
+
+
+def _bench_loss(extra):
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"), "--gpus", "1",
+         "--steps", "6", "--warmup", "3", "--batch", "512"] + extra,
+        capture_output=True, text=True, timeout=600, cwd=ROOT,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    return json.loads(line)["loss"]
+
+
+def test_bench_graph_matches_eager_semantics():
+    """The hipGraph-captured step must run the SAME training math as the
+    eager step: identical seeds/data give closely matching final losses
+    (loose tolerance: the wgrad fp32 atomics make reduction order, and
+    hence bf16 rounding, nondeterministic across any two runs).  This is
+    the regression guard for capture bugs that silently freeze weights or
+    skip work — a frozen capture pins loss near ln(10)=2.30 while eager
+    descends."""
+    lg = _bench_loss([])           # graph default at N=1
+    le = _bench_loss(["--no-graph"])
+    assert abs(lg - le) < 0.35, (lg, le)
