@@ -140,6 +140,16 @@ def main():
                 inner_step()
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
+        # The pre-capture warmups ran at the schedule's INITIAL lr (0 with
+        # linear warmup), so parameters are untouched — but their backward
+        # passes accumulated MOMENTUM.  Reset optimizer state so replays
+        # start from the same state an eager run starts from (without this
+        # the first real steps carry a ~3x momentum kick and the graph and
+        # eager trajectories diverge systematically —
+        # tests/test_gpu_bench.py::test_bench_graph_matches_eager_semantics).
+        for st in opt.state.values():
+            if "momentum_buffer" in st:
+                st["momentum_buffer"].zero_()
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph, stream=side):
             loss_st = inner_step()

@@ -35,8 +35,11 @@ class MultiHeadAttention(nn.Module):
         N, S, D = x.shape
         h, dh = self.heads, D // self.heads
         qkv = self.qkv(x)  # (N, S, 3D)
+        # the flash forward/backward tile K/V, so S is uncapped (the old
+        # S<=224 limit belonged to the P-materializing two-pass kernel;
+        # fused-vs-composed grads are tested at S=320)
         if (
-            x.is_cuda and dh == 64 and S <= 224
+            x.is_cuda and dh == 64
             and x.dtype in (torch.bfloat16, torch.float16)
             and native_available()
         ):

@@ -459,11 +459,12 @@ def test_attn_fwd_fused_matches_composed():
     ).max() < 2e-2
 
 
-def test_attention_qkv_grads_match_composed():
+@pytest.mark.parametrize("S", [197, 320])  # 320: beyond the old S<=224 cap
+def test_attention_qkv_grads_match_composed(S):
     from pytorch_ddp_template_amd.ops.functional import attention, attention_qkv
 
     torch.manual_seed(4)
-    N, S, h, dh = 2, 197, 4, 64
+    N, h, dh = 2, 4, 64
     qkv = (torch.randn(N, S, 3 * h * dh) * 0.5).to(torch.bfloat16).to(DEV)
     a = qkv.clone().requires_grad_(True)
     b = qkv.clone().requires_grad_(True)
