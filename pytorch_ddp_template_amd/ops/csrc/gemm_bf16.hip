@@ -975,8 +975,9 @@ void gemm_wgrad_tr_kernel(
 // issue/latency-bound at ~425 TF.  128-B rows cut the request count 4x.
 // Everything else (tr-read fragments, f32 atomic writeback, zsplit, XCD
 // remap, optional fused bias grad) matches the kernel above.
-template <typename T16, int TI = 4, int TJ = 4, int WI = 2, int WJ = 2>
-__global__ __launch_bounds__(WI * WJ * 64, 2) void gemm_tn_plain_kernel(
+template <typename T16, int TI = 4, int TJ = 4, int WI = 2, int WJ = 2,
+          int MINB = 3>
+__global__ __launch_bounds__(WI * WJ * 64, MINB) void gemm_tn_plain_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
     float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot, int I,
     int J, long long ldc, long long sA, long long sB, long long sC,
@@ -1627,10 +1628,17 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
   g16::ConvMeta cm{};
   auto& zp = zero_page(A.device(), A.scalar_type());
   static const char* tn_tile2 = getenv("PDT_TN_TILE");
+  // 256x256 tiles on the 128-B-row tn_plain kernel (PDT_TN_TILE=3):
+  // measured SLOWER than 128 tiles (qkv 380 vs 414 TF, proj 219 vs 352 —
+  // the drop to one 8-wave block/CU costs more than the halved re-read
+  // traffic, which the per-XCD L2s largely absorb).  Kept as a documented
+  // negative result; '2' = the older xwide TR route (also slower).
   const bool xwide = (tn_tile2 && tn_tile2[0] == '2') && I >= 256 &&
                      J >= 256 && M >= 4096;
+  const bool big256 = (tn_tile2 && tn_tile2[0] == '3') && I >= 256 &&
+                      J >= 256 && M >= 4096;
   const bool wide = I >= 128 && J >= 128;
-  int tiles = xwide ? ((J + 255) / 256) * ((I + 255) / 256)
+  int tiles = (xwide || big256) ? ((J + 255) / 256) * ((I + 255) / 256)
               : wide ? ((J + 127) / 128) * ((I + 127) / 128)
                      : ((J + 63) / 64) * ((I + 63) / 64);
   int z = std::max(
@@ -1638,6 +1646,17 @@ std::vector<torch::Tensor> bmm_tn_bias_bf16(torch::Tensor A, torch::Tensor B) {
                   ((xwide ? 256 : 1024) + tiles - 1) / std::max(1, tiles)));
   auto run = [&](auto tag) {
     using t16 = decltype(tag);
+    if (big256) {
+      dim3 grid((J + 255) / 256, (I + 255) / 256, (unsigned)z);
+      hipLaunchKernelGGL(
+          (g16::gemm_tn_plain_kernel<t16, 8, 8, 2, 4, 1>), grid,
+          dim3(512), 0, stream,
+          reinterpret_cast<const t16*>(A.data_ptr()),
+          reinterpret_cast<const t16*>(B.data_ptr()), C.data_ptr<float>(),
+          reinterpret_cast<const t16*>(zp.data_ptr()), M, I, J, J, 0, 0,
+          0, z, db.data_ptr<float>());
+      return;
+    }
     if (xwide) {
       dim3 grid((J + 255) / 256, (I + 255) / 256, (unsigned)z);
       hipLaunchKernelGGL(
