@@ -82,6 +82,29 @@ class SGD(torch.optim.Optimizer):
         )
         super().__init__(params, defaults)
 
+    def load_state_dict(self, state_dict):
+        """torch's Optimizer.load_state_dict casts floating-point state to
+        each param's dtype — with bf16 params that silently rounds the fp32
+        master weights and momentum through bf16 (destroying the master
+        precision) and then crashes the fused kernel, which requires fp32
+        state.  Restore the fp32 slots from the ORIGINAL (on-disk) tensors
+        after the structural load."""
+        super().load_state_dict(state_dict)
+        id_map = {}
+        for g_saved, g in zip(state_dict["param_groups"], self.param_groups):
+            for pid, p in zip(g_saved["params"], g["params"]):
+                id_map[pid] = p
+        for pid, st_saved in state_dict["state"].items():
+            p = id_map.get(pid)
+            if p is None:
+                continue
+            st = self.state[p]
+            for k in ("momentum_buffer", "master"):
+                if k in st_saved and isinstance(st_saved[k], torch.Tensor):
+                    st[k] = st_saved[k].to(
+                        device=p.device, dtype=torch.float32
+                    )
+
     @torch.no_grad()
     def step(self, closure=None, guard=None, skip_count=None, lr_tensor=None):
         """``guard`` (f32 device scalar, e.g. the pre-clip grad norm): when

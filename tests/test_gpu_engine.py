@@ -61,3 +61,15 @@ def test_engine_hip_graph_e2e(tmp_path):
     assert moved > 10, f"only {moved} tensors changed — frozen capture?"
     for v in sd.values():
         assert torch.isfinite(v.float()).all()
+
+
+def test_engine_resume_steps_after_load(tmp_path):
+    """Resume must TRAIN after loading (regression: torch's state-dtype
+    cast crashed the fused SGD on the first post-resume step with bf16
+    params — the old roundtrip test loaded but never stepped)."""
+    out = _run(tmp_path, ["--bf16", "--max_steps", "4", "--save_steps", "2",
+                          "--momentum", "0.9"])
+    assert (out / "checkpoint-2").exists()
+    # second run resumes from step 2 and must complete steps 3..4
+    _run(tmp_path, ["--bf16", "--max_steps", "4", "--save_steps", "0",
+                    "--momentum", "0.9", "--global-step", "2"])

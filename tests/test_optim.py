@@ -98,3 +98,30 @@ def test_linear_warmup_schedule_shape():
     assert math.isclose(lrs[10], 1.0)
     assert lrs[50] < 1.0
     assert math.isclose(lrs[99], (100 - 99) / 90, rel_tol=1e-6)
+
+
+def test_sgd_state_dict_roundtrip_keeps_fp32_state():
+    """torch's Optimizer.load_state_dict casts state to the param dtype —
+    with bf16 params that rounded the fp32 master/momentum through bf16
+    (and crashed the fused GPU kernel on the first post-resume step).  Our
+    override must restore the fp32 slots bit-exactly from the saved
+    tensors."""
+    import torch
+
+    from pytorch_ddp_template_amd.optim import SGD
+
+    p = torch.nn.Parameter(torch.randn(64).to(torch.bfloat16))
+    opt = SGD([p], lr=0.1, momentum=0.9, master_weights=True)
+    p.grad = torch.randn_like(p)
+    opt.step()
+    sd = opt.state_dict()
+    # fresh optimizer, same param object
+    opt2 = SGD([p], lr=0.1, momentum=0.9, master_weights=True)
+    opt2.load_state_dict(sd)
+    st = opt2.state[p]
+    assert st["momentum_buffer"].dtype == torch.float32
+    assert st["master"].dtype == torch.float32
+    assert torch.equal(st["master"], opt.state[p]["master"])
+    assert torch.equal(st["momentum_buffer"], opt.state[p]["momentum_buffer"])
+    p.grad = torch.randn_like(p)
+    opt2.step()  # must not crash / degrade
