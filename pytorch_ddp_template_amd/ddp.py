@@ -144,7 +144,8 @@ def save_model(model, output_dir):
 
 
 def save_checkpoint(
-    args, model, optimizer, scheduler, global_step, epoch=0, batches_in_epoch=0
+    args, model, optimizer, scheduler, global_step, epoch=0,
+    batches_in_epoch=0, scaler=None
 ):
     """Full checkpoint in the reference layout (ddp.py:255-277), plus
     ``training_state.pt`` (epoch / in-epoch position / RNG states) so a
@@ -164,6 +165,10 @@ def save_checkpoint(
             "random": random.getstate(),
         },
     }
+    if scaler is not None:
+        # fp16 dynamic loss scale: resume at the calibrated scale instead
+        # of re-converging from the default through overflow skips
+        state["scaler_scale"] = float(scaler.scale)
     if torch.cuda.is_available() and args.n_gpu > 0:
         state["rng"]["cuda"] = torch.cuda.get_rng_state_all()
     torch.save(state, os.path.join(output_dir, "training_state.pt"))
@@ -215,14 +220,16 @@ def load_checkpoint(args, model, optimizer=None, scheduler=None, path=None):
     return step
 
 
-def load_training_state(args, path):
-    """Restore epoch / in-epoch position / RNG states saved by
-    save_checkpoint.  Returns (epoch, batches_in_epoch); (0, 0) when the
-    checkpoint predates training_state.pt."""
+def load_training_state(args, path, scaler=None):
+    """Restore epoch / in-epoch position / RNG states (and the fp16 loss
+    scale) saved by save_checkpoint.  Returns (epoch, batches_in_epoch);
+    (0, 0) when the checkpoint predates training_state.pt."""
     f = os.path.join(path, "training_state.pt")
     if not os.path.exists(f):
         return 0, 0
     state = torch.load(f, map_location="cpu", weights_only=False)
+    if scaler is not None and "scaler_scale" in state:
+        scaler.scale = float(state["scaler_scale"])
     rng = state.get("rng", {})
     if "torch" in rng:
         torch.set_rng_state(rng["torch"])
@@ -526,6 +533,12 @@ def train(args, model):
             find_unused_parameters=args.find_unused_parameters,
         )
 
+    # Native fp16 loss scaling (replaces the reference's broken apex branch,
+    # ddp.py:165-181: static scale when --loss_scale > 0, else dynamic).
+    scaler = (
+        LossScaler(args.loss_scale) if (args.fp16 and not args.bf16) else None
+    )
+
     global_step = 0
     resume_epoch, resume_batches = 0, 0
     if args.resume_from or args.global_step:
@@ -540,7 +553,9 @@ def train(args, model):
             path = find_latest_checkpoint(args.output_dir)
         if path is not None and global_step > 0:
             # mid-epoch position + RNG states (true resume, not data replay)
-            resume_epoch, resume_batches = load_training_state(args, path)
+            resume_epoch, resume_batches = load_training_state(
+                args, path, scaler=scaler
+            )
 
     logger.info(
         "***** Running training *****",
@@ -556,11 +571,6 @@ def train(args, model):
         ),
     )
 
-    # Native fp16 loss scaling (replaces the reference's broken apex branch,
-    # ddp.py:165-181: static scale when --loss_scale > 0, else dynamic).
-    scaler = (
-        LossScaler(args.loss_scale) if (args.fp16 and not args.bf16) else None
-    )
 
     # hipGraph-captured step (opt-in): ws=1, GPU, no accumulation, no fp16
     # scaler (the distributed reducer's in-backward RCCL launches and the
@@ -749,7 +759,7 @@ def train(args, model):
                 ):
                     save_checkpoint(
                         args, model, optimizer, scheduler, global_step,
-                        epoch=epoch, batches_in_epoch=step + 1,
+                        epoch=epoch, batches_in_epoch=step + 1, scaler=scaler,
                     )
 
                 if args.max_steps > 0 and global_step >= args.max_steps:

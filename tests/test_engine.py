@@ -256,3 +256,26 @@ def test_hip_graph_flag_ignored_on_cpu(tmp_path):
         "--num_workers", "0", "--no_tensorboard", "--no_progress_bar",
         "--logging_steps", "1", "--save_steps", "0",
     ])
+
+def test_fp16_scaler_scale_survives_checkpoint(tmp_path):
+    """Dynamic loss scale persists through save/resume (a resumed fp16 run
+    otherwise re-converges its scale from the default via overflow skips)."""
+    import torch
+
+    from pytorch_ddp_template_amd.ddp import (
+        LossScaler, load_training_state, save_checkpoint,
+    )
+
+    class A:  # minimal args shim
+        output_dir = str(tmp_path)
+        n_gpu = 0
+
+    m = torch.nn.Linear(4, 4)
+    opt = torch.optim.SGD(m.parameters(), lr=0.1)
+    sch = torch.optim.lr_scheduler.LambdaLR(opt, lambda s: 1.0)
+    sc = LossScaler(0)
+    sc.scale = 1024.0
+    save_checkpoint(A, m, opt, sch, 7, scaler=sc)
+    sc2 = LossScaler(0)
+    load_training_state(A, str(tmp_path / "checkpoint-7"), scaler=sc2)
+    assert sc2.scale == 1024.0
