@@ -266,9 +266,9 @@ def test_fp16_scaler_scale_survives_checkpoint(tmp_path):
         LossScaler, load_training_state, save_checkpoint,
     )
 
-    class A:  # minimal args shim
-        output_dir = str(tmp_path)
-        n_gpu = 0
+    import argparse
+
+    A = argparse.Namespace(output_dir=str(tmp_path), n_gpu=0)
 
     m = torch.nn.Linear(4, 4)
     opt = torch.optim.SGD(m.parameters(), lr=0.1)
