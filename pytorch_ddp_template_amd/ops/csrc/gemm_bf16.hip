@@ -1711,7 +1711,12 @@ namespace g16 {
 // of 144 lifts occupancy 2 -> 3 waves/SIMD, and each wave reads only its
 // own rt's x tiles (per-CU tr-read traffic -40%).  dw writes were already
 // atomicAdd, so the rt-partials need no extra reduction.
-template <typename T16, int TSPLIT = 1>
+// RING (WO==32 only, TSPLIT==1): one chunk = exactly one output row, so
+// chunk ch+1's kernel-row-r x tile IS chunk ch's row r+1 tile — a 4-slot
+// ring restages ONE tile per chunk (12 glds) instead of three (28),
+// except at image boundaries (ho wrap: full 3-tile restage behind an
+// extra barrier, 1/HO of chunks).
+template <typename T16, int TSPLIT = 1, bool RING = false>
 __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
                              TSPLIT == 3 ? 1 : 2) void gemm_wgrad_seg_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
@@ -1725,7 +1730,10 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
   constexpr int TILE_X = 4 * IMG_X;       // 8 KiB per kernel row r
   typedef short v4s __attribute__((ext_vector_type(4)));
   using vec16 = typename M16<T16>::vec;
-  __shared__ __attribute__((aligned(16))) T16 lds[2 * (TILE_A + 3 * TILE_X)];
+  // RING layout: [2 dy bufs][4 x rt-slots] = 40 KB; else the classic
+  // [2][dy + 3 x tiles] = 56 KB
+  __shared__ __attribute__((aligned(16)))
+      T16 lds[RING ? (2 * TILE_A + 4 * TILE_X) : 2 * (TILE_A + 3 * TILE_X)];
 
   // XCD-contiguous remap (private per-XCD L2s — see gemm_wgrad_tr_kernel):
   // blocks sharing a zidx read the same dy/x chunks; keep them on one L2.
@@ -1770,7 +1778,56 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
   f32x16 acc[TSPLIT == 3 ? 3 : 9] = {};
 
   // ---- stage chunk: dy (4 glds) + 3 x row-tiles (8 glds each) ----
+  // RING helpers: dy buffers at lds[0..2*TILE_A), x slots after
+  auto stage_dy_ring = [&](int buf, int ch) {
+    if (!RING) return;
+    const int m0 = ch * BMC;
+    for (int u = wave; u < 4; u += 4) {
+      const T16* src = zpad;
+      const int gm = m0 + sm;
+      const int ii = i0 + u * 16 + sh8;
+      if (gm < Mtot && ii < I) src = dy + (long long)gm * I + ii;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)(
+              lds + buf * TILE_A + u * IMG_A),
+          16, 0, 0);
+    }
+  };
+  auto stage_x_slot = [&](int slot, int ch, int rt) {
+    if (!RING) return;
+    const int m0 = ch * BMC;
+    T16* dst0 = lds + 2 * TILE_A + slot * TILE_X;
+    for (int u = wave & 3; u < 8; u += 4) {
+      const int g2 = (u >> 2) & 1;
+      const int ig = u & 3;
+      const int q = g2 * 32 + sm;
+      const T16* src = zpad;
+      if (q < E) {
+        const int seg = (q * rcpL2) >> 16;
+        const int c = q - seg * (L + 2);
+        const long long m_seg = (long long)m0 + (long long)seg * L;
+        if (m_seg < Mtot) {
+          const int wo0 = (int)(m_seg & ((1 << lgWO) - 1));
+          const long long t = m_seg >> lgWO;
+          const int ho = (int)(t & ((1 << lgHO) - 1));
+          const int n = (int)(t >> lgHO);
+          const int hi = ho - cm.pad + rt;
+          const int wi = wo0 - cm.pad + c;
+          const int jj = j0 + ig * 16 + sh8;
+          if (hi >= 0 && hi < cm.H && wi >= 0 && wi < cm.W && jj < Cin)
+            src = x + (((long long)n * cm.H + hi) * cm.W + wi) * Cin + jj;
+        }
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)(
+              dst0 + ig * IMG_X + g2 * (32 * 16)),
+          16, 0, 0);
+    }
+  };
   auto stage = [&](int buf, int ch) {
+    if (RING) return;
     const int m0 = ch * BMC;
     T16* base = lds + buf * (TILE_A + 3 * TILE_X);
     for (int u = wave; u < 4 + 24; u += (TSPLIT == 3 ? 12 : 4)) {
@@ -1828,14 +1885,35 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
   ((unsigned)(unsigned long long)(__attribute__((            \
       address_space(3))) const T16*)(p))
 
-  stage(0, ch0);
+  int r0 = 0;  // RING rotation base
+  if (RING) {
+    stage_dy_ring(0, ch0);
+    stage_x_slot(0, ch0, 0);
+    stage_x_slot(1, ch0, 1);
+    stage_x_slot(2, ch0, 2);
+  } else {
+    stage(0, ch0);
+  }
   __syncthreads();
 
   for (int ch = ch0; ch < ch1; ++ch) {
     const int buf = (ch - ch0) & 1;
     const bool more = ch + 1 < ch1;
-    if (more) stage(buf ^ 1, ch + 1);
-    const T16* base = lds + buf * (TILE_A + 3 * TILE_X);
+    // RING: does the NEXT chunk start a new image (ho wrap)?  Then the
+    // shift-by-one identity breaks and all three tiles restage.
+    const bool wrap =
+        RING && ((int)((((long long)ch * BMC) >> lgWO) &
+                       ((1 << lgHO) - 1)) == (1 << lgHO) - 1);
+    if (more) {
+      if (RING) {
+        stage_dy_ring(buf ^ 1, ch + 1);
+        if (!wrap) stage_x_slot((r0 + 3) & 3, ch + 1, 2);
+      } else {
+        stage(buf ^ 1, ch + 1);
+      }
+    }
+    const T16* base = RING ? lds + buf * TILE_A
+                           : lds + buf * (TILE_A + 3 * TILE_X);
 
     // A fragments (dy): 4 tr reads issued, drained by the FIRST counted
     // wait of the B pipeline below (not a full lgkmcnt(0) drain)
@@ -1875,7 +1953,10 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
         const int rt = TSPLIT == 3 ? rtw : idx / 6;
         const int rem = idx % 6, s2 = rem >> 1, kh = rem & 1;
         const T16* timg =
-            base + TILE_A + rt * TILE_X + (img_sel + (wn >> 4)) * IMG_X;
+            RING ? lds + 2 * TILE_A + ((r0 + rt) & 3) * TILE_X +
+                       (img_sel + (wn >> 4)) * IMG_X
+                 : base + TILE_A + rt * TILE_X +
+                       (img_sel + (wn >> 4)) * IMG_X;
         const unsigned b0 =
             LDS_BYTE(timg) + (unsigned)((qw[kh] + s2) * 32) + lane_b;
         asm volatile(
@@ -1909,6 +1990,19 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
       }
     }
     if (more) __syncthreads();
+    if (RING && more) {
+      if (wrap) {
+        // new image: restage all three rt tiles (slots now all free) and
+        // reset the rotation — one extra barrier every HO chunks
+        stage_x_slot(0, ch + 1, 0);
+        stage_x_slot(1, ch + 1, 1);
+        stage_x_slot(2, ch + 1, 2);
+        r0 = 0;
+        __syncthreads();
+      } else {
+        r0 = (r0 + 1) & 3;
+      }
+    }
   }
 
   // ---- writeback (32x32x16 C/D layout; TSPLIT: only this wave's taps) ----
@@ -1976,7 +2070,20 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
       // blocks interleave around each other's glds drains.  Kept behind
       // PDT_WGRAD_TSPLIT=3 as a documented negative result.
       static const char* e_ts = getenv("PDT_WGRAD_TSPLIT");
-      if (e_ts && e_ts[0] == '3')
+      // rolling-ring x reuse (WO==32): default ON — L1 wgrad 1.30 ->
+      // 0.94 ms (+38%), r18 bench 114.9k -> 118.8k same-box.
+      // PDT_WGRAD_RING=0 reverts.
+      static const char* e_rg = getenv("PDT_WGRAD_RING");
+      const bool ring = !(e_rg && e_rg[0] == '0') && lgWO == 5 && lgHO >= 1;
+      if (ring) {
+        hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16, 1, true>), grid,
+                           dim3(g16::THREADS), 0, stream,
+                           reinterpret_cast<const t16*>(dy.data_ptr()),
+                           reinterpret_cast<const t16*>(x.data_ptr()),
+                           dw.data_ptr<float>(),
+                           reinterpret_cast<const t16*>(zp.data_ptr()), M,
+                           Kout, Cin, ldc, cm, lgWO, lgHO, (int)grid.z);
+      } else if (e_ts && e_ts[0] == '3')
         hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16, 3>), grid,
                            dim3(768), 0, stream,
                            reinterpret_cast<const t16*>(dy.data_ptr()),

@@ -201,7 +201,9 @@ def test_conv2d_dgrad_bf16(n, h, c, k, r, stride, pad):
      (2, 4, 64, 64, 3, 1, 1), (2, 8, 128, 64, 3, 1, 1), (3, 12, 64, 64, 3, 1, 1),
      # generic RxS taps-on-J path (MODE_CONVJ): the ResNet-50 stem shape
      # (7x7 s2 p3, Cin-padded-8) and a 5x5 to cover non-square tap counts
-     (2, 28, 8, 64, 7, 2, 3), (2, 14, 16, 32, 5, 1, 2)],
+     (2, 28, 8, 64, 7, 2, 3), (2, 14, 16, 32, 5, 1, 2),
+     # WO=32 (one output row per seg chunk): the rolling-ring reuse domain
+     (4, 32, 64, 64, 3, 1, 1)],
 )
 def test_conv2d_wgrad_bf16(n, h, c, k, r, stride, pad):
     ho = (h + 2 * pad - r) // stride + 1
