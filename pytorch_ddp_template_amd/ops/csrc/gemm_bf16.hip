@@ -1711,18 +1711,20 @@ namespace g16 {
 // of 144 lifts occupancy 2 -> 3 waves/SIMD, and each wave reads only its
 // own rt's x tiles (per-CU tr-read traffic -40%).  dw writes were already
 // atomicAdd, so the rt-partials need no extra reduction.
-// RING (WO==32 only, TSPLIT==1): one chunk = exactly one output row, so
-// chunk ch+1's kernel-row-r x tile IS chunk ch's row r+1 tile — a 4-slot
-// ring restages ONE tile per chunk (12 glds) instead of three (28),
-// except at image boundaries (ho wrap: full 3-tile restage behind an
-// extra barrier, 1/HO of chunks).
-template <typename T16, int TSPLIT = 1, bool RING = false>
+// RSHIFT (TSPLIT==1 only): rolling x-tile ring.  One chunk advances the
+// output row by RSHIFT (= 32/WO), so chunk ch+1's kernel-row-r tile IS
+// chunk ch's row r+RSHIFT tile — a (3+RSHIFT)-slot ring restages RSHIFT
+// tiles per chunk instead of three (WO=32: 12 glds vs 28; WO=16: 20),
+// except at image boundaries (full restage behind an extra barrier).
+template <typename T16, int TSPLIT = 1, int RSHIFT = 0>
 __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
                              TSPLIT == 3 ? 1 : 2) void gemm_wgrad_seg_kernel(
     const T16* __restrict__ dy, const T16* __restrict__ x,
     float* __restrict__ dw, const T16* __restrict__ zpad, int Mtot,
     int I /*Kout*/, int J /*Cin*/, long long ldc, ConvMeta cm, int lgWO,
     int lgHO, int zsplit) {
+  constexpr bool RING = RSHIFT > 0;
+  constexpr int SLOTS = 3 + RSHIFT;
   constexpr int BI = 64, BJ = 64, BMC = 32;
   constexpr int IMG_A = 32 * 16;   // dy images: [32 m][16 ch]
   constexpr int IMG_X = 64 * 16;   // x images: [<=48 q rows][16 ch], padded
@@ -1733,7 +1735,8 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
   // RING layout: [2 dy bufs][4 x rt-slots] = 40 KB; else the classic
   // [2][dy + 3 x tiles] = 56 KB
   __shared__ __attribute__((aligned(16)))
-      T16 lds[RING ? (2 * TILE_A + 4 * TILE_X) : 2 * (TILE_A + 3 * TILE_X)];
+      T16 lds[RING ? (2 * TILE_A + SLOTS * TILE_X)
+                   : 2 * (TILE_A + 3 * TILE_X)];
 
   // XCD-contiguous remap (private per-XCD L2s — see gemm_wgrad_tr_kernel):
   // blocks sharing a zidx read the same dy/x chunks; keep them on one L2.
@@ -1902,12 +1905,16 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
     // RING: does the NEXT chunk start a new image (ho wrap)?  Then the
     // shift-by-one identity breaks and all three tiles restage.
     const bool wrap =
-        RING && ((int)((((long long)ch * BMC) >> lgWO) &
+        RING && ((int)((((long long)ch * BMC + BMC - 1) >> lgWO) &
                        ((1 << lgHO) - 1)) == (1 << lgHO) - 1);
     if (more) {
       if (RING) {
         stage_dy_ring(buf ^ 1, ch + 1);
-        if (!wrap) stage_x_slot((r0 + 3) & 3, ch + 1, 2);
+        if (!wrap) {
+#pragma unroll
+          for (int rr = 3 - RSHIFT; rr < 3; ++rr)
+            stage_x_slot((r0 + rr + RSHIFT) % SLOTS, ch + 1, rr);
+        }
       } else {
         stage(buf ^ 1, ch + 1);
       }
@@ -1953,7 +1960,7 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
         const int rt = TSPLIT == 3 ? rtw : idx / 6;
         const int rem = idx % 6, s2 = rem >> 1, kh = rem & 1;
         const T16* timg =
-            RING ? lds + 2 * TILE_A + ((r0 + rt) & 3) * TILE_X +
+            RING ? lds + 2 * TILE_A + ((r0 + rt) % SLOTS) * TILE_X +
                        (img_sel + (wn >> 4)) * IMG_X
                  : base + TILE_A + rt * TILE_X +
                        (img_sel + (wn >> 4)) * IMG_X;
@@ -2000,7 +2007,7 @@ __global__ __launch_bounds__(TSPLIT == 3 ? 768 : THREADS,
         r0 = 0;
         __syncthreads();
       } else {
-        r0 = (r0 + 1) & 3;
+        r0 = (r0 + RSHIFT) % SLOTS;
       }
     }
   }
@@ -2074,9 +2081,18 @@ torch::Tensor conv2d_wgrad_bf16(torch::Tensor dy, torch::Tensor x,
       // 0.94 ms (+38%), r18 bench 114.9k -> 118.8k same-box.
       // PDT_WGRAD_RING=0 reverts.
       static const char* e_rg = getenv("PDT_WGRAD_RING");
-      const bool ring = !(e_rg && e_rg[0] == '0') && lgWO == 5 && lgHO >= 1;
-      if (ring) {
-        hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16, 1, true>), grid,
+      const bool ring =
+          !(e_rg && e_rg[0] == '0') && lgWO >= 4 && lgHO >= 1;
+      if (ring && lgWO == 5) {
+        hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16, 1, 1>), grid,
+                           dim3(g16::THREADS), 0, stream,
+                           reinterpret_cast<const t16*>(dy.data_ptr()),
+                           reinterpret_cast<const t16*>(x.data_ptr()),
+                           dw.data_ptr<float>(),
+                           reinterpret_cast<const t16*>(zp.data_ptr()), M,
+                           Kout, Cin, ldc, cm, lgWO, lgHO, (int)grid.z);
+      } else if (ring && lgWO == 4) {
+        hipLaunchKernelGGL((g16::gemm_wgrad_seg_kernel<t16, 1, 2>), grid,
                            dim3(g16::THREADS), 0, stream,
                            reinterpret_cast<const t16*>(dy.data_ptr()),
                            reinterpret_cast<const t16*>(x.data_ptr()),
